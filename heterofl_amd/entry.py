@@ -1,0 +1,162 @@
+"""Shared entry-point machinery: CLI flags mirroring every cfg key, the
+experiment loop, checkpointing with the reference's exact layout
+(./output/model/{tag}_checkpoint.pt + {tag}_best.pt, reference:
+src/train_classifier_fed.py:37-96).
+"""
+import argparse
+import ast
+import os
+import shutil
+
+import torch
+
+from .config import cfg as global_cfg
+from .control import process_control, CONTROL_FIELDS
+from .data import fetch_dataset, split_dataset
+from .fed import FedRunner
+from .logger import Logger
+from .metrics import Metric
+from .models import make_model
+from .utils import (save, process_dataset, make_optimizer, make_scheduler,
+                    resume, model_tag_of)
+
+
+def build_parser(cfg):
+    parser = argparse.ArgumentParser(description='cfg')
+    for k, v in cfg.items():
+        if isinstance(v, (dict, list, bool)):
+            parser.add_argument(f'--{k}', default=v, type=_literal)
+        else:
+            parser.add_argument(f'--{k}', default=v, type=type(v) if v is not None else str)
+    parser.add_argument('--control_name', default=None, type=str)
+    parser.add_argument('--synthetic', default=0, type=int,
+                        help='use deterministic synthetic data (no-network envs)')
+    return parser
+
+
+def _literal(x):
+    if isinstance(x, str):
+        try:
+            return ast.literal_eval(x)
+        except (ValueError, SyntaxError):
+            return x
+    return x
+
+
+def parse_args(argv=None, cfg=None):
+    cfg = dict(global_cfg if cfg is None else cfg)
+    parser = build_parser(cfg)
+    args = vars(parser.parse_args(argv))
+    for k in cfg:
+        cfg[k] = args[k]
+    if args.get('control_name') and args['control_name'] != 'None':
+        cfg['control'] = dict(zip(CONTROL_FIELDS, args['control_name'].split('_')))
+    cfg['control_name'] = '_'.join([str(cfg['control'][k]) for k in cfg['control']])
+    cfg['synthetic'] = bool(args.get('synthetic', 0))
+    return cfg
+
+
+def run_fed_experiment(cfg, pivot_metric, pivot_sign, metric_name):
+    """One full federated experiment per seed (reference:
+    src/train_classifier_fed.py:48-96).  pivot_sign=+1 maximizes (accuracy),
+    -1 minimizes (perplexity)."""
+    cfg['pivot_metric'] = pivot_metric
+    cfg['metric_name'] = metric_name
+    process_control(cfg)
+    seeds = list(range(cfg['init_seed'], cfg['init_seed'] + cfg['num_experiments']))
+    for seed in seeds:
+        cfg['model_tag'] = model_tag_of(seed, cfg)
+        cfg['pivot'] = -float('inf') * pivot_sign if pivot_sign > 0 else float('inf')
+        print('Experiment: {}'.format(cfg['model_tag']))
+        _run_one(cfg, seed, pivot_sign)
+
+
+def _run_one(cfg, seed, pivot_sign):
+    torch.manual_seed(seed)
+    if torch.cuda.is_available():
+        torch.cuda.manual_seed(seed)
+    if cfg['device'].startswith('cuda') and not torch.cuda.is_available():
+        print('[heterofl_amd] no GPU visible; falling back to cpu')
+        cfg['device'] = 'cpu'
+    dataset = fetch_dataset(cfg['data_name'], cfg['subset'],
+                            synthetic=cfg.get('synthetic', False))
+    process_dataset(dataset, cfg)
+    model = make_model(cfg, model_rate=cfg['global_model_rate']).to(cfg['device'])
+    optimizer = make_optimizer(model, cfg['lr'], cfg)
+    scheduler = make_scheduler(optimizer, cfg)
+    if cfg['resume_mode'] == 1:
+        last_epoch, data_split, label_split, model, optimizer, scheduler, logger = resume(
+            model, cfg['model_tag'], optimizer, scheduler)
+    elif cfg['resume_mode'] == 2:
+        last_epoch = 1
+        _, data_split, label_split, model, _, _, _ = resume(model, cfg['model_tag'])
+        logger = Logger(os.path.join('output', 'runs', cfg['model_tag']))
+    else:
+        last_epoch = 1
+        data_split, label_split = None, None
+        logger = Logger(os.path.join('output', 'runs', 'train_{}'.format(cfg['model_tag'])))
+    if data_split is None:
+        data_split, label_split = split_dataset(
+            dataset, cfg['num_users'], cfg['data_split_mode'],
+            classes_size=cfg.get('classes_size'))
+    runner = FedRunner(cfg, dataset, data_split, label_split, model, optimizer,
+                       logger=logger)
+    for epoch in range(last_epoch, cfg['num_epochs']['global'] + 1):
+        logger.safe(True)
+        runner.train_round(epoch)
+        test_model = runner.stats()
+        runner.test(test_model, epoch)
+        if cfg['scheduler_name'] == 'ReduceLROnPlateau':
+            scheduler.step(metrics=logger.mean['train/{}'.format(cfg['pivot_metric'])])
+        else:
+            scheduler.step()
+        logger.safe(False)
+        save_result = {
+            'cfg': cfg, 'epoch': epoch + 1, 'data_split': data_split,
+            'label_split': label_split, 'model_dict': model.state_dict(),
+            'optimizer_dict': optimizer.state_dict(),
+            'scheduler_dict': scheduler.state_dict(), 'logger': logger}
+        save(save_result, './output/model/{}_checkpoint.pt'.format(cfg['model_tag']))
+        cur = logger.mean['test/{}'.format(cfg['pivot_metric'])]
+        better = cur > cfg['pivot'] if pivot_sign > 0 else cur < cfg['pivot']
+        if better:
+            cfg['pivot'] = cur
+            shutil.copy('./output/model/{}_checkpoint.pt'.format(cfg['model_tag']),
+                        './output/model/{}_best.pt'.format(cfg['model_tag']))
+        logger.reset()
+    logger.safe(False)
+
+
+def run_fed_eval(cfg, metric_name, result_key='test'):
+    """Evaluation entry (reference: src/test_classifier_fed.py:41-60): load
+    {tag}_best.pt, re-run sBN stats (vision), evaluate, save
+    ./output/result/{tag}.pt."""
+    cfg['metric_name'] = metric_name
+    process_control(cfg)
+    seeds = list(range(cfg['init_seed'], cfg['init_seed'] + cfg['num_experiments']))
+    for seed in seeds:
+        cfg['model_tag'] = model_tag_of(seed, cfg)
+        print('Eval: {}'.format(cfg['model_tag']))
+        torch.manual_seed(seed)
+        dataset = fetch_dataset(cfg['data_name'], cfg['subset'],
+                                synthetic=cfg.get('synthetic', False))
+        process_dataset(dataset, cfg)
+        if cfg['device'].startswith('cuda') and not torch.cuda.is_available():
+            cfg['device'] = 'cpu'
+        model = make_model(cfg, model_rate=cfg['global_model_rate']).to(cfg['device'])
+        last_epoch, data_split, label_split, model, _, _, train_logger = resume(
+            model, cfg['model_tag'], load_tag='best', strict=False)
+        if data_split is None:
+            data_split, label_split = split_dataset(
+                dataset, cfg['num_users'], cfg['data_split_mode'],
+                classes_size=cfg.get('classes_size'))
+        logger = Logger(os.path.join('output', 'runs', 'test_{}'.format(cfg['model_tag'])))
+        runner = FedRunner(cfg, dataset, data_split, label_split, model,
+                           make_optimizer(model, cfg['lr'], cfg), logger=logger)
+        logger.safe(True)
+        test_model = runner.stats()
+        runner.test(test_model, last_epoch - 1)
+        logger.safe(False)
+        result = {'cfg': cfg, 'epoch': last_epoch, 'logger': {'train': train_logger,
+                                                              'test': logger}}
+        save(result, './output/result/{}.pt'.format(cfg['model_tag']))

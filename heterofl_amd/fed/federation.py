@@ -1,0 +1,381 @@
+"""Federation core: width-sliced subnetwork distribute / combine.
+
+Re-implements the semantics of the reference Federation (reference:
+src/fed.py:8-298) with a typed index-map representation instead of
+per-parameter torch.meshgrid gathers:
+
+- every slice the HeteroFL rules produce is either FULL (whole axis),
+  PREFIX(n) (the first n indices — conv/resnet channel slicing and
+  transformer non-qkv slicing), or GATHER(tensor) (transformer per-head
+  slicing and label-split-filtered output rows).
+- distribute() turns PREFIX maps into contiguous narrow+clone (no gather);
+- combine() accumulates into an fp32 accumulator + count per parameter and
+  writes the masked average in place into the global tensors, exactly like
+  reference src/fed.py:180-298 (including label_split filtering of the
+  output layers).
+
+The slicing rule sets per architecture mirror reference src/fed.py:26-159,
+including the transformer quirks: per-head q/k/v slicing (fed.py:124-131)
+and the idx_i reset after linear_q/linear_k biases (fed.py:147-151).
+"""
+import math
+from collections import OrderedDict
+
+import torch
+
+FULL = 'full'      # whole axis
+PREFIX = 'prefix'  # first n indices
+GATHER = 'gather'  # explicit index tensor
+
+
+class Axis:
+    """One axis of a parameter slice."""
+    __slots__ = ('kind', 'n', 'idx')
+
+    def __init__(self, kind, n=None, idx=None):
+        self.kind = kind
+        self.n = n
+        self.idx = idx
+
+    @staticmethod
+    def full(n):
+        return Axis(FULL, n=n)
+
+    @staticmethod
+    def prefix(n):
+        return Axis(PREFIX, n=n)
+
+    @staticmethod
+    def gather(idx):
+        return Axis(GATHER, n=int(idx.numel()), idx=idx)
+
+    def indices(self, device=None):
+        """Materialize as an index tensor (for tests / RCCL padding masks)."""
+        if self.kind == GATHER:
+            return self.idx if device is None else self.idx.to(device)
+        return torch.arange(self.n, device=device)
+
+    def is_dense_prefix(self):
+        return self.kind in (FULL, PREFIX)
+
+    def __repr__(self):
+        return f'Axis({self.kind}, n={self.n})'
+
+
+class ParamSlice:
+    """Slice spec for one parameter: out axis and (for dim>1) in axis."""
+    __slots__ = ('out', 'inp')
+
+    def __init__(self, out, inp=None):
+        self.out = out
+        self.inp = inp
+
+    def __repr__(self):
+        return f'ParamSlice(out={self.out}, inp={self.inp})'
+
+
+def _ceil(x):
+    return int(math.ceil(x))
+
+
+class Federation:
+    def __init__(self, global_parameters, rate, label_split, cfg):
+        self.global_parameters = global_parameters
+        self.rate = rate
+        self.label_split = label_split
+        self.cfg = cfg
+        self.model_rate = None
+        self.make_model_rate()
+
+    # ------------------------------------------------------------------ rates
+    def make_model_rate(self, generator=None):
+        """Dynamic mode resamples each user's level per round
+        (reference: src/fed.py:15-24).  Pass a seeded generator to keep the
+        sampling identical across ranks without communication."""
+        cfg = self.cfg
+        if cfg['model_split_mode'] == 'dynamic':
+            proportion = torch.tensor(cfg['proportion'])
+            rate_idx = torch.multinomial(proportion, num_samples=cfg['num_users'],
+                                         replacement=True, generator=generator).tolist()
+            self.model_rate = [cfg['model_rate'][i] for i in rate_idx]
+        elif cfg['model_split_mode'] == 'fix':
+            self.model_rate = list(self.rate)
+        else:
+            raise ValueError('Not valid model split mode')
+
+    # ------------------------------------------------------------- split maps
+    def split_model(self, user_idx):
+        name = self.cfg['model_name']
+        if name == 'conv':
+            return [self._split_conv(u) for u in user_idx]
+        if 'resnet' in name:
+            return [self._split_resnet(u) for u in user_idx]
+        if name == 'transformer':
+            return [self._split_transformer(u) for u in user_idx]
+        raise ValueError('Not valid model name')
+
+    def _scaler_rate(self, user):
+        return self.model_rate[user] / self.cfg['global_model_rate']
+
+    def _split_conv(self, user):
+        """reference: src/fed.py:27-62 — slice output channels of every
+        weight with dim>1; the final classifier keeps full rows."""
+        gp = self.global_parameters
+        weight_keys = [k for k in gp if 'weight' in k]
+        bias_keys = [k for k in gp if 'bias' in k]
+        output_weight_name = weight_keys[-1]
+        output_bias_name = bias_keys[-1]
+        rate = self._scaler_rate(user)
+        idx = OrderedDict()
+        idx_i = None  # running output axis
+        for k, v in gp.items():
+            ptype = k.split('.')[-1]
+            if 'weight' in ptype:
+                if v.dim() > 1:
+                    if idx_i is None:
+                        idx_i = Axis.full(v.size(1))
+                    inp = idx_i
+                    if k == output_weight_name:
+                        out = Axis.full(v.size(0))
+                    else:
+                        out = Axis.prefix(_ceil(v.size(0) * rate))
+                    idx[k] = ParamSlice(out, inp)
+                    idx_i = out
+                else:
+                    idx[k] = ParamSlice(idx_i)
+            elif 'bias' in ptype:
+                idx[k] = ParamSlice(idx_i)
+            # non weight/bias keys are skipped (distributed fully)
+        return idx
+
+    def _split_resnet(self, user):
+        """reference: src/fed.py:63-103 — conv1/conv2 sliced; shortcut reuses
+        conv1's input index; linear keeps full output rows."""
+        gp = self.global_parameters
+        rate = self._scaler_rate(user)
+        idx = OrderedDict()
+        idx_i = None
+        for k, v in gp.items():
+            ptype = k.split('.')[-1]
+            if 'weight' in ptype:
+                if v.dim() > 1:
+                    if 'conv1' in k or 'conv2' in k:
+                        if idx_i is None:
+                            idx_i = Axis.full(v.size(1))
+                        inp = idx_i
+                        out = Axis.prefix(_ceil(v.size(0) * rate))
+                        idx_i = out
+                    elif 'shortcut' in k:
+                        inp = idx[k.replace('shortcut', 'conv1')].inp
+                        out = idx_i
+                    elif 'linear' in k:
+                        inp = idx_i
+                        out = Axis.full(v.size(0))
+                    else:
+                        raise ValueError('Not valid k')
+                    idx[k] = ParamSlice(out, inp)
+                else:
+                    idx[k] = ParamSlice(idx_i)
+            elif 'bias' in ptype:
+                if 'linear' in k:
+                    idx[k] = ParamSlice(Axis.full(v.size(0)))
+                else:
+                    idx[k] = ParamSlice(idx_i)
+        return idx
+
+    def _split_transformer(self, user):
+        """reference: src/fed.py:104-156 — embedding slices the embedding dim
+        keeping full vocab rows; q/k/v slice per head; decoder.linear2 keeps
+        full vocab output; idx_i resets to the layer input after linear_q /
+        linear_k biases so the residual stream stays at the sliced width."""
+        gp = self.global_parameters
+        num_heads = self.cfg['transformer']['num_heads']
+        rate = self._scaler_rate(user)
+        idx = OrderedDict()
+        idx_i = None
+        for k, v in gp.items():
+            ptype = k.split('.')[-1]
+            if 'weight' in ptype:
+                if v.dim() > 1:
+                    if 'embedding' in k.split('.')[-2]:
+                        out = Axis.full(v.size(0))
+                        inp = Axis.prefix(_ceil(v.size(1) * rate))
+                        idx[k] = ParamSlice(out, inp)
+                        idx_i = inp
+                    elif 'decoder' in k and 'linear2' in k:
+                        idx[k] = ParamSlice(Axis.full(v.size(0)), idx_i)
+                    elif ('linear_q' in k) or ('linear_k' in k) or ('linear_v' in k):
+                        inp = idx_i
+                        head_dim = v.size(0) // num_heads
+                        local_hd = _ceil(head_dim * rate)
+                        out_idx = torch.arange(v.size(0)).reshape(
+                            num_heads, -1)[:, :local_hd].reshape(-1)
+                        out = Axis.gather(out_idx)
+                        idx[k] = ParamSlice(out, inp)
+                        idx_i = out
+                    else:
+                        inp = idx_i
+                        out = Axis.prefix(_ceil(v.size(0) * rate))
+                        idx[k] = ParamSlice(out, inp)
+                        idx_i = out
+                else:
+                    idx[k] = ParamSlice(idx_i)
+            elif 'bias' in ptype:
+                if 'decoder' in k and 'linear2' in k:
+                    idx[k] = ParamSlice(Axis.full(v.size(0)))
+                elif ('linear_q' in k) or ('linear_k' in k) or ('linear_v' in k):
+                    idx[k] = ParamSlice(idx_i)
+                    if 'linear_v' not in k:
+                        # reset the running index to this layer's INPUT axis
+                        # (reference: src/fed.py:147-151)
+                        idx_i = idx[k.replace('bias', 'weight')].inp
+                else:
+                    idx[k] = ParamSlice(idx_i)
+        return idx
+
+    # -------------------------------------------------------------- transport
+    @staticmethod
+    def _slice_value(v, ps):
+        """Extract the slice a client receives."""
+        if ps is None:
+            return v.clone()
+        if v.dim() > 1 and ps.inp is not None:
+            out, inp = ps.out, ps.inp
+            x = v
+            if out.kind == PREFIX:
+                x = x.narrow(0, 0, out.n)
+            elif out.kind == GATHER:
+                x = x.index_select(0, out.idx.to(v.device))
+            if inp.kind == PREFIX:
+                x = x.narrow(1, 0, inp.n)
+            elif inp.kind == GATHER:
+                x = x.index_select(1, inp.idx.to(v.device))
+            return x.clone()
+        out = ps.out
+        if out.kind == FULL:
+            return v.clone()
+        if out.kind == PREFIX:
+            return v.narrow(0, 0, out.n).clone()
+        return v.index_select(0, out.idx.to(v.device)).clone()
+
+    def distribute(self, user_idx, resample=True, generator=None):
+        """reference: src/fed.py:161-178 — resample rates, build index maps,
+        and hand each client its parameter slices.  Set resample=False when
+        the engine already resampled rates (e.g. with a shared seeded
+        generator for multi-rank determinism)."""
+        if resample:
+            self.make_model_rate(generator)
+        param_idx = self.split_model(user_idx)
+        local_parameters = []
+        for m in range(len(user_idx)):
+            lp = OrderedDict()
+            for k, v in self.global_parameters.items():
+                ptype = k.split('.')[-1]
+                if 'weight' in ptype or 'bias' in ptype:
+                    lp[k] = self._slice_value(v, param_idx[m][k])
+                else:
+                    lp[k] = v.clone()
+            local_parameters.append(lp)
+        return local_parameters, param_idx
+
+    # --------------------------------------------------------------- combine
+    def _output_layer_kind(self, k, weight_keys, bias_keys):
+        """Which parameters get label-split row filtering in combine
+        (reference: src/fed.py:193-198, 228-233, 263-274)."""
+        name = self.cfg['model_name']
+        if name == 'conv':
+            if k == weight_keys[-1] or k == bias_keys[-1]:
+                return True
+        elif 'resnet' in name:
+            if 'linear' in k:
+                return True
+        elif name == 'transformer':
+            if k.split('.')[-2] == 'embedding' and k.split('.')[-1] == 'weight':
+                return True
+            if 'decoder' in k and 'linear2' in k:
+                return True
+        return False
+
+    @staticmethod
+    def _accumulate(tmp, cnt, ps, val):
+        """tmp[slice] += val; cnt[slice] += 1 for one client's parameter."""
+        if ps is None or ps.out is None:
+            tmp += val
+            cnt += 1
+            return
+        out, inp = ps.out, ps.inp
+        if inp is not None and tmp.dim() > 1:
+            if out.is_dense_prefix() and inp.is_dense_prefix():
+                tmp[:out.n, :inp.n] += val
+                cnt[:out.n, :inp.n] += 1
+            elif inp.is_dense_prefix():
+                oidx = out.indices(tmp.device)
+                view_t = tmp.narrow(1, 0, inp.n)
+                view_c = cnt.narrow(1, 0, inp.n)
+                view_t.index_add_(0, oidx, val.to(view_t.dtype))
+                view_c.index_add_(0, oidx, torch.ones_like(val, dtype=view_c.dtype))
+            else:
+                oidx = out.indices(tmp.device).unsqueeze(1)
+                iidx = inp.indices(tmp.device).unsqueeze(0)
+                tmp.index_put_((oidx, iidx), val.to(tmp.dtype), accumulate=True)
+                cnt.index_put_((oidx, iidx),
+                               torch.ones_like(val, dtype=cnt.dtype), accumulate=True)
+        else:
+            if out.is_dense_prefix():
+                tmp[:out.n] += val
+                cnt[:out.n] += 1
+            else:
+                oidx = out.indices(tmp.device)
+                tmp.index_add_(0, oidx, val.to(tmp.dtype))
+                cnt.index_add_(0, oidx, torch.ones_like(val, dtype=cnt.dtype))
+
+    def accumulate(self, local_parameters, param_idx, user_idx, slots=None):
+        """Build the fp32 zero-padded accumulator + count for a subset of the
+        round's clients.  ``slots`` indexes into user_idx/param_idx;
+        local_parameters maps slot -> state_dict (a list works too).
+
+        This is the per-rank half of the padded-RCCL combine (SURVEY §2b
+        K14): accumulators are global-shaped, so summing them across ranks
+        and dividing reproduces the sequential combine exactly.
+        """
+        gp = self.global_parameters
+        weight_keys = [k for k in gp if 'weight' in k]
+        bias_keys = [k for k in gp if 'bias' in k]
+        if slots is None:
+            slots = range(len(param_idx))
+        tmp_d, cnt_d = OrderedDict(), OrderedDict()
+        for k, v in gp.items():
+            ptype = k.split('.')[-1]
+            tmp = torch.zeros(v.size(), dtype=torch.float32, device=v.device)
+            cnt = torch.zeros(v.size(), dtype=torch.float32, device=v.device)
+            is_output = self._output_layer_kind(k, weight_keys, bias_keys)
+            for m in slots:
+                val = local_parameters[m][k]
+                if 'weight' in ptype or 'bias' in ptype:
+                    ps = param_idx[m][k]
+                    if is_output and ps is not None and ps.out is not None:
+                        label_split = torch.tensor(self.label_split[user_idx[m]])
+                        out_idx = ps.out.indices()[label_split]
+                        ps = ParamSlice(Axis.gather(out_idx), ps.inp)
+                        val = val[label_split]
+                    self._accumulate(tmp, cnt, ps, val.float())
+                else:
+                    tmp += val.float()
+                    cnt += 1
+            tmp_d[k] = tmp
+            cnt_d[k] = cnt
+        return tmp_d, cnt_d
+
+    def finalize(self, tmp_d, cnt_d):
+        """Masked count-weighted average written in place into the global
+        tensors (reference: src/fed.py:217-218)."""
+        for k, v in self.global_parameters.items():
+            cnt = cnt_d[k]
+            mask = cnt > 0
+            avg = tmp_d[k] / cnt.clamp(min=1)
+            v.copy_(torch.where(mask, avg, v.float()).to(v.dtype))
+
+    def combine(self, local_parameters, param_idx, user_idx):
+        """Sequential (single-process) combine (reference: src/fed.py:180-298)."""
+        tmp_d, cnt_d = self.accumulate(local_parameters, param_idx, user_idx)
+        self.finalize(tmp_d, cnt_d)

@@ -1,0 +1,186 @@
+"""Federated round orchestration.
+
+Drives the reference round structure (src/train_classifier_fed.py:74-94):
+sample active clients -> distribute slices -> local training -> combine ->
+sBN statistics pass -> evaluation -> checkpoint.  The local-training engine
+is pluggable: 'sequential' (reference-faithful oracle) or 'batched'
+(MI355X fast path, clients of one rate trained in one grouped model).
+
+Multi-rank operation (one process per GPU over RCCL): active clients are
+sharded across ranks; every rank samples identical user_idx and dynamic
+rates from a round-seeded generator, trains its shard, and combine runs as
+a padded all-reduce (see parallel/dist.py).
+"""
+import math
+
+import numpy as np
+import torch
+
+from ..data import SplitDataset, BatchDataset, make_data_loader
+from ..metrics import Metric
+from ..models import make_model
+from ..utils import collate, to_device
+from .federation import Federation
+from .sequential import SequentialClientTrainer
+
+
+def sample_active_users(cfg, epoch=None, generator=None):
+    """ceil(frac*num_users) clients by randperm
+    (reference: src/train_classifier_fed.py:172-181).  With a generator the
+    sampling is reproducible across ranks and GPU counts."""
+    num_active = int(np.ceil(cfg['frac'] * cfg['num_users']))
+    perm = torch.randperm(cfg['num_users'], generator=generator)
+    return perm[:num_active].tolist()
+
+
+class FedRunner:
+    def __init__(self, cfg, dataset, data_split, label_split, model, optimizer,
+                 logger=None, dist_ctx=None):
+        self.cfg = cfg
+        self.dataset = dataset
+        self.data_split = data_split
+        self.label_split = label_split
+        self.global_model = model
+        self.optimizer = optimizer
+        self.logger = logger
+        self.dist_ctx = dist_ctx  # parallel.DistContext or None
+        self.federation = Federation(model.state_dict(), cfg['model_rate'],
+                                     label_split, cfg)
+        self.is_lm = cfg['model_name'] == 'transformer'
+        if cfg.get('engine', 'sequential') == 'batched':
+            from .batched import BatchedClientTrainer
+            self.trainer = BatchedClientTrainer(cfg)
+        else:
+            self.trainer = SequentialClientTrainer(cfg)
+        self._round_gen = None
+
+    # ------------------------------------------------------------------ round
+    def _round_generator(self, epoch):
+        """Seeded per round so all ranks agree on sampling without
+        communication (SURVEY §7 reproducibility contract)."""
+        seed = int(self.cfg.get('round_seed_base', self.cfg.get('init_seed', 0)))
+        g = torch.Generator()
+        g.manual_seed((seed * 1000003 + epoch) % (2 ** 63 - 1))
+        return g
+
+    def _make_loader(self, user):
+        cfg = self.cfg
+        if self.is_lm:
+            # LM clients own rows of the batchified token matrix
+            # (reference: src/train_transformer_fed.py:158-162)
+            return BatchDataset(SplitDataset(self.dataset['train'],
+                                             self.data_split['train'][user]),
+                                cfg['bptt'])
+        loader = make_data_loader(
+            {'train': SplitDataset(self.dataset['train'],
+                                   self.data_split['train'][user])}, cfg)['train']
+        return loader
+
+    def train_round(self, epoch):
+        cfg = self.cfg
+        self.global_model.load_state_dict(self.federation.global_parameters)
+        self.global_model.train(True)
+        lr = self.optimizer.param_groups[0]['lr']
+
+        g = self._round_generator(epoch)
+        if cfg['world_size'] > 1 or self.dist_ctx is not None:
+            user_idx = sample_active_users(cfg, epoch, generator=g)
+            self.federation.make_model_rate(generator=g)
+            resample = False
+        else:
+            user_idx = sample_active_users(cfg, epoch)
+            resample = True
+        local_parameters, param_idx = self.federation.distribute(
+            user_idx, resample=resample)
+
+        # shard clients across ranks (identity shard when single rank)
+        rank, world = (0, 1) if self.dist_ctx is None else \
+            (self.dist_ctx.rank, self.dist_ctx.world_size)
+        my_clients = list(range(rank, len(user_idx), world))
+
+        trained = dict(self.trainer.train_clients(
+            my_clients, user_idx, local_parameters, self.federation.model_rate,
+            self._make_loader, self.label_split, lr, self.logger))
+
+        if self.dist_ctx is None:
+            ordered = [trained[m] for m in range(len(user_idx))]
+            self.federation.combine(ordered, param_idx, user_idx)
+        else:
+            from ..parallel.dist import distributed_combine
+            distributed_combine(self.federation, trained, param_idx, user_idx,
+                                self.dist_ctx)
+        self.global_model.load_state_dict(self.federation.global_parameters)
+        return user_idx
+
+    # ------------------------------------------------------------------ stats
+    def stats(self):
+        """sBN statistics pass: rebuild the global model with tracked running
+        stats and run the full train set in train mode under no_grad
+        (reference: src/train_classifier_fed.py:127-138)."""
+        cfg = self.cfg
+        if self.is_lm:
+            return self.global_model  # LM path has no sBN pass (reference:
+            # src/train_transformer_fed.py:77)
+        with torch.no_grad():
+            test_model = make_model(cfg, model_rate=cfg['global_model_rate'],
+                                    track=True).to(cfg['device'])
+            test_model.load_state_dict(self.global_model.state_dict(), strict=False)
+            test_model.train(True)
+            loader = make_data_loader({'train': self.dataset['train']}, cfg)['train']
+            for input in loader:
+                input = collate(input)
+                input = to_device(input, cfg['device'])
+                test_model(input)
+        return test_model
+
+    # ------------------------------------------------------------------- test
+    def test(self, test_model, epoch):
+        """Per-client Local metrics + Global metrics
+        (reference: src/train_classifier_fed.py:141-169)."""
+        cfg = self.cfg
+        metric = Metric()
+        logger = self.logger
+        with torch.no_grad():
+            test_model.train(False)
+            if not self.is_lm:
+                for m in range(cfg['num_users']):
+                    loader = make_data_loader(
+                        {'test': SplitDataset(self.dataset['test'],
+                                              self.data_split['test'][m])}, cfg)['test']
+                    for input in loader:
+                        input = collate(input)
+                        input_size = input['label'].size(0)
+                        input['label_split'] = torch.tensor(self.label_split[m])
+                        input = to_device(input, cfg['device'])
+                        output = test_model(input)
+                        ev = metric.evaluate(cfg['metric_name']['test']['Local'],
+                                             input, output)
+                        if logger:
+                            logger.append(ev, 'test', input_size)
+                loader = make_data_loader({'test': self.dataset['test']}, cfg)['test']
+                for input in loader:
+                    input = collate(input)
+                    input_size = input['label'].size(0)
+                    input = to_device(input, cfg['device'])
+                    output = test_model(input)
+                    ev = metric.evaluate(cfg['metric_name']['test']['Global'],
+                                         input, output)
+                    if logger:
+                        logger.append(ev, 'test', input_size)
+            else:
+                ds = BatchDataset(self.dataset['test'], cfg['bptt'])
+                for i in range(len(ds)):
+                    input = ds[i]
+                    input_size = input['label'].size(0)
+                    input = to_device(input, cfg['device'])
+                    output = test_model(input)
+                    ev = metric.evaluate(cfg['metric_name']['test']['Global'],
+                                         input, output)
+                    if logger:
+                        logger.append(ev, 'test', input_size)
+            if logger:
+                info = {'info': ['Test Epoch: {}({:.0f}%)'.format(epoch, 100.)]}
+                logger.append(info, 'test', mean=False)
+                names = cfg['metric_name']['test']
+                flat = sum(names.values(), []) if isinstance(names, dict) else names
+                logger.write('test', flat)

@@ -1,0 +1,91 @@
+"""control_name grammar tests against hand-computed expansions
+(reference semantics: src/utils.py:113-215)."""
+import numpy as np
+import pytest
+
+from heterofl_amd.config import default_config
+from heterofl_amd.control import (process_control, parse_control_name,
+                                  parse_model_mode, MODEL_SPLIT_RATE,
+                                  CONTROL_FIELDS)
+
+
+def _cfg(control_name, data_name='CIFAR10'):
+    cfg = default_config()
+    cfg['control'] = parse_control_name(control_name)
+    cfg['data_name'] = data_name
+    process_control(cfg)
+    return cfg
+
+
+def test_rates():
+    assert MODEL_SPLIT_RATE == {'a': 1, 'b': 0.5, 'c': 0.25, 'd': 0.125, 'e': 0.0625}
+
+
+def test_parse_fields():
+    c = parse_control_name('1_100_0.1_iid_fix_a1-e1_bn_1_1')
+    assert c['num_users'] == '100' and c['frac'] == '0.1'
+    assert c['model_mode'] == 'a1-e1' and c['norm'] == 'bn'
+    with pytest.raises(ValueError):
+        parse_control_name('1_100_0.1_iid')
+
+
+def test_fix_partition_a1_e1():
+    cfg = _cfg('1_100_0.1_iid_fix_a1-e1_bn_1_1')
+    # 100 users // (1+1) = 50 per term: first 50 at rate 1, next 50 at 1/16
+    assert cfg['model_rate'][:50] == [1] * 50
+    assert cfg['model_rate'][50:] == [0.0625] * 50
+    assert cfg['global_model_rate'] == 1
+
+
+def test_fix_partition_remainder():
+    cfg = _cfg('1_10_0.5_iid_fix_a1-b2_bn_1_1')
+    # 10 // 3 = 3 per unit: 3 at a, 6 at b, 1 leftover gets last rate (b)
+    assert cfg['model_rate'] == [1] * 3 + [0.5] * 7
+
+
+def test_dynamic_proportion():
+    cfg = _cfg('1_100_0.1_iid_dynamic_a1-b1-c2_bn_1_1')
+    assert cfg['model_rate'] == [1, 0.5, 0.25]
+    assert np.allclose(cfg['proportion'], [0.25, 0.25, 0.5])
+
+
+def test_hyperparameters_cifar_iid():
+    cfg = _cfg('1_100_0.1_iid_fix_a1_bn_1_1', 'CIFAR10')
+    assert cfg['lr'] == 0.1 and cfg['optimizer_name'] == 'SGD'
+    assert cfg['num_epochs'] == {'global': 400, 'local': 5}
+    assert cfg['batch_size'] == {'train': 10, 'test': 50}
+    assert cfg['milestones'] == [150, 250]
+    assert cfg['data_shape'] == [3, 32, 32]
+
+
+def test_hyperparameters_cifar_noniid():
+    cfg = _cfg('1_100_0.1_non-iid-2_fix_a1_gn_1_1', 'CIFAR10')
+    assert cfg['num_epochs'] == {'global': 800, 'local': 5}
+    assert cfg['milestones'] == [300, 500]
+    assert cfg['norm'] == 'gn'
+
+
+def test_hyperparameters_mnist():
+    cfg = _cfg('1_100_0.1_iid_fix_a1_bn_1_1', 'MNIST')
+    assert cfg['lr'] == 1e-2
+    assert cfg['num_epochs'] == {'global': 200, 'local': 5}
+
+
+def test_hyperparameters_wikitext2():
+    cfg = _cfg('1_100_0.1_iid_fix_a1-e1_bn_1_1', 'WikiText2')
+    assert cfg['bptt'] == 64 and cfg['mask_rate'] == 0.15
+    assert cfg['num_epochs'] == {'global': 200, 'local': 1}
+    assert cfg['batch_size'] == {'train': 100, 'test': 10}
+    assert cfg['transformer'] == {'embedding_size': 256, 'num_heads': 8,
+                                  'hidden_size': 512, 'num_layers': 4,
+                                  'dropout': 0.2}
+
+
+def test_scale_mask_flags():
+    cfg = _cfg('1_10_0.1_iid_fix_a1_bn_0_0')
+    assert cfg['scale'] is False and cfg['mask'] is False
+
+
+def test_model_mode_parse():
+    rates, props = parse_model_mode('a5-b10-e1')
+    assert rates == [1, 0.5, 0.0625] and props == [5, 10, 1]

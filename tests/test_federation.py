@@ -1,0 +1,212 @@
+"""Federation slicing and aggregation tests.
+
+Oracle: the slicing rules at reference src/fed.py:26-159 are
+exactly specifiable — these tests pin them with hand-computed index maps,
+the distribute->combine round-trip property, and count-weighted averaging
+identities.  (The implementation was additionally verified once against the
+reference's own Federation on identical state dicts: exact match for conv,
+resnet18 and transformer.)
+"""
+import math
+
+import numpy as np
+import pytest
+import torch
+
+from heterofl_amd.fed import Federation
+from heterofl_amd.fed.federation import PREFIX, FULL, GATHER
+from heterofl_amd.models import make_model
+from tests.conftest import make_cfg
+
+
+def _fed(cfg, model, rates=None, label_split=None):
+    rates = rates if rates is not None else cfg['model_rate']
+    num_users = cfg['num_users']
+    if label_split is None:
+        label_split = {i: list(range(10)) for i in range(num_users)}
+    return Federation(model.state_dict(), rates, label_split, cfg)
+
+
+# ------------------------------------------------------------------- resnet
+def test_resnet_split_rules(base_cfg):
+    cfg = make_cfg(base_cfg, '1_4_1_iid_fix_a1-b1-c1-e1_bn_1_1',
+                   data_name='CIFAR10', model_name='resnet18')
+    model = make_model(cfg, model_rate=1)
+    fed = _fed(cfg, model)
+    idx = fed.split_model([1])  # user 1 -> rate 0.5
+    m = idx[0]
+    # stem conv: out prefix ceil(64*0.5)=32, input full 3
+    ps = m['conv1.weight']
+    assert ps.out.kind == PREFIX and ps.out.n == 32
+    assert ps.inp.kind == FULL and ps.inp.n == 3
+    # first block conv1: 32 in, 32 out
+    ps = m['layer1.0.conv1.weight']
+    assert ps.inp.n == 32 and ps.out.n == 32
+    # layer2.0 downsamples 64->128: sliced to 32->64
+    ps = m['layer2.0.conv1.weight']
+    assert ps.inp.n == 32 and ps.out.n == 64
+    # shortcut reuses conv1's INPUT index; out follows conv2's out
+    ps = m['layer2.0.shortcut.weight']
+    assert ps.inp.n == 32 and ps.out.n == 64
+    # classifier keeps full rows, input sliced
+    ps = m['linear.weight']
+    assert ps.out.kind == FULL and ps.out.n == 10
+    assert ps.inp.n == 256  # ceil(512*0.5)
+    assert m['linear.bias'].out.kind == FULL
+    # norm weight follows the running index of its block input
+    ps = m['layer1.0.n1.weight']
+    assert ps.out.n == 32
+
+
+def test_resnet_split_tiny_rate(base_cfg):
+    # ragged widths: rate 1/16 -> ceil(64/16)=4 channels
+    cfg = make_cfg(base_cfg, '1_2_1_iid_fix_e1_bn_1_1',
+                   data_name='CIFAR10', model_name='resnet18')
+    cfg['global_model_rate'] = 1.0  # slice from a full-width global model
+    model = make_model(cfg, model_rate=1)
+    fed = _fed(cfg, model, rates=[0.0625, 0.0625])
+    m = fed.split_model([0])[0]
+    assert m['conv1.weight'].out.n == 4
+    assert m['layer4.1.conv2.weight'].out.n == 32  # ceil(512/16)
+    assert m['linear.weight'].inp.n == 32
+
+
+# --------------------------------------------------------------------- conv
+def test_conv_split_rules(base_cfg):
+    cfg = make_cfg(base_cfg, '1_2_1_iid_fix_a1-b1_bn_1_1',
+                   data_name='MNIST', model_name='conv')
+    model = make_model(cfg, model_rate=1)
+    fed = _fed(cfg, model)
+    m = fed.split_model([1])[0]  # rate 0.5
+    assert m['blocks.0.weight'].out.n == 32 and m['blocks.0.weight'].inp.n == 1
+    assert m['blocks.0.bias'].out.n == 32
+    # norm (BN) weight/bias follow
+    assert m['blocks.2.weight'].out.n == 32
+    # final linear: full out rows, sliced input
+    keys = [k for k in model.state_dict() if 'weight' in k]
+    ps = m[keys[-1]]
+    assert ps.out.kind == FULL and ps.inp.n == 256
+
+
+# -------------------------------------------------------------- transformer
+def test_transformer_split_rules(base_cfg):
+    cfg = make_cfg(base_cfg, '1_2_1_iid_fix_a1-b1_bn_1_1',
+                   data_name='WikiText2', model_name='transformer',
+                   num_tokens=100)
+    model = make_model(cfg, model_rate=1)
+    fed = _fed(cfg, model)
+    m = fed.split_model([1])[0]  # rate 0.5
+    # embedding: full vocab rows, embedding dim sliced to 128
+    ps = m['transformer_embedding.embedding.weight']
+    assert ps.out.kind == FULL and ps.out.n == 101
+    assert ps.inp.n == 128
+    # positional embedding also slices the embedding dim
+    ps = m['transformer_embedding.positional_embedding.positional_embedding.weight']
+    assert ps.out.kind == FULL and ps.inp.n == 128
+    # q/k/v: per-head slicing — head_dim 32 -> 16 per head, 8 heads
+    ps = m['transformer_encoder.layers.0.mha.linear_q.weight']
+    assert ps.out.kind == GATHER
+    expect = torch.arange(256).reshape(8, 32)[:, :16].reshape(-1)
+    assert torch.equal(ps.out.idx, expect)
+    assert ps.inp.n == 128
+    # after q bias the running index resets to the layer input (fed.py:147-151):
+    # linear_k input must be the embedding prefix, not q's per-head gather
+    ps = m['transformer_encoder.layers.0.mha.linear_k.weight']
+    assert ps.inp.kind == PREFIX and ps.inp.n == 128
+    # linear_o input = v's per-head out index
+    ps = m['transformer_encoder.layers.0.mha.linear_o.weight']
+    assert ps.inp.kind == GATHER and ps.inp.n == 128
+    # decoder.linear2 keeps full vocab rows
+    ps = m['decoder.linear2.weight']
+    assert ps.out.kind == FULL and ps.out.n == 100
+    assert m['decoder.linear2.bias'].out.kind == FULL
+
+
+# ---------------------------------------------------------------- transport
+@pytest.mark.parametrize('model_name,data_name', [
+    ('conv', 'MNIST'), ('resnet18', 'CIFAR10'), ('transformer', 'WikiText2')])
+def test_distribute_shapes_match_local_model(base_cfg, model_name, data_name):
+    """Each distributed slice must load into a model built at that rate."""
+    cfg = make_cfg(base_cfg, '1_4_1_iid_fix_a1-b1-c1-e1_bn_1_1',
+                   data_name=data_name, model_name=model_name, num_tokens=100)
+    global_model = make_model(cfg, model_rate=1)
+    fed = _fed(cfg, global_model)
+    user_idx = [0, 1, 2, 3]
+    local_parameters, _ = fed.distribute(user_idx)
+    for m, u in enumerate(user_idx):
+        local = make_model(cfg, model_rate=fed.model_rate[u])
+        local.load_state_dict(local_parameters[m])  # raises on shape mismatch
+
+
+def test_rate1_round_trip(base_cfg):
+    """distribute -> combine with a single rate-1 client is an exact
+    round-trip (SURVEY §4 property 2)."""
+    cfg = make_cfg(base_cfg, '1_1_1_iid_fix_a1_bn_1_1',
+                   data_name='CIFAR10', model_name='resnet18')
+    model = make_model(cfg, model_rate=1)
+    before = {k: v.clone() for k, v in model.state_dict().items()}
+    fed = _fed(cfg, model, rates=[1.0])
+    local, pidx = fed.distribute([0])
+    fed.combine(local, pidx, [0])
+    after = fed.global_parameters
+    for k in before:
+        assert torch.allclose(before[k].float(), after[k].float(), atol=1e-6), k
+
+
+def test_combine_count_weighted_average(base_cfg):
+    """Two clients at different rates: overlapping entries average, the
+    full-width client owns the tail alone."""
+    cfg = make_cfg(base_cfg, '1_2_1_iid_fix_a1-b1_bn_1_1',
+                   data_name='CIFAR10', model_name='resnet18')
+    model = make_model(cfg, model_rate=1)
+    fed = _fed(cfg, model, rates=[1.0, 0.5])
+    user_idx = [0, 1]
+    local, pidx = fed.distribute(user_idx)
+    # client 0 (full) contributes 1s, client 1 (half) contributes 3s
+    for m, fill in [(0, 1.0), (1, 3.0)]:
+        for k in local[m]:
+            if local[m][k].is_floating_point():
+                local[m][k] = torch.full_like(local[m][k], fill)
+    fed.combine(local, pidx, user_idx)
+    w = fed.global_parameters['layer1.0.conv1.weight']  # (64,64,3,3)
+    assert torch.allclose(w[:32, :32], torch.full_like(w[:32, :32], 2.0))
+    assert torch.allclose(w[32:, :], torch.full_like(w[32:, :], 1.0))
+    assert torch.allclose(w[:32, 32:], torch.full_like(w[:32, 32:], 1.0))
+
+
+def test_combine_label_split_filtering(base_cfg):
+    """Output-layer rows are aggregated only over clients holding that label
+    (reference: src/fed.py:193-198)."""
+    cfg = make_cfg(base_cfg, '1_2_1_iid_fix_a1-a1_bn_1_1',
+                   data_name='CIFAR10', model_name='resnet18')
+    model = make_model(cfg, model_rate=1)
+    label_split = {0: [0, 1], 1: [1, 2]}
+    fed = _fed(cfg, model, rates=[1.0, 1.0], label_split=label_split)
+    user_idx = [0, 1]
+    local, pidx = fed.distribute(user_idx)
+    before = fed.global_parameters['linear.weight'].clone()
+    for m, fill in [(0, 1.0), (1, 3.0)]:
+        for k in local[m]:
+            if local[m][k].is_floating_point():
+                local[m][k] = torch.full_like(local[m][k], fill)
+    fed.combine(local, pidx, user_idx)
+    w = fed.global_parameters['linear.weight']
+    assert torch.allclose(w[0], torch.full_like(w[0], 1.0))   # only client 0
+    assert torch.allclose(w[1], torch.full_like(w[1], 2.0))   # both -> mean
+    assert torch.allclose(w[2], torch.full_like(w[2], 3.0))   # only client 1
+    # rows 3..9: nobody contributed -> unchanged
+    assert torch.equal(w[3:], before[3:])
+
+
+def test_dynamic_resampling_with_generator(base_cfg):
+    cfg = make_cfg(base_cfg, '1_20_0.5_iid_dynamic_a1-e1_bn_1_1',
+                   data_name='CIFAR10', model_name='resnet18')
+    model = make_model(cfg, model_rate=1)
+    fed = _fed(cfg, model, rates=cfg['model_rate'])
+    g1 = torch.Generator().manual_seed(7)
+    fed.make_model_rate(generator=g1)
+    r1 = list(fed.model_rate)
+    g2 = torch.Generator().manual_seed(7)
+    fed.make_model_rate(generator=g2)
+    assert r1 == fed.model_rate
+    assert set(fed.model_rate) <= {1, 0.0625}
