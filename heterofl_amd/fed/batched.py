@@ -1,0 +1,479 @@
+"""Batched-client training engine — the MI355X-native hot path.
+
+The reference trains the round's active clients strictly one at a time with
+batch 10 on 32x32 images (reference: src/train_classifier_fed.py:106-121) —
+on a 256-CU MI355X that leaves >95% of the chip idle and is launch-latency
+bound.  Here all active clients of one width rate are trained in ONE model:
+
+- every client becomes a GROUP of a grouped convolution: R clients' conv
+  weights stack to (R*Cout, Cin, kh, kw), the input batch stacks channelwise
+  to (N, R*Cin, H, W), and one conv kernel launch computes all R clients;
+- static BatchNorm over R*C channels IS per-client sBN (batch stats are
+  per-channel, channels never mix across groups);
+- the classifier becomes a per-client bmm; cross-entropy is computed per
+  client and summed (gradients stay exactly per-client because groups are
+  disjoint);
+- grad clipping is the per-client global-L2 clip at 1.0, vectorized over
+  the client dimension;
+- the batched state_dict keys MIRROR the local model's keys with stacked
+  shapes, so pack/unpack between Federation slices and the batched model is
+  mechanical.
+
+Numerics: identical to the sequential engine up to fp reduction order
+(tested in tests/test_batched.py).
+"""
+import math
+
+import numpy as np
+import torch
+import torch.nn as nn
+import torch.nn.functional as F
+
+
+# --------------------------------------------------------------------- ops
+class BConv2d(nn.Module):
+    """R clients' Conv2d as one grouped conv."""
+
+    def __init__(self, R, in_ch, out_ch, kernel, stride, padding, bias):
+        super().__init__()
+        self.R, self.in_ch, self.out_ch = R, in_ch, out_ch
+        self.stride, self.padding = stride, padding
+        self.weight = nn.Parameter(torch.empty(R * out_ch, in_ch, kernel, kernel))
+        self.bias = nn.Parameter(torch.empty(R * out_ch)) if bias else None
+
+    def forward(self, x):
+        return F.conv2d(x, self.weight, self.bias, stride=self.stride,
+                        padding=self.padding, groups=self.R)
+
+
+class BBatchNorm2d(nn.Module):
+    """Per-client static BatchNorm (momentum=None, no running stats)."""
+
+    def __init__(self, R, ch):
+        super().__init__()
+        self.weight = nn.Parameter(torch.ones(R * ch))
+        self.bias = nn.Parameter(torch.zeros(R * ch))
+
+    def forward(self, x):
+        return F.batch_norm(x, None, None, self.weight, self.bias,
+                            training=True, eps=1e-5)
+
+
+class BGroupNorm(nn.Module):
+    """Per-client GroupNorm: local GroupNorm(G, C) -> batched (R*G, R*C)."""
+
+    def __init__(self, R, groups, ch):
+        super().__init__()
+        self.num_groups = R * groups
+        self.weight = nn.Parameter(torch.ones(R * ch))
+        self.bias = nn.Parameter(torch.zeros(R * ch))
+
+    def forward(self, x):
+        return F.group_norm(x, self.num_groups, self.weight, self.bias, eps=1e-5)
+
+
+class BLinear(nn.Module):
+    """R clients' Linear: weight (R, out, in), input (N, R, in)."""
+
+    def __init__(self, R, in_f, out_f):
+        super().__init__()
+        self.weight = nn.Parameter(torch.empty(R, out_f, in_f))
+        self.bias = nn.Parameter(torch.zeros(R, out_f))
+
+    def forward(self, x):
+        # x: (N, R, in) -> (N, R, out)
+        return torch.einsum('nri,roi->nro', x, self.weight) + self.bias
+
+
+def batched_norm(norm, R, ch):
+    if norm == 'bn':
+        return BBatchNorm2d(R, ch)
+    if norm == 'in':
+        return BGroupNorm(R, ch, ch)
+    if norm == 'ln':
+        return BGroupNorm(R, 1, ch)
+    if norm == 'gn':
+        return BGroupNorm(R, 4, ch)
+    if norm == 'none':
+        return nn.Identity()
+    raise ValueError('Not valid norm')
+
+
+class BScaler(nn.Module):
+    def __init__(self, rate):
+        super().__init__()
+        self.rate = rate
+
+    def forward(self, x):
+        return x / self.rate if self.training else x
+
+
+# ------------------------------------------------------------------ models
+class BBlock(nn.Module):
+    expansion = 1
+
+    def __init__(self, R, in_planes, planes, stride, rate, norm, scale):
+        super().__init__()
+        self.n1 = batched_norm(norm, R, in_planes)
+        self.conv1 = BConv2d(R, in_planes, planes, 3, stride, 1, bias=False)
+        self.n2 = batched_norm(norm, R, planes)
+        self.conv2 = BConv2d(R, planes, planes, 3, 1, 1, bias=False)
+        self.scaler = BScaler(rate) if scale else nn.Identity()
+        if stride != 1 or in_planes != self.expansion * planes:
+            self.shortcut = BConv2d(R, in_planes, self.expansion * planes,
+                                    1, stride, 0, bias=False)
+
+    def forward(self, x):
+        out = F.relu(self.n1(self.scaler(x)))
+        shortcut = self.shortcut(out) if hasattr(self, 'shortcut') else x
+        out = self.conv1(out)
+        out = self.conv2(F.relu(self.n2(self.scaler(out))))
+        out += shortcut
+        return out
+
+
+class BatchedResNet(nn.Module):
+    """R same-rate clients' ResNet18/34 in one grouped model.  state_dict
+    keys mirror models.resnet.ResNet with stacked shapes."""
+
+    def __init__(self, R, data_shape, hidden_size, num_blocks, num_classes,
+                 rate, norm, scale):
+        super().__init__()
+        self.R = R
+        self.num_classes = num_classes
+        self.data_ch = data_shape[0]
+        self.in_planes = hidden_size[0]
+        self.conv1 = BConv2d(R, data_shape[0], hidden_size[0], 3, 1, 1, bias=False)
+        self.layer1 = self._make_layer(R, hidden_size[0], num_blocks[0], 1, rate, norm, scale)
+        self.layer2 = self._make_layer(R, hidden_size[1], num_blocks[1], 2, rate, norm, scale)
+        self.layer3 = self._make_layer(R, hidden_size[2], num_blocks[2], 2, rate, norm, scale)
+        self.layer4 = self._make_layer(R, hidden_size[3], num_blocks[3], 2, rate, norm, scale)
+        self.n4 = batched_norm(norm, R, hidden_size[3])
+        self.scaler = BScaler(rate) if scale else nn.Identity()
+        self.linear = BLinear(R, hidden_size[3], num_classes)
+
+    def _make_layer(self, R, planes, num_blocks, stride, rate, norm, scale):
+        strides = [stride] + [1] * (num_blocks - 1)
+        layers = []
+        for s in strides:
+            layers.append(BBlock(R, self.in_planes, planes, s, rate, norm, scale))
+            self.in_planes = planes
+        return nn.Sequential(*layers)
+
+    def forward(self, x):
+        # x: (N, R*data_ch, H, W) -> scores (N, R, classes)
+        out = self.conv1(x)
+        out = self.layer1(out)
+        out = self.layer2(out)
+        out = self.layer3(out)
+        out = self.layer4(out)
+        out = F.relu(self.n4(self.scaler(out)))
+        out = F.adaptive_avg_pool2d(out, 1)
+        out = out.view(out.size(0), self.R, -1)
+        return self.linear(out)
+
+
+class BatchedConv(nn.Module):
+    """R same-rate clients' 4-block CNN.  Keys mirror models.conv.Conv
+    (blocks.i.*) with stacked shapes."""
+
+    def __init__(self, R, data_shape, hidden_size, num_classes, rate, norm, scale):
+        super().__init__()
+        self.R = R
+        self.num_classes = num_classes
+        blocks = []
+        in_ch = data_shape[0]
+        for i, out_ch in enumerate(hidden_size):
+            blocks.append(BConv2d(R, in_ch, out_ch, 3, 1, 1, bias=True))
+            blocks.append(BScaler(rate) if scale else nn.Identity())
+            blocks.append(batched_norm(norm, R, out_ch))
+            blocks.append(nn.ReLU(inplace=True))
+            if i != len(hidden_size) - 1:
+                blocks.append(nn.MaxPool2d(2))
+            in_ch = out_ch
+        self.blocks = nn.Sequential(*blocks)
+        # final linear occupies the tail indices like the local model
+        self._feat = hidden_size[-1]
+        self.head = BLinear(R, hidden_size[-1], num_classes)
+
+    def forward(self, x):
+        out = self.blocks(x)
+        out = F.adaptive_avg_pool2d(out, 1).view(out.size(0), self.R, -1)
+        return self.head(out)
+
+
+# ------------------------------------------------------- pack / unpack
+def pack_states(batched, locals_list):
+    """Load R per-client state dicts into the batched model, by key."""
+    R = batched.R
+    bsd = batched.state_dict()
+    key_map = _key_map(batched, locals_list[0])
+    with torch.no_grad():
+        for bk, lk in key_map.items():
+            bt = bsd[bk]
+            vals = [locals_list[r][lk] for r in range(R)]
+            if bt.dim() == 3:            # BLinear weight (R, O, I)
+                src = torch.stack(vals, 0)
+            elif bt.dim() == 2:          # BLinear bias (R, O)
+                src = torch.stack(vals, 0)
+            else:                        # stacked along channel dim 0
+                src = torch.cat(vals, 0)
+            bt.copy_(src.to(bt.device, bt.dtype))
+
+
+def unpack_states(batched, template_keys):
+    """Split the batched model back into R per-client state dicts."""
+    R = batched.R
+    bsd = batched.state_dict()
+    key_map = _key_map_from_keys(batched, template_keys)
+    outs = [dict() for _ in range(R)]
+    for bk, lk in key_map.items():
+        bt = bsd[bk].detach()
+        # BLinear params are (R, ...) stacks; everything else is a dim-0 cat
+        # (conv weights (R*O, I, kh, kw), norm weights/biases (R*C,)).
+        if bt.dim() in (2, 3):
+            for r in range(R):
+                outs[r][lk] = bt[r].clone()
+        else:
+            per = bt.size(0) // R
+            for r in range(R):
+                outs[r][lk] = bt[r * per:(r + 1) * per].clone()
+    return outs
+
+
+def _key_map(batched, local_sd):
+    return _key_map_from_keys(batched, list(local_sd.keys()))
+
+
+def _key_map_from_keys(batched, local_keys):
+    """Map batched state_dict keys -> local model keys.  BatchedResNet keys
+    match exactly; BatchedConv maps head.* to the local blocks tail linear."""
+    bkeys = list(batched.state_dict().keys())
+    if isinstance(batched, BatchedConv):
+        lin_w = [k for k in local_keys if 'weight' in k][-1]
+        lin_b = [k for k in local_keys if 'bias' in k][-1]
+        m = {}
+        for bk in bkeys:
+            if bk == 'head.weight':
+                m[bk] = lin_w
+            elif bk == 'head.bias':
+                m[bk] = lin_b
+            else:
+                m[bk] = bk
+        return m
+    return {bk: bk for bk in bkeys}
+
+
+# ------------------------------------------------------------- loss / clip
+def batched_masked_ce(scores, labels, label_masks):
+    """scores (N, R, C), labels (N, R), label_masks (R, C) in {0,1} or None.
+    Returns per-client mean-CE losses (R,).  Matches the reference's
+    mask-then-CE order (src/models/resnet.py:152-157)."""
+    N, R, C = scores.shape
+    if label_masks is not None:
+        scores = scores.masked_fill(label_masks.unsqueeze(0) == 0, 0)
+    logp = F.log_softmax(scores, dim=2)
+    nll = -logp.gather(2, labels.unsqueeze(2)).squeeze(2)  # (N, R)
+    return nll.mean(0)
+
+
+def per_client_clip_(params, R, max_norm=1.0):
+    """Vectorized per-client global-L2 grad clip at max_norm
+    (reference: src/train_classifier_fed.py:205)."""
+    sq = None
+    views = []
+    for p in params:
+        if p.grad is None:
+            continue
+        g = p.grad
+        if g.size(0) % R != 0:
+            raise RuntimeError(f'grad shape {tuple(g.shape)} not divisible by R={R}')
+        v = g.view(R, -1)
+        views.append((g, v))
+        s = (v.float() ** 2).sum(dim=1)
+        sq = s if sq is None else sq + s
+    if sq is None:
+        return None
+    total = sq.sqrt()
+    scale = (max_norm / (total + 1e-6)).clamp(max=1.0)
+    for g, v in views:
+        v.mul_(scale.unsqueeze(1).to(v.dtype))
+    return total
+
+
+# ------------------------------------------------------------ augmentation
+class DeviceAugment:
+    """GPU-side per-sample augmentation + normalization matching the
+    reference's torchvision pipeline distributionally
+    (reference: src/data.py:14-27)."""
+
+    def __init__(self, data_name, device):
+        self.data_name = data_name
+        if data_name in ('MNIST', 'FashionMNIST'):
+            mean, std = (0.1307,), (0.3081,)
+            self.crop = False
+        else:
+            mean, std = (0.4914, 0.4822, 0.4465), (0.2023, 0.1994, 0.2010)
+            self.crop = True
+        self.mean = torch.tensor(mean, device=device).view(1, -1, 1, 1)
+        self.std = torch.tensor(std, device=device).view(1, -1, 1, 1)
+
+    def __call__(self, img_u8, train=True):
+        """img_u8: (n, H, W, C) uint8 -> (n, C, H, W) float normalized."""
+        x = img_u8.permute(0, 3, 1, 2).float().div_(255.0)
+        n, C, H, W = x.shape
+        if train and self.crop:
+            pad = 4
+            xp = F.pad(x, [pad] * 4)
+            dy = torch.randint(0, 2 * pad + 1, (n,), device=x.device)
+            dx = torch.randint(0, 2 * pad + 1, (n,), device=x.device)
+            rows = dy.view(n, 1) + torch.arange(H, device=x.device).view(1, H)
+            idx_r = rows.view(n, 1, H, 1).expand(n, C, H, W + 2 * pad)
+            xp = xp.gather(2, idx_r)
+            cols = dx.view(n, 1) + torch.arange(W, device=x.device).view(1, W)
+            idx_c = cols.view(n, 1, 1, W).expand(n, C, H, W)
+            x = xp.gather(3, idx_c)
+            flip = torch.rand(n, device=x.device) < 0.5
+            x = torch.where(flip.view(n, 1, 1, 1), x.flip(-1), x)
+        return (x - self.mean) / self.std
+
+
+# ----------------------------------------------------------------- trainer
+class BatchedClientTrainer:
+    """Trains the round's active clients grouped by (rate, batch schedule).
+
+    Interface matches SequentialClientTrainer.train_clients; data access goes
+    through shard tensors staged on device (set_data must be called by the
+    runner before the first round)."""
+
+    def __init__(self, cfg):
+        self.cfg = cfg
+        self.device = torch.device(cfg['device'])
+        self.dataset = None
+        self.data_split = None
+        self.augment = None
+        # bf16 local training (master weights fp32, compute bf16)
+        self._amp = (cfg.get('compute_dtype') == 'bfloat16'
+                     and self.device.type == 'cuda')
+        self._shard_cache = {}
+        self._model_cache = {}
+
+    # data staging ---------------------------------------------------------
+    def set_data(self, dataset, data_split):
+        self.dataset = dataset['train']
+        self.data_split = data_split['train']
+        self.augment = DeviceAugment(self.cfg['data_name'], self.device)
+        self._shard_cache.clear()
+
+    def _shard(self, user):
+        if user not in self._shard_cache:
+            idx = torch.tensor(self.data_split[user], dtype=torch.long)
+            img = self.dataset.img[idx]          # uint8 (n, H, W[, C])
+            if img.dim() == 3:
+                img = img.unsqueeze(-1)
+            labels = torch.tensor([self.dataset.target[i] for i in idx.tolist()],
+                                  dtype=torch.long)
+            self._shard_cache[user] = (img.to(self.device, non_blocking=True),
+                                       labels.to(self.device, non_blocking=True))
+        return self._shard_cache[user]
+
+    # model construction ---------------------------------------------------
+    def _batched_model(self, rate, R):
+        key = (rate, R)
+        if key not in self._model_cache:
+            cfg = self.cfg
+            name = cfg['model_name']
+            scaler_rate = rate / cfg['global_model_rate']
+            if 'resnet' in name:
+                hidden = [int(np.ceil(rate * h)) for h in cfg['resnet']['hidden_size']]
+                nb = {'resnet18': [2, 2, 2, 2], 'resnet34': [3, 4, 6, 3]}[name]
+                model = BatchedResNet(R, cfg['data_shape'], hidden, nb,
+                                      cfg['classes_size'], scaler_rate,
+                                      cfg['norm'], cfg['scale'])
+            elif name == 'conv':
+                hidden = [int(np.ceil(rate * h)) for h in cfg['conv']['hidden_size']]
+                model = BatchedConv(R, cfg['data_shape'], hidden,
+                                    cfg['classes_size'], scaler_rate,
+                                    cfg['norm'], cfg['scale'])
+            else:
+                raise NotImplementedError(name)
+            self._model_cache[key] = model.to(self.device)
+        return self._model_cache[key]
+
+    # training -------------------------------------------------------------
+    def train_clients(self, client_slots, user_idx, local_parameters,
+                      model_rate, make_loader, label_split, lr, logger=None):
+        if self.dataset is None:
+            raise RuntimeError('BatchedClientTrainer.set_data was not called')
+        cfg = self.cfg
+        B = cfg['batch_size']['train']
+        # group slots by (rate, batch schedule)
+        groups = {}
+        for m in client_slots:
+            u = user_idx[m]
+            n = len(self.data_split[u])
+            sched = tuple([B] * (n // B) + ([n % B] if n % B else []))
+            groups.setdefault((model_rate[u], sched), []).append(m)
+        out = []
+        for (rate, sched), slots in groups.items():
+            out.extend(self._train_group(rate, sched, slots, user_idx,
+                                         local_parameters, label_split, lr,
+                                         logger))
+        return out
+
+    def _train_group(self, rate, sched, slots, user_idx, local_parameters,
+                     label_split, lr, logger=None):
+        cfg = self.cfg
+        R = len(slots)
+        device = self.device
+        model = self._batched_model(rate, R)
+        locals_list = [local_parameters[m] for m in slots]
+        pack_states(model, locals_list)
+        model.train(True)
+        params = [p for p in model.parameters() if p.requires_grad]
+        opt = torch.optim.SGD(params, lr=lr, momentum=cfg['momentum'],
+                              weight_decay=cfg['weight_decay'])
+        # label masks (R, classes)
+        masks = None
+        if cfg['mask']:
+            masks = torch.zeros(R, cfg['classes_size'], device=device)
+            for i, m in enumerate(slots):
+                masks[i, label_split[user_idx[m]]] = 1
+        shards = [self._shard(user_idx[m]) for m in slots]
+        n = sum(sched)
+        steps = len(sched)
+        local_metrics = []
+        for _ in range(cfg['num_epochs']['local']):
+            # per-client shuffle + epoch-wide augmented gather
+            xs, ys = [], []
+            for img, lab in shards:
+                perm = torch.randperm(img.size(0), device=device)
+                xs.append(self.augment(img[perm], train=True))   # (n, C, H, W)
+                ys.append(lab[perm])
+            x_all = torch.stack(xs, 1)   # (n, R, C, H, W)
+            x_all = x_all.reshape(n, -1, x_all.size(-2), x_all.size(-1))
+            y_all = torch.stack(ys, 1)   # (n, R)
+            off = 0
+            for bs in sched:
+                xb = x_all[off:off + bs]
+                yb = y_all[off:off + bs]
+                off += bs
+                opt.zero_grad(set_to_none=True)
+                with torch.autocast('cuda', torch.bfloat16, enabled=self._amp):
+                    scores = model(xb)
+                    losses = batched_masked_ce(scores.float(), yb, masks)
+                losses.sum().backward()
+                per_client_clip_(params, R, 1.0)
+                opt.step()
+                if logger is not None:
+                    with torch.no_grad():
+                        pred = scores.argmax(dim=2)
+                        acc = (pred == yb).float().mean(0) * 100.0
+                        for i in range(R):
+                            logger.append({'Local-Loss': losses[i].item(),
+                                           'Local-Accuracy': acc[i].item()},
+                                          'train', n=bs)
+        template_keys = list(locals_list[0].keys())
+        states = unpack_states(model, template_keys)
+        cpu_or_dev = [{k: v for k, v in st.items()} for st in states]
+        return list(zip(slots, cpu_or_dev))

@@ -47,9 +47,12 @@ class FedRunner:
         self.federation = Federation(model.state_dict(), cfg['model_rate'],
                                      label_split, cfg)
         self.is_lm = cfg['model_name'] == 'transformer'
-        if cfg.get('engine', 'sequential') == 'batched':
+        use_batched = (cfg.get('engine', 'sequential') == 'batched'
+                       and not self.is_lm)
+        if use_batched:
             from .batched import BatchedClientTrainer
             self.trainer = BatchedClientTrainer(cfg)
+            self.trainer.set_data(dataset, data_split)
         else:
             self.trainer = SequentialClientTrainer(cfg)
         self._round_gen = None

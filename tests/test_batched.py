@@ -1,0 +1,162 @@
+"""Batched-client engine equivalence vs the sequential oracle.
+
+The grouped-model trainer must reproduce per-client training exactly
+(up to fp32 reduction order): groups are disjoint, so forward, backward,
+per-client clip and SGD decompose per client.
+"""
+import numpy as np
+import torch
+import torch.nn.functional as F
+
+from heterofl_amd.fed.batched import (BatchedResNet, BatchedConv, pack_states,
+                                      unpack_states, batched_masked_ce,
+                                      per_client_clip_, BatchedClientTrainer)
+from heterofl_amd.models import make_model
+from heterofl_amd.models.functional import masked_cross_entropy
+from tests.conftest import make_cfg
+
+R = 3
+CLASSES = 10
+
+
+def _local_models(cfg, rate, n=R):
+    models = []
+    for i in range(n):
+        torch.manual_seed(100 + i)
+        models.append(make_model(cfg, model_rate=rate))
+    return models
+
+
+def _batched_resnet(cfg, rate, n=R):
+    hidden = [int(np.ceil(rate * h)) for h in cfg['resnet']['hidden_size']]
+    return BatchedResNet(n, cfg['data_shape'], hidden, [2, 2, 2, 2],
+                         CLASSES, rate / cfg['global_model_rate'],
+                         cfg['norm'], cfg['scale'])
+
+
+def test_pack_unpack_roundtrip(base_cfg):
+    cfg = make_cfg(base_cfg, '1_3_1_iid_fix_b1_bn_1_1',
+                   data_name='CIFAR10', model_name='resnet18')
+    cfg['global_model_rate'] = 1.0
+    locals_ = _local_models(cfg, 0.5)
+    bm = _batched_resnet(cfg, 0.5)
+    sds = [m.state_dict() for m in locals_]
+    pack_states(bm, sds)
+    outs = unpack_states(bm, list(sds[0].keys()))
+    for r in range(R):
+        for k in sds[r]:
+            assert torch.equal(outs[r][k], sds[r][k]), (r, k)
+
+
+def _train_sequential(model, data, labels, mask, lr, steps):
+    model.train(True)
+    opt = torch.optim.SGD(model.parameters(), lr=lr, momentum=0.9,
+                          weight_decay=5e-4)
+    for s in range(steps):
+        opt.zero_grad()
+        out = model.features(data[s])
+        score = model.linear(out)
+        score, loss = masked_cross_entropy(score, labels[s], mask, CLASSES)
+        loss.backward()
+        torch.nn.utils.clip_grad_norm_(model.parameters(), 1)
+        opt.step()
+    return model.state_dict()
+
+
+def test_batched_step_equivalence(base_cfg):
+    """K SGD steps on the batched model == the same steps on each client's
+    own model, same data, to fp32 tolerance."""
+    cfg = make_cfg(base_cfg, '1_3_1_iid_fix_b1_bn_1_1',
+                   data_name='CIFAR10', model_name='resnet18')
+    cfg['global_model_rate'] = 1.0
+    rate, lr, steps, n = 0.5, 0.1, 3, 6
+    locals_ = _local_models(cfg, rate)
+    torch.manual_seed(0)
+    data = [torch.randn(n, R, 3, 32, 32) for _ in range(steps)]
+    labels = [torch.randint(0, CLASSES, (n, R)) for _ in range(steps)]
+    masks = torch.zeros(R, CLASSES)
+    for r in range(R):
+        masks[r, [r, r + 1, r + 2]] = 1
+    # sequential oracle
+    seq_states = []
+    for r in range(R):
+        d = [data[s][:, r] for s in range(steps)]
+        l = [labels[s][:, r] for s in range(steps)]
+        seq_states.append(_train_sequential(locals_[r], d, l,
+                                            torch.tensor([r, r + 1, r + 2]),
+                                            lr, steps))
+    # batched
+    locals2 = _local_models(cfg, rate)
+    bm = _batched_resnet(cfg, rate)
+    pack_states(bm, [m.state_dict() for m in locals2])
+    bm.train(True)
+    params = list(bm.parameters())
+    opt = torch.optim.SGD(params, lr=lr, momentum=0.9, weight_decay=5e-4)
+    for s in range(steps):
+        xb = data[s].permute(0, 1, 2, 3, 4).reshape(n, R * 3, 32, 32)
+        opt.zero_grad()
+        scores = bm(xb)
+        losses = batched_masked_ce(scores, labels[s], masks)
+        losses.sum().backward()
+        per_client_clip_(params, R, 1.0)
+        opt.step()
+    outs = unpack_states(bm, list(seq_states[0].keys()))
+    for r in range(R):
+        for k in seq_states[r]:
+            a, b = seq_states[r][k], outs[r][k]
+            diff = (a - b).abs().max().item()
+            scale = a.abs().max().item() + 1e-8
+            assert diff / max(scale, 1.0) < 2e-4, (r, k, diff, scale)
+
+
+def test_batched_masked_ce_matches_reference():
+    torch.manual_seed(1)
+    n = 8
+    scores = torch.randn(n, R, CLASSES)
+    labels = torch.randint(0, 3, (n, R))
+    masks = torch.zeros(R, CLASSES)
+    for r in range(R):
+        masks[r, :4] = 1
+    losses = batched_masked_ce(scores.clone(), labels, masks)
+    for r in range(R):
+        _, ref = masked_cross_entropy(scores[:, r].clone(), labels[:, r],
+                                      torch.arange(4), CLASSES)
+        assert torch.allclose(losses[r], ref, atol=1e-6), r
+
+
+def test_per_client_clip_matches_torch():
+    torch.manual_seed(2)
+    cfgR = 4
+    p1 = torch.nn.Parameter(torch.randn(cfgR * 8, 3, 3, 3))
+    p2 = torch.nn.Parameter(torch.randn(cfgR, 10, 8))
+    p1.grad = torch.randn_like(p1) * 3
+    p2.grad = torch.randn_like(p2) * 3
+    g1, g2 = p1.grad.clone(), p2.grad.clone()
+    per_client_clip_([p1, p2], cfgR, 1.0)
+    for r in range(cfgR):
+        gr1, gr2 = g1.view(cfgR, -1)[r], g2.view(cfgR, -1)[r]
+        total = (gr1.pow(2).sum() + gr2.pow(2).sum()).sqrt()
+        coef = min(1.0, 1.0 / (total.item() + 1e-6))
+        assert torch.allclose(p1.grad.view(cfgR, -1)[r], gr1 * coef, atol=1e-6)
+        assert torch.allclose(p2.grad.view(cfgR, -1)[r], gr2 * coef, atol=1e-6)
+
+
+def test_batched_engine_e2e(base_cfg):
+    """Full federated round with engine='batched' on CPU."""
+    from heterofl_amd.data import fetch_dataset, split_dataset
+    from heterofl_amd.fed import FedRunner
+    from heterofl_amd.utils import process_dataset, make_optimizer
+    cfg = make_cfg(base_cfg, '1_4_0.5_iid_fix_a1-e1_bn_1_1',
+                   data_name='CIFAR10', model_name='resnet18')
+    cfg['engine'] = 'batched'
+    cfg['num_epochs'] = {'global': 1, 'local': 1}
+    torch.manual_seed(0)
+    ds = fetch_dataset('CIFAR10', synthetic=True, synthetic_size=80)
+    process_dataset(ds, cfg)
+    data_split, label_split = split_dataset(ds, 4, 'iid', cfg['classes_size'])
+    model = make_model(cfg)
+    opt = make_optimizer(model, cfg['lr'], cfg)
+    runner = FedRunner(cfg, ds, data_split, label_split, model, opt)
+    runner.train_round(1)
+    tm = runner.stats()
+    runner.test(tm, 1)
