@@ -39,7 +39,12 @@ class BConv2d(nn.Module):
         self.R, self.in_ch, self.out_ch = R, in_ch, out_ch
         self.stride, self.padding = stride, padding
         self.weight = nn.Parameter(torch.empty(R * out_ch, in_ch, kernel, kernel))
-        self.bias = nn.Parameter(torch.empty(R * out_ch)) if bias else None
+        nn.init.kaiming_uniform_(self.weight, a=math.sqrt(5))
+        if bias:
+            bound = 1 / math.sqrt(in_ch * kernel * kernel)
+            self.bias = nn.Parameter(torch.empty(R * out_ch).uniform_(-bound, bound))
+        else:
+            self.bias = None
 
     def forward(self, x):
         return F.conv2d(x, self.weight, self.bias, stride=self.stride,
@@ -78,6 +83,7 @@ class BLinear(nn.Module):
     def __init__(self, R, in_f, out_f):
         super().__init__()
         self.weight = nn.Parameter(torch.empty(R, out_f, in_f))
+        nn.init.kaiming_uniform_(self.weight, a=math.sqrt(5))
         self.bias = nn.Parameter(torch.zeros(R, out_f))
 
     def forward(self, x):
@@ -357,6 +363,7 @@ class BatchedClientTrainer:
                      and self.device.type == 'cuda')
         self._shard_cache = {}
         self._model_cache = {}
+        self._graph_cache = {}
 
     # data staging ---------------------------------------------------------
     def set_data(self, dataset, data_split):
@@ -421,13 +428,36 @@ class BatchedClientTrainer:
                                          logger))
         return out
 
+    def _graph_step(self, rate, sched, R, lr):
+        """Get/create the captured training step for this group shape."""
+        from .graphs import GraphedGroupStep
+        key = (rate, R, tuple(sched), lr)
+        if key not in self._graph_cache:
+            cfg = self.cfg
+            model = self._batched_model(rate, R)
+            # graphs own their model instance: evict from the eager cache so
+            # shapes don't alias
+            self._model_cache.pop((rate, R), None)
+            model.train(True)
+            capacity = sum(sched) * cfg['num_epochs']['local']
+            self._graph_cache[key] = GraphedGroupStep(
+                model, R, sched[0], lr, cfg['momentum'], cfg['weight_decay'],
+                cfg['classes_size'], capacity, self._amp, self.device)
+        return self._graph_cache[key]
+
     def _train_group(self, rate, sched, slots, user_idx, local_parameters,
                      label_split, lr, logger=None):
         cfg = self.cfg
         R = len(slots)
         device = self.device
-        model = self._batched_model(rate, R)
+        use_graph = (device.type == 'cuda' and cfg.get('hip_graphs', True)
+                     and len(set(sched)) == 1)
         locals_list = [local_parameters[m] for m in slots]
+        if use_graph:
+            return self._train_group_graphed(rate, sched, slots, user_idx,
+                                             locals_list, label_split, lr,
+                                             logger)
+        model = self._batched_model(rate, R)
         pack_states(model, locals_list)
         model.train(True)
         params = [p for p in model.parameters() if p.requires_grad]
@@ -477,3 +507,43 @@ class BatchedClientTrainer:
         states = unpack_states(model, template_keys)
         cpu_or_dev = [{k: v for k, v in st.items()} for st in states]
         return list(zip(slots, cpu_or_dev))
+
+    def _train_group_graphed(self, rate, sched, slots, user_idx, locals_list,
+                             label_split, lr, logger=None):
+        """hipGraph path: pack -> stage all epochs' augmented data -> replay
+        the captured step n_steps times -> unpack + device-side metrics."""
+        cfg = self.cfg
+        R = len(slots)
+        device = self.device
+        gs = self._graph_step(rate, sched, R, lr)
+        pack_states(gs.model, locals_list)
+        masks = None
+        if cfg['mask']:
+            masks = torch.zeros(R, cfg['classes_size'], device=device)
+            for i, m in enumerate(slots):
+                masks[i, label_split[user_idx[m]]] = 1
+        gs.begin_round(masks)
+        shards = [self._shard(user_idx[m]) for m in slots]
+        n = sum(sched)
+        E = cfg['num_epochs']['local']
+        xs_ep, ys_ep = [], []
+        for _ in range(E):
+            xs, ys = [], []
+            for img, lab in shards:
+                perm = torch.randperm(img.size(0), device=device)
+                xs.append(self.augment(img[perm], train=True))
+                ys.append(lab[perm])
+            x_all = torch.stack(xs, 1)
+            xs_ep.append(x_all.reshape(n, -1, x_all.size(-2), x_all.size(-1)))
+            ys_ep.append(torch.stack(ys, 1))
+        gs.run_epochs(torch.cat(xs_ep, 0), torch.cat(ys_ep, 0), len(sched))
+        if logger is not None:
+            m = gs.metrics.detach().cpu()
+            for i in range(R):
+                cnt = max(int(m[i, 2].item()), 1)
+                logger.append({'Local-Loss': m[i, 0].item() / cnt,
+                               'Local-Accuracy': m[i, 1].item() / cnt * 100.0},
+                              'train', n=cnt)
+        template_keys = list(locals_list[0].keys())
+        states = unpack_states(gs.model, template_keys)
+        return list(zip(slots, [dict(st) for st in states]))

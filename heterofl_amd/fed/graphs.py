@@ -1,0 +1,142 @@
+"""hipGraph-captured local-training steps.
+
+HeteroFL's hot loop is thousands of tiny steps (batch 10 on 32x32 images,
+~400 kernels each) — on MI355X that is launch-latency bound, not FLOP bound
+(SURVEY §7 'tiny-work kernels').  We capture ONE training step of the
+batched group model in a hipGraph (torch.cuda.CUDAGraph == hipGraph on ROCm)
+with a DEVICE-side step counter: the graph gathers its own batch from the
+round's pre-staged epoch buffer via the counter, so an entire local-training
+run (5 epochs x 50 steps) is 250 back-to-back graph replays with no host
+work in between.
+
+Graphs are cached per (rate, R, batch, lr, n_steps_capacity); weights,
+momentum and the label masks live in stable buffers the graph reads, so a
+new round only repacks buffers and replays.
+"""
+import math
+
+import torch
+
+from .batched import batched_masked_ce
+
+
+class GraphedGroupStep:
+    """One captured training step for a (rate, R) client group."""
+
+    def __init__(self, model, R, batch, lr, momentum, weight_decay, classes,
+                 capacity, amp, device):
+        self.model = model
+        self.R = R
+        self.batch = batch
+        self.lr = lr
+        self.momentum = momentum
+        self.weight_decay = weight_decay
+        self.amp = amp
+        self.device = device
+        self.params = [p for p in model.parameters() if p.requires_grad]
+        self.bufs = [torch.zeros_like(p) for p in self.params]
+        self.capacity = capacity  # max samples in the epoch buffer
+        shape = self._input_shape()
+        self.x_all = torch.zeros(capacity, *shape, device=device)
+        self.y_all = torch.zeros(capacity, R, dtype=torch.long, device=device)
+        self.masks = torch.ones(R, classes, device=device)
+        self.counter = torch.zeros(1, dtype=torch.long, device=device)
+        # device metric accumulators: [loss_sum*bs, correct, samples] per client
+        self.metrics = torch.zeros(R, 3, device=device)
+        self.graph = None
+        self._arange = torch.arange(batch, device=device)
+
+    def _input_shape(self):
+        m = self.model
+        ch = getattr(m, 'data_ch', 3)
+        # infer from first conv weight
+        w = next(p for p in m.parameters() if p.dim() == 4)
+        in_ch = w.size(1)
+        hw = 32 if in_ch == 3 else 28
+        return (self.R * in_ch, hw, hw)
+
+    def _one_step(self):
+        idx = self.counter * self.batch + self._arange
+        xb = self.x_all.index_select(0, idx)
+        yb = self.y_all.index_select(0, idx)
+        with torch.autocast('cuda', torch.bfloat16, enabled=self.amp):
+            scores = self.model(xb)
+            losses = batched_masked_ce(scores.float(), yb, self.masks)
+        loss = losses.sum()
+        grads = torch.autograd.grad(loss, self.params)
+        # per-client global-L2 clip at 1.0
+        sq = None
+        views = [g.view(self.R, -1) for g in grads]
+        for v in views:
+            s = (v.float() ** 2).sum(dim=1)
+            sq = s if sq is None else sq + s
+        scale = (1.0 / (sq.sqrt() + 1e-6)).clamp(max=1.0)
+        for v in views:
+            v.mul_(scale.unsqueeze(1).to(v.dtype))
+        # momentum SGD (wd before momentum, dampening 0 — torch SGD semantics)
+        grads = list(grads)
+        torch._foreach_add_(grads, self.params, alpha=self.weight_decay)
+        torch._foreach_mul_(self.bufs, self.momentum)
+        torch._foreach_add_(self.bufs, grads)
+        torch._foreach_add_(self.params, self.bufs, alpha=-self.lr)
+        # metrics
+        with torch.no_grad():
+            pred = scores.argmax(dim=2)
+            correct = (pred == yb).float().sum(0)
+            self.metrics[:, 0] += losses.detach() * self.batch
+            self.metrics[:, 1] += correct
+            self.metrics[:, 2] += self.batch
+        self.counter += 1
+
+    def capture(self):
+        """Warmup + capture mutate params/bufs/metrics — snapshot and
+        restore so lazy capture inside a round is state-neutral."""
+        saved_p = [p.detach().clone() for p in self.params]
+        saved_b = [b.clone() for b in self.bufs]
+        saved_m = self.metrics.clone()
+        torch.cuda.synchronize()
+        s = torch.cuda.Stream()
+        s.wait_stream(torch.cuda.current_stream())
+        with torch.cuda.stream(s):
+            for _ in range(3):  # warmup allocations / autotune
+                self.counter.zero_()
+                self._one_step()
+        torch.cuda.current_stream().wait_stream(s)
+        self.counter.zero_()
+        self.graph = torch.cuda.CUDAGraph()
+        with torch.cuda.graph(self.graph):
+            self._one_step()
+        with torch.no_grad():
+            for p, sp in zip(self.params, saved_p):
+                p.copy_(sp)
+            for b, sb in zip(self.bufs, saved_b):
+                b.copy_(sb)
+            self.metrics.copy_(saved_m)
+            self.counter.zero_()
+        torch.cuda.synchronize()
+
+    # ------------------------------------------------------------------ API
+    def begin_round(self, masks):
+        """Load label masks, zero momentum and metrics.  (Weights are packed
+        directly into self.model's parameters — stable graph addresses.)"""
+        with torch.no_grad():
+            if masks is not None:
+                self.masks.copy_(masks)
+            else:
+                self.masks.fill_(1)
+            for b in self.bufs:
+                b.zero_()
+            self.metrics.zero_()
+
+    def run_epochs(self, x_epochs, y_epochs, steps_per_epoch):
+        """x_epochs: (total, R*C, H, W) pre-augmented, all epochs concat."""
+        total = x_epochs.size(0)
+        with torch.no_grad():
+            self.x_all[:total].copy_(x_epochs)
+            self.y_all[:total].copy_(y_epochs)
+            self.counter.zero_()
+        n_steps = total // self.batch
+        if self.graph is None:
+            self.capture()
+        for _ in range(n_steps):
+            self.graph.replay()

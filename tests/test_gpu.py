@@ -104,6 +104,51 @@ def test_bf16_batched_step(base_cfg):
 
 
 @needs_gpu
+def test_graph_path_matches_eager(base_cfg):
+    """The hipGraph-captured step must train identically to the eager
+    batched path (same device RNG seed -> same data order/augmentation)."""
+    from heterofl_amd.data import fetch_dataset, split_dataset
+    from heterofl_amd.fed.batched import BatchedClientTrainer
+    from heterofl_amd.models import make_model
+    from heterofl_amd.utils import process_dataset
+    cfg = make_cfg(base_cfg, '1_4_1_iid_fix_a1_bn_1_1',
+                   data_name='CIFAR10', model_name='resnet18')
+    cfg['device'] = 'cuda:0'
+    cfg['engine'] = 'batched'
+    cfg['compute_dtype'] = 'float32'
+    cfg['num_epochs'] = {'global': 1, 'local': 2}
+    torch.manual_seed(0)
+    ds = fetch_dataset('CIFAR10', synthetic=True, synthetic_size=160)
+    process_dataset(ds, cfg)
+    torch.manual_seed(1)
+    data_split, label_split = split_dataset(ds, 4, 'iid', 10)
+    model = make_model(cfg).to('cuda:0')
+    gp = model.state_dict()
+    user_idx = [0, 1, 2, 3]
+    locals_ = [{k: v.clone() for k, v in gp.items()} for _ in user_idx]
+    rates = [1.0] * 4
+
+    results = {}
+    for use_graph in (False, True):
+        cfg2 = dict(cfg)
+        cfg2['hip_graphs'] = use_graph
+        tr = BatchedClientTrainer(cfg2)
+        tr.set_data(ds, data_split)
+        torch.manual_seed(42)
+        torch.cuda.manual_seed_all(42)
+        out = tr.train_clients(list(range(4)), user_idx,
+                               [dict(l) for l in locals_], rates,
+                               None, label_split, 0.1)
+        results[use_graph] = dict(out)
+    for m in range(4):
+        for k in results[False][m]:
+            a = results[False][m][k].float().cpu()
+            b = results[True][m][k].float().cpu()
+            diff = (a - b).abs().max().item()
+            assert diff < 1e-4, (m, k, diff)
+
+
+@needs_gpu
 def test_fed_round_gpu(base_cfg):
     """One full federated round on GPU with the batched engine."""
     from heterofl_amd.data import fetch_dataset, split_dataset
