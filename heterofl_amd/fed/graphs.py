@@ -63,7 +63,7 @@ class GraphedGroupStep:
             scores = self.model(xb)
             losses = batched_masked_ce(scores.float(), yb, self.masks)
         loss = losses.sum()
-        grads = torch.autograd.grad(loss, self.params)
+        grads = [g.contiguous() for g in torch.autograd.grad(loss, self.params)]
         # per-client global-L2 clip at 1.0
         sq = None
         views = [g.view(self.R, -1) for g in grads]
