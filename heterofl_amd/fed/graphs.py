@@ -63,30 +63,30 @@ class GraphedGroupStep:
             scores = self.model(xb)
             losses = batched_masked_ce(scores.float(), yb, self.masks)
         loss = losses.sum()
-        grads = [g.contiguous() for g in torch.autograd.grad(loss, self.params)]
-        # per-client global-L2 clip at 1.0
-        sq = None
-        views = [g.view(self.R, -1) for g in grads]
-        for v in views:
-            s = (v.float() ** 2).sum(dim=1)
-            sq = s if sq is None else sq + s
-        scale = (1.0 / (sq.sqrt() + 1e-6)).clamp(max=1.0)
-        for v in views:
-            v.mul_(scale.unsqueeze(1).to(v.dtype))
-        # momentum SGD (wd before momentum, dampening 0 — torch SGD semantics)
-        grads = list(grads)
-        torch._foreach_add_(grads, self.params, alpha=self.weight_decay)
-        torch._foreach_mul_(self.bufs, self.momentum)
-        torch._foreach_add_(self.bufs, grads)
-        torch._foreach_add_(self.params, self.bufs, alpha=-self.lr)
-        # metrics
+        raw = torch.autograd.grad(loss, self.params)
         with torch.no_grad():
+            grads = [g.contiguous() for g in raw]
+            # per-client global-L2 clip at 1.0
+            sq = None
+            views = [g.view(self.R, -1) for g in grads]
+            for v in views:
+                s = (v.float() ** 2).sum(dim=1)
+                sq = s if sq is None else sq + s
+            scale = (1.0 / (sq.sqrt() + 1e-6)).clamp(max=1.0)
+            for v in views:
+                v.mul_(scale.unsqueeze(1).to(v.dtype))
+            # momentum SGD (wd before momentum, dampening 0 — torch semantics)
+            torch._foreach_add_(grads, self.params, alpha=self.weight_decay)
+            torch._foreach_mul_(self.bufs, self.momentum)
+            torch._foreach_add_(self.bufs, grads)
+            torch._foreach_add_(self.params, self.bufs, alpha=-self.lr)
+            # metrics
             pred = scores.argmax(dim=2)
             correct = (pred == yb).float().sum(0)
             self.metrics[:, 0] += losses.detach() * self.batch
             self.metrics[:, 1] += correct
             self.metrics[:, 2] += self.batch
-        self.counter += 1
+            self.counter += 1
 
     def capture(self):
         """Warmup + capture mutate params/bufs/metrics — snapshot and
