@@ -113,6 +113,11 @@ def main():
         torch.distributed.all_reduce(t, op=torch.distributed.ReduceOp.MAX)
         elapsed = t.item()
 
+    if os.environ.get('HETEROFL_TIMING') == '1' and rank == 0:
+        from heterofl_amd.fed.runner import _phase_timer
+        import sys
+        print('[timing]', json.dumps(_phase_timer.report()), file=sys.stderr)
+
     active = int(torch.tensor(cfg['frac'] * cfg['num_users']).ceil())
     samples_per_user = len(ds['train']) // cfg['num_users']
     samples_per_round = active * cfg['num_epochs']['local'] * samples_per_user

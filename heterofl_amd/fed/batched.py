@@ -512,11 +512,14 @@ class BatchedClientTrainer:
                              label_split, lr, logger=None):
         """hipGraph path: pack -> stage all epochs' augmented data -> replay
         the captured step n_steps times -> unpack + device-side metrics."""
+        from .runner import _phase_timer
         cfg = self.cfg
         R = len(slots)
         device = self.device
-        gs = self._graph_step(rate, sched, R, lr)
-        pack_states(gs.model, locals_list)
+        with _phase_timer('2a.graph_build'):
+            gs = self._graph_step(rate, sched, R, lr)
+        with _phase_timer('2b.pack'):
+            pack_states(gs.model, locals_list)
         masks = None
         if cfg['mask']:
             masks = torch.zeros(R, cfg['classes_size'], device=device)
@@ -526,17 +529,21 @@ class BatchedClientTrainer:
         shards = [self._shard(user_idx[m]) for m in slots]
         n = sum(sched)
         E = cfg['num_epochs']['local']
-        xs_ep, ys_ep = [], []
-        for _ in range(E):
-            xs, ys = [], []
-            for img, lab in shards:
-                perm = torch.randperm(img.size(0), device=device)
-                xs.append(self.augment(img[perm], train=True))
-                ys.append(lab[perm])
-            x_all = torch.stack(xs, 1)
-            xs_ep.append(x_all.reshape(n, -1, x_all.size(-2), x_all.size(-1)))
-            ys_ep.append(torch.stack(ys, 1))
-        gs.run_epochs(torch.cat(xs_ep, 0), torch.cat(ys_ep, 0), len(sched))
+        with _phase_timer('2c.stage_augment'):
+            xs_ep, ys_ep = [], []
+            for _ in range(E):
+                xs, ys = [], []
+                for img, lab in shards:
+                    perm = torch.randperm(img.size(0), device=device)
+                    xs.append(self.augment(img[perm], train=True))
+                    ys.append(lab[perm])
+                x_all = torch.stack(xs, 1)
+                xs_ep.append(x_all.reshape(n, -1, x_all.size(-2), x_all.size(-1)))
+                ys_ep.append(torch.stack(ys, 1))
+            x_cat = torch.cat(xs_ep, 0)
+            y_cat = torch.cat(ys_ep, 0)
+        with _phase_timer('2d.replay'):
+            gs.run_epochs(x_cat, y_cat, len(sched))
         if logger is not None:
             m = gs.metrics.detach().cpu()
             for i in range(R):
@@ -544,6 +551,7 @@ class BatchedClientTrainer:
                 logger.append({'Local-Loss': m[i, 0].item() / cnt,
                                'Local-Accuracy': m[i, 1].item() / cnt * 100.0},
                               'train', n=cnt)
-        template_keys = list(locals_list[0].keys())
-        states = unpack_states(gs.model, template_keys)
+        with _phase_timer('2e.unpack'):
+            template_keys = list(locals_list[0].keys())
+            states = unpack_states(gs.model, template_keys)
         return list(zip(slots, [dict(st) for st in states]))

@@ -12,9 +12,38 @@ rates from a round-seeded generator, trains its shard, and combine runs as
 a padded all-reduce (see parallel/dist.py).
 """
 import math
+import os
+import time
 
 import numpy as np
 import torch
+
+_TIMING = os.environ.get('HETEROFL_TIMING') == '1'
+
+
+class _phase_timer:
+    acc = {}
+
+    def __init__(self, name):
+        self.name = name
+
+    def __enter__(self):
+        if _TIMING:
+            if torch.cuda.is_available():
+                torch.cuda.synchronize()
+            self.t0 = time.perf_counter()
+        return self
+
+    def __exit__(self, *a):
+        if _TIMING:
+            if torch.cuda.is_available():
+                torch.cuda.synchronize()
+            dt = time.perf_counter() - self.t0
+            _phase_timer.acc[self.name] = _phase_timer.acc.get(self.name, 0.0) + dt
+
+    @classmethod
+    def report(cls):
+        return {k: round(v, 4) for k, v in sorted(cls.acc.items())}
 
 from ..data import SplitDataset, BatchDataset, make_data_loader
 from ..metrics import Metric
@@ -93,25 +122,29 @@ class FedRunner:
         else:
             user_idx = sample_active_users(cfg, epoch)
             resample = True
-        local_parameters, param_idx = self.federation.distribute(
-            user_idx, resample=resample)
+        with _phase_timer('1.distribute'):
+            local_parameters, param_idx = self.federation.distribute(
+                user_idx, resample=resample)
 
         # shard clients across ranks (identity shard when single rank)
         rank, world = (0, 1) if self.dist_ctx is None else \
             (self.dist_ctx.rank, self.dist_ctx.world_size)
         my_clients = list(range(rank, len(user_idx), world))
 
-        trained = dict(self.trainer.train_clients(
-            my_clients, user_idx, local_parameters, self.federation.model_rate,
-            self._make_loader, self.label_split, lr, self.logger))
+        with _phase_timer('2.local_train'):
+            trained = dict(self.trainer.train_clients(
+                my_clients, user_idx, local_parameters,
+                self.federation.model_rate, self._make_loader,
+                self.label_split, lr, self.logger))
 
-        if self.dist_ctx is None:
-            ordered = [trained[m] for m in range(len(user_idx))]
-            self.federation.combine(ordered, param_idx, user_idx)
-        else:
-            from ..parallel.dist import distributed_combine
-            distributed_combine(self.federation, trained, param_idx, user_idx,
-                                self.dist_ctx)
+        with _phase_timer('3.combine'):
+            if self.dist_ctx is None:
+                ordered = [trained[m] for m in range(len(user_idx))]
+                self.federation.combine(ordered, param_idx, user_idx)
+            else:
+                from ..parallel.dist import distributed_combine
+                distributed_combine(self.federation, trained, param_idx,
+                                    user_idx, self.dist_ctx)
         self.global_model.load_state_dict(self.federation.global_parameters)
         return user_idx
 
