@@ -79,7 +79,11 @@ def main():
     cfg['world_size'] = world_size
 
     torch.manual_seed(0)
-    ds = fetch_dataset('CIFAR10', synthetic=True)
+    # real CIFAR10 shape: 500 samples per user (50k/100); HETEROFL_BENCH_SPU
+    # shrinks it for plumbing tests only
+    spu = int(os.environ.get('HETEROFL_BENCH_SPU', '500'))
+    ds = fetch_dataset('CIFAR10', synthetic=True,
+                       synthetic_size=spu * cfg['num_users'])
     process_dataset(ds, cfg)
     # deterministic split across ranks
     g = torch.Generator().manual_seed(1234)
