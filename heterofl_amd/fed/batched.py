@@ -29,6 +29,8 @@ import torch
 import torch.nn as nn
 import torch.nn.functional as F
 
+from .. import ops as native_ops
+
 
 # --------------------------------------------------------------------- ops
 class BConv2d(nn.Module):
@@ -105,6 +107,47 @@ def batched_norm(norm, R, ch):
     raise ValueError('Not valid norm')
 
 
+class BNormReLU(nn.Module):
+    """Fused Scaler -> per-client norm -> ReLU (the prefix of every block,
+    reference: src/models/resnet.py:44-50).  On GPU this is ONE hand-written
+    HIP kernel forward and one backward (ops/csrc/fused_norm.hip); on CPU the
+    explicit scaler/norm/relu torch ops run (the oracle).  The Scaler's x/rate
+    cancels inside a train-mode norm, so the kernel omits it; with norm='none'
+    the explicit ops always run."""
+
+    def __init__(self, R, ch, norm, rate, scale):
+        super().__init__()
+        self.norm = norm
+        self.rate = rate if scale else 1.0
+        if norm == 'in':
+            self.groups = R * ch
+        elif norm == 'ln':
+            self.groups = R
+        elif norm == 'gn':
+            self.groups = R * 4
+        else:
+            self.groups = 0
+        if norm != 'none':
+            self.weight = nn.Parameter(torch.ones(R * ch))
+            self.bias = nn.Parameter(torch.zeros(R * ch))
+
+    def forward(self, x):
+        if self.norm == 'none':
+            if self.training and self.rate != 1.0:
+                x = x / self.rate
+            return F.relu(x)
+        if native_ops.use_native(x):
+            from ..ops.fused import fused_norm_relu
+            kind = 'bn' if self.norm == 'bn' else 'gn'
+            return fused_norm_relu(x, self.weight, self.bias, kind,
+                                   self.groups)
+        from ..ops.fused import eager_scaler_norm_relu
+        kind = 'bn' if self.norm == 'bn' else 'gn'
+        r = self.rate if self.training else 1.0
+        return eager_scaler_norm_relu(x, self.weight, self.bias, kind,
+                                      self.groups, r)
+
+
 class BScaler(nn.Module):
     def __init__(self, rate):
         super().__init__()
@@ -120,20 +163,19 @@ class BBlock(nn.Module):
 
     def __init__(self, R, in_planes, planes, stride, rate, norm, scale):
         super().__init__()
-        self.n1 = batched_norm(norm, R, in_planes)
+        self.n1 = BNormReLU(R, in_planes, norm, rate, scale)
         self.conv1 = BConv2d(R, in_planes, planes, 3, stride, 1, bias=False)
-        self.n2 = batched_norm(norm, R, planes)
+        self.n2 = BNormReLU(R, planes, norm, rate, scale)
         self.conv2 = BConv2d(R, planes, planes, 3, 1, 1, bias=False)
-        self.scaler = BScaler(rate) if scale else nn.Identity()
         if stride != 1 or in_planes != self.expansion * planes:
             self.shortcut = BConv2d(R, in_planes, self.expansion * planes,
                                     1, stride, 0, bias=False)
 
     def forward(self, x):
-        out = F.relu(self.n1(self.scaler(x)))
+        out = self.n1(x)
         shortcut = self.shortcut(out) if hasattr(self, 'shortcut') else x
         out = self.conv1(out)
-        out = self.conv2(F.relu(self.n2(self.scaler(out))))
+        out = self.conv2(self.n2(out))
         out += shortcut
         return out
 
@@ -154,8 +196,7 @@ class BatchedResNet(nn.Module):
         self.layer2 = self._make_layer(R, hidden_size[1], num_blocks[1], 2, rate, norm, scale)
         self.layer3 = self._make_layer(R, hidden_size[2], num_blocks[2], 2, rate, norm, scale)
         self.layer4 = self._make_layer(R, hidden_size[3], num_blocks[3], 2, rate, norm, scale)
-        self.n4 = batched_norm(norm, R, hidden_size[3])
-        self.scaler = BScaler(rate) if scale else nn.Identity()
+        self.n4 = BNormReLU(R, hidden_size[3], norm, rate, scale)
         self.linear = BLinear(R, hidden_size[3], num_classes)
 
     def _make_layer(self, R, planes, num_blocks, stride, rate, norm, scale):
@@ -173,7 +214,7 @@ class BatchedResNet(nn.Module):
         out = self.layer2(out)
         out = self.layer3(out)
         out = self.layer4(out)
-        out = F.relu(self.n4(self.scaler(out)))
+        out = self.n4(out)
         out = F.adaptive_avg_pool2d(out, 1)
         out = out.view(out.size(0), self.R, -1)
         return self.linear(out)
@@ -191,9 +232,11 @@ class BatchedConv(nn.Module):
         in_ch = data_shape[0]
         for i, out_ch in enumerate(hidden_size):
             blocks.append(BConv2d(R, in_ch, out_ch, 3, 1, 1, bias=True))
-            blocks.append(BScaler(rate) if scale else nn.Identity())
-            blocks.append(batched_norm(norm, R, out_ch))
-            blocks.append(nn.ReLU(inplace=True))
+            # placeholder keeps the norm at sequential index 4k+2 so keys
+            # mirror the local model (src/models/conv.py:29-33)
+            blocks.append(nn.Identity())
+            blocks.append(BNormReLU(R, out_ch, norm, rate, scale))
+            blocks.append(nn.Identity())
             if i != len(hidden_size) - 1:
                 blocks.append(nn.MaxPool2d(2))
             in_ch = out_ch
@@ -271,16 +314,28 @@ def _key_map_from_keys(batched, local_keys):
 
 
 # ------------------------------------------------------------- loss / clip
-def batched_masked_ce(scores, labels, label_masks):
+def batched_masked_ce(scores, labels, label_masks, metrics=None):
     """scores (N, R, C), labels (N, R), label_masks (R, C) in {0,1} or None.
     Returns per-client mean-CE losses (R,).  Matches the reference's
-    mask-then-CE order (src/models/resnet.py:152-157)."""
+    mask-then-CE order (src/models/resnet.py:152-157).  On GPU this is the
+    hand-written kernel (ops/csrc/masked_ce.hip), which also accumulates
+    device-side (loss-sum, correct, count) into `metrics` (R, 3) if given."""
+    if native_ops.use_native(scores):
+        from ..ops.fused import fused_masked_ce
+        return fused_masked_ce(scores, labels, label_masks, metrics)
     N, R, C = scores.shape
     if label_masks is not None:
         scores = scores.masked_fill(label_masks.unsqueeze(0) == 0, 0)
     logp = F.log_softmax(scores, dim=2)
     nll = -logp.gather(2, labels.unsqueeze(2)).squeeze(2)  # (N, R)
-    return nll.mean(0)
+    losses = nll.mean(0)
+    if metrics is not None:
+        with torch.no_grad():
+            correct = (scores.argmax(2) == labels).float().sum(0)
+            metrics[:, 0] += losses.detach() * N
+            metrics[:, 1] += correct
+            metrics[:, 2] += N
+    return losses
 
 
 def per_client_clip_(params, R, max_norm=1.0):

@@ -9,6 +9,12 @@ round's pre-staged epoch buffer via the counter, so an entire local-training
 run (5 epochs x 50 steps) is 250 back-to-back graph replays with no host
 work in between.
 
+With the native extension the step is: fused norm+relu kernels in the model,
+fused masked-CE (which also accumulates device-side metrics), backward into
+preallocated .grad buffers, then the two-kernel fused per-client
+clip(1.0)+momentum-SGD (ops/csrc/clip_sgd.hip) over a chunk table built once
+at capture.  Without it (CPU/debug) the pure-torch step runs.
+
 Graphs are cached per (rate, R, batch, lr, n_steps_capacity); weights,
 momentum and the label masks live in stable buffers the graph reads, so a
 new round only repacks buffers and replays.
@@ -17,7 +23,8 @@ import math
 
 import torch
 
-from .batched import batched_masked_ce
+from .. import ops as native_ops
+from .batched import batched_masked_ce, per_client_clip_
 
 
 class GraphedGroupStep:
@@ -41,15 +48,22 @@ class GraphedGroupStep:
         self.y_all = torch.zeros(capacity, R, dtype=torch.long, device=device)
         self.masks = torch.ones(R, classes, device=device)
         self.counter = torch.zeros(1, dtype=torch.long, device=device)
-        # device metric accumulators: [loss_sum*bs, correct, samples] per client
+        # device metric accumulators: [loss_sum, correct, samples] per client
         self.metrics = torch.zeros(R, 3, device=device)
         self.graph = None
         self._arange = torch.arange(batch, device=device)
+        self._native = (device.type == 'cuda' and native_ops.use_native(device))
+        if self._native:
+            native_ops.require_native()
+            for p in self.params:
+                p.grad = torch.zeros_like(p)
+            from ..ops.fused import FusedClipSGD
+            self.opt = FusedClipSGD(self.params,
+                                    [p.grad for p in self.params],
+                                    self.bufs, R, device)
 
     def _input_shape(self):
         m = self.model
-        ch = getattr(m, 'data_ch', 3)
-        # infer from first conv weight
         w = next(p for p in m.parameters() if p.dim() == 4)
         in_ch = w.size(1)
         hw = 32 if in_ch == 3 else 28
@@ -59,6 +73,18 @@ class GraphedGroupStep:
         idx = self.counter * self.batch + self._arange
         xb = self.x_all.index_select(0, idx)
         yb = self.y_all.index_select(0, idx)
+        if self._native:
+            with torch.no_grad():
+                torch._foreach_zero_([p.grad for p in self.params])
+            with torch.autocast('cuda', torch.bfloat16, enabled=self.amp):
+                scores = self.model(xb)
+            losses = batched_masked_ce(scores.float(), yb, self.masks,
+                                       metrics=self.metrics)
+            losses.sum().backward()
+            self.opt.step(1.0, self.lr, self.momentum, self.weight_decay)
+            with torch.no_grad():
+                self.counter += 1
+            return
         with torch.autocast('cuda', torch.bfloat16, enabled=self.amp):
             scores = self.model(xb)
             losses = batched_masked_ce(scores.float(), yb, self.masks)

@@ -1,0 +1,37 @@
+// Python bindings for the gfx950 HIP kernels (_heterofl_hip).
+#include <torch/extension.h>
+
+std::vector<at::Tensor> bn_relu_fwd(at::Tensor x, at::Tensor gamma,
+                                    at::Tensor beta, double eps);
+std::vector<at::Tensor> bn_relu_bwd(at::Tensor dy, at::Tensor x,
+                                    at::Tensor gamma, at::Tensor beta,
+                                    at::Tensor mean, at::Tensor invstd);
+std::vector<at::Tensor> gn_relu_fwd(at::Tensor x, at::Tensor gamma,
+                                    at::Tensor beta, int64_t G, double eps);
+std::vector<at::Tensor> gn_relu_bwd(at::Tensor dy, at::Tensor x,
+                                    at::Tensor gamma, at::Tensor beta,
+                                    at::Tensor mean, at::Tensor invstd,
+                                    int64_t G);
+std::vector<at::Tensor> masked_ce_fwd(at::Tensor scores, at::Tensor labels,
+                                      at::Tensor mask, at::Tensor metrics);
+at::Tensor masked_ce_bwd(at::Tensor scores, at::Tensor labels, at::Tensor mask,
+                         at::Tensor up);
+std::vector<at::Tensor> build_chunk_table(std::vector<at::Tensor> grads,
+                                          std::vector<at::Tensor> params,
+                                          std::vector<at::Tensor> bufs,
+                                          int64_t R, int64_t chunk_elems);
+void clip_sgd_step(at::Tensor table_blob, int64_t n_chunks, at::Tensor normsq,
+                   double max_norm, double lr, double momentum,
+                   double weight_decay);
+
+PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+    m.def("bn_relu_fwd", &bn_relu_fwd, "fused sBN+ReLU forward");
+    m.def("bn_relu_bwd", &bn_relu_bwd, "fused sBN+ReLU backward");
+    m.def("gn_relu_fwd", &gn_relu_fwd, "fused GroupNorm+ReLU forward");
+    m.def("gn_relu_bwd", &gn_relu_bwd, "fused GroupNorm+ReLU backward");
+    m.def("masked_ce_fwd", &masked_ce_fwd, "batched masked CE forward");
+    m.def("masked_ce_bwd", &masked_ce_bwd, "batched masked CE backward");
+    m.def("build_chunk_table", &build_chunk_table,
+          "build clip+SGD chunk table");
+    m.def("clip_sgd_step", &clip_sgd_step, "fused per-client clip+SGD step");
+}

@@ -1,0 +1,114 @@
+"""Autograd wrappers for the hand-written gfx950 kernels, with pure-torch
+reference implementations (the CPU path and the numerics oracle in tests).
+
+Semantics mirrored from the reference:
+- Scaler -> norm -> ReLU prefix of every block (src/models/resnet.py:44-50,
+  src/models/conv.py:29-33).  With a train-mode norm the Scaler division
+  cancels exactly (norm(x/r) == norm(x)), so the fused kernels omit it; the
+  eager path keeps the explicit division as the oracle.
+- masked cross-entropy (src/models/resnet.py:152-157).
+- per-client clip(1.0) + momentum-SGD (src/train_classifier_fed.py:205-206).
+"""
+import torch
+import torch.nn.functional as F
+
+from . import require_native, use_native
+
+
+class _FusedNormReLU(torch.autograd.Function):
+    """y = relu(norm(x) * w + b) with batch (kind='bn') or group stats."""
+
+    @staticmethod
+    def forward(ctx, x, weight, bias, kind, groups, eps):
+        ext = require_native()
+        x = x.contiguous()
+        if kind == 'bn':
+            y, mean, invstd = ext.bn_relu_fwd(x, weight, bias, eps)
+        else:
+            y, mean, invstd = ext.gn_relu_fwd(x, weight, bias, groups, eps)
+        ctx.save_for_backward(x, weight, bias, mean, invstd)
+        ctx.kind = kind
+        ctx.groups = groups
+        return y
+
+    @staticmethod
+    def backward(ctx, dy):
+        ext = require_native()
+        x, weight, bias, mean, invstd = ctx.saved_tensors
+        if ctx.kind == 'bn':
+            dx, dgamma, dbeta = ext.bn_relu_bwd(dy, x, weight, bias, mean,
+                                                invstd)
+        else:
+            dx, dgamma, dbeta = ext.gn_relu_bwd(dy, x, weight, bias, mean,
+                                                invstd, ctx.groups)
+        return dx, dgamma, dbeta, None, None, None
+
+
+def fused_norm_relu(x, weight, bias, kind, groups=1, eps=1e-5):
+    return _FusedNormReLU.apply(x, weight, bias, kind, groups, eps)
+
+
+def eager_scaler_norm_relu(x, weight, bias, kind, groups, rate, eps=1e-5):
+    """Reference path: scaler -> norm -> relu with explicit torch ops."""
+    if rate != 1.0:
+        x = x / rate
+    if kind == 'bn':
+        x = F.batch_norm(x, None, None, weight, bias, training=True, eps=eps)
+    else:
+        x = F.group_norm(x, groups, weight, bias, eps=eps)
+    return F.relu(x)
+
+
+class _FusedMaskedCE(torch.autograd.Function):
+    """Per-client mean masked-CE over scores (N, R, C); optionally
+    accumulates (sum-loss, correct, count) into a device metrics buffer."""
+
+    @staticmethod
+    def forward(ctx, scores, labels, mask, metrics):
+        ext = require_native()
+        scores = scores.contiguous().float()
+        m = mask if mask is not None else torch.Tensor()
+        (losses,) = ext.masked_ce_fwd(
+            scores, labels.contiguous(),
+            m if mask is not None else torch.Tensor(),
+            metrics if metrics is not None else torch.Tensor())
+        ctx.save_for_backward(scores, labels,
+                              mask if mask is not None else torch.Tensor())
+        ctx.has_mask = mask is not None
+        return losses
+
+    @staticmethod
+    def backward(ctx, up):
+        ext = require_native()
+        scores, labels, mask = ctx.saved_tensors
+        dscores = ext.masked_ce_bwd(scores, labels,
+                                    mask if ctx.has_mask else torch.Tensor(),
+                                    up)
+        return dscores, None, None, None
+
+
+def fused_masked_ce(scores, labels, mask=None, metrics=None):
+    return _FusedMaskedCE.apply(scores, labels, mask, metrics)
+
+
+class FusedClipSGD:
+    """Chunk-table driven per-client clip + momentum-SGD over the batched
+    model's parameters.  Build once per captured graph (pointer stability is
+    guaranteed by holding refs to grads/params/bufs)."""
+
+    CHUNK = 65536
+
+    def __init__(self, params, grads, bufs, R, device):
+        ext = require_native()
+        self._refs = (list(params), list(grads), list(bufs))
+        blob, n = ext.build_chunk_table(self._refs[1], self._refs[0],
+                                        self._refs[2], R, self.CHUNK)
+        self.table = blob
+        self.n_chunks = int(n.item())
+        self.normsq = torch.zeros(R, dtype=torch.float32, device=device)
+
+    def step(self, max_norm, lr, momentum, weight_decay):
+        ext = require_native()
+        self.normsq.zero_()
+        ext.clip_sgd_step(self.table, self.n_chunks, self.normsq, max_norm,
+                          lr, momentum, weight_decay)

@@ -172,3 +172,115 @@ def test_fed_round_gpu(base_cfg):
     for v in runner.federation.global_parameters.values():
         if v.is_floating_point():
             assert torch.isfinite(v).all()
+
+
+# ---------------------------------------------------------- native kernels
+@needs_gpu
+def test_fused_bn_relu_matches_torch(base_cfg):
+    """HIP fused sBN+ReLU fwd/bwd vs plain torch fp32 reference."""
+    from heterofl_amd.ops.fused import fused_norm_relu
+    torch.manual_seed(0)
+    for N, C, HW in [(10, 64, 32 * 32), (10, 2560, 16), (3, 20, 49)]:
+        x = torch.randn(N, C, int(HW ** 0.5) if int(HW ** 0.5) ** 2 == HW else 1,
+                        HW // (int(HW ** 0.5) if int(HW ** 0.5) ** 2 == HW else 1),
+                        device='cuda', requires_grad=True)
+        w = torch.rand(C, device='cuda', requires_grad=True) + 0.5
+        b = torch.randn(C, device='cuda', requires_grad=True)
+        y = fused_norm_relu(x, w, b, 'bn', 0)
+        x2 = x.detach().clone().requires_grad_(True)
+        w2 = w.detach().clone().requires_grad_(True)
+        b2 = b.detach().clone().requires_grad_(True)
+        import torch.nn.functional as F
+        ref = F.relu(F.batch_norm(x2, None, None, w2, b2, training=True,
+                                  eps=1e-5))
+        assert (y - ref).abs().max().item() < 1e-4
+        g = torch.randn_like(y)
+        y.backward(g)
+        ref.backward(g)
+        assert (x.grad - x2.grad).abs().max().item() < 1e-4, (N, C, HW)
+        assert (w.grad - w2.grad).abs().max().item() < 2e-3
+        assert (b.grad - b2.grad).abs().max().item() < 2e-3
+
+
+@needs_gpu
+def test_fused_gn_relu_matches_torch(base_cfg):
+    from heterofl_amd.ops.fused import fused_norm_relu
+    import torch.nn.functional as F
+    torch.manual_seed(0)
+    for N, C, H, G in [(10, 40, 8, 20), (10, 64, 16, 4), (4, 16, 4, 16)]:
+        x = torch.randn(N, C, H, H, device='cuda', requires_grad=True)
+        w = torch.rand(C, device='cuda', requires_grad=True) + 0.5
+        b = torch.randn(C, device='cuda', requires_grad=True)
+        y = fused_norm_relu(x, w, b, 'gn', G)
+        x2 = x.detach().clone().requires_grad_(True)
+        w2 = w.detach().clone().requires_grad_(True)
+        b2 = b.detach().clone().requires_grad_(True)
+        ref = F.relu(F.group_norm(x2, G, w2, b2, eps=1e-5))
+        assert (y - ref).abs().max().item() < 1e-4
+        g = torch.randn_like(y)
+        y.backward(g)
+        ref.backward(g)
+        assert (x.grad - x2.grad).abs().max().item() < 1e-4, (N, C, G)
+        assert (w.grad - w2.grad).abs().max().item() < 2e-3
+        assert (b.grad - b2.grad).abs().max().item() < 2e-3
+
+
+@needs_gpu
+def test_fused_masked_ce_matches_torch(base_cfg):
+    from heterofl_amd.ops.fused import fused_masked_ce
+    import torch.nn.functional as F
+    torch.manual_seed(0)
+    N, R, C = 10, 5, 10
+    scores = torch.randn(N, R, C, device='cuda', requires_grad=True)
+    labels = torch.randint(0, C, (N, R), device='cuda')
+    mask = (torch.rand(R, C, device='cuda') > 0.3).float()
+    for r in range(R):  # labels must be unmasked (reference invariant)
+        mask[r, labels[:, r]] = 1
+    metrics = torch.zeros(R, 3, device='cuda')
+    losses = fused_masked_ce(scores, labels, mask, metrics)
+    s2 = scores.detach().clone().requires_grad_(True)
+    ref_masked = s2.masked_fill(mask.unsqueeze(0) == 0, 0)
+    logp = F.log_softmax(ref_masked, dim=2)
+    ref = -logp.gather(2, labels.unsqueeze(2)).squeeze(2).mean(0)
+    assert (losses - ref).abs().max().item() < 1e-5
+    losses.sum().backward()
+    ref.sum().backward()
+    assert (scores.grad - s2.grad).abs().max().item() < 1e-5
+    # metrics: loss sums and counts
+    assert (metrics[:, 0] - ref.detach() * N).abs().max().item() < 1e-4
+    assert (metrics[:, 2] == N).all()
+    correct = (ref_masked.argmax(2) == labels).float().sum(0)
+    assert (metrics[:, 1] - correct).abs().max().item() < 1e-4
+
+
+@needs_gpu
+def test_fused_clip_sgd_matches_torch(base_cfg):
+    from heterofl_amd.ops.fused import FusedClipSGD
+    torch.manual_seed(0)
+    R, lr, mom, wd = 3, 0.1, 0.9, 5e-4
+    shapes = [(R * 8, 4, 3, 3), (R * 8,), (R, 10, 8), (R, 10)]
+    params = [torch.randn(*s, device='cuda') for s in shapes]
+    grads = [torch.randn(*s, device='cuda') for s in shapes]
+    bufs = [torch.zeros_like(p) for p in params]
+    # torch reference: per-client clip + SGD
+    ref_p = [p.clone() for p in params]
+    ref_b = [b.clone() for b in bufs]
+    ref_g = [g.clone() for g in grads]
+    sq = None
+    for g in ref_g:
+        s = (g.view(R, -1) ** 2).sum(1)
+        sq = s if sq is None else sq + s
+    scale = (1.0 / (sq.sqrt() + 1e-6)).clamp(max=1.0)
+    for i, g in enumerate(ref_g):
+        g.view(R, -1).mul_(scale.unsqueeze(1))
+        g.add_(ref_p[i], alpha=wd)
+        ref_b[i].mul_(mom).add_(g)
+        ref_p[i].add_(ref_b[i], alpha=-lr)
+    opt = FusedClipSGD(params, grads, bufs, R, torch.device('cuda'))
+    for step in range(2):
+        opt.step(1.0, lr, mom, wd)
+        if step == 0:
+            for i in range(len(params)):
+                assert (params[i] - ref_p[i]).abs().max().item() < 1e-5, i
+                assert (bufs[i] - ref_b[i]).abs().max().item() < 1e-5, i
+    torch.cuda.synchronize()
