@@ -127,6 +127,117 @@ def _run_one(cfg, seed, pivot_sign):
     logger.safe(False)
 
 
+def run_centralized_experiment(cfg, pivot_metric, pivot_sign, metric_name):
+    """Non-federated baseline: train ONE model at the global rate on the full
+    dataset (reference: src/train_classifier.py:48-133).  Where the reference
+    uses torch.nn.DataParallel for multi-GPU (src/train_classifier.py:65-66),
+    the MI355X-native path is one process per GPU with DDP over RCCL (run
+    under torchrun); single-process runs need no wrapper."""
+    from .data import make_data_loader
+    from .metrics import Metric
+    from .utils import collate, to_device
+    cfg['pivot_metric'] = pivot_metric
+    cfg['metric_name'] = metric_name
+    process_control(cfg)
+    seeds = list(range(cfg['init_seed'], cfg['init_seed'] + cfg['num_experiments']))
+    for seed in seeds:
+        cfg['model_tag'] = model_tag_of(seed, cfg)
+        cfg['pivot'] = -float('inf') * pivot_sign if pivot_sign > 0 else float('inf')
+        print('Experiment: {}'.format(cfg['model_tag']))
+        torch.manual_seed(seed)
+        if cfg['device'].startswith('cuda') and not torch.cuda.is_available():
+            cfg['device'] = 'cpu'
+        dataset = fetch_dataset(cfg['data_name'], cfg['subset'],
+                                synthetic=cfg.get('synthetic', False))
+        process_dataset(dataset, cfg)
+        model = make_model(cfg, model_rate=cfg['global_model_rate']).to(cfg['device'])
+        optimizer = make_optimizer(model, cfg['lr'], cfg)
+        scheduler = make_scheduler(optimizer, cfg)
+        ddp_model = model
+        if int(os.environ.get('WORLD_SIZE', '1')) > 1:
+            from .parallel import init_distributed
+            ctx = init_distributed()
+            ddp_model = torch.nn.parallel.DistributedDataParallel(
+                model, device_ids=[ctx.local_rank] if ctx.device.type == 'cuda'
+                else None)
+        if cfg['resume_mode'] == 1:
+            last_epoch, _, _, model, optimizer, scheduler, logger = resume(
+                model, cfg['model_tag'], optimizer, scheduler)
+        else:
+            last_epoch = 1
+            logger = Logger(os.path.join('output', 'runs',
+                                         'train_{}'.format(cfg['model_tag'])))
+        metric = Metric()
+        is_lm = cfg['model_name'] == 'transformer'
+        num_epochs = cfg['num_epochs']
+        if isinstance(num_epochs, dict):
+            num_epochs = num_epochs.get('global', 200)
+        for epoch in range(last_epoch, num_epochs + 1):
+            logger.safe(True)
+            ddp_model.train(True)
+            if is_lm:
+                from .data import BatchDataset
+                train_iter = BatchDataset(dataset['train'], cfg['bptt'])
+                batches = (train_iter[i] for i in range(len(train_iter)))
+            else:
+                loader = make_data_loader({'train': dataset['train']}, cfg)['train']
+                batches = (collate(b) for b in loader)
+            for input in batches:
+                input = to_device(input, cfg['device'])
+                optimizer.zero_grad()
+                output = ddp_model(input)
+                output['loss'].backward()
+                torch.nn.utils.clip_grad_norm_(model.parameters(), 1)
+                optimizer.step()
+                n = input['label'].size(0)
+                ev = metric.evaluate(cfg['metric_name']['train'], input, output)
+                logger.append(ev, 'train', n=n)
+            # sBN stats pass + evaluation (vision; reference
+            # src/train_classifier.py:123-133)
+            with torch.no_grad():
+                if not is_lm:
+                    test_model = make_model(cfg, model_rate=cfg['global_model_rate'],
+                                            track=True).to(cfg['device'])
+                    test_model.load_state_dict(model.state_dict(), strict=False)
+                    test_model.train(True)
+                    loader = make_data_loader({'train': dataset['train']}, cfg)['train']
+                    for input in loader:
+                        input = collate(input)
+                        test_model(to_device(input, cfg['device']))
+                else:
+                    test_model = model
+                test_model.train(False)
+                if is_lm:
+                    from .data import BatchDataset
+                    ds = BatchDataset(dataset['test'], cfg['bptt'])
+                    test_batches = (ds[i] for i in range(len(ds)))
+                else:
+                    loader = make_data_loader({'test': dataset['test']}, cfg)['test']
+                    test_batches = (collate(b) for b in loader)
+                for input in test_batches:
+                    input = to_device(input, cfg['device'])
+                    output = test_model(input)
+                    ev = metric.evaluate(cfg['metric_name']['test'], input, output)
+                    logger.append(ev, 'test', n=input['label'].size(0))
+            logger.write('test', cfg['metric_name']['test'])
+            scheduler.step()
+            logger.safe(False)
+            save_result = {
+                'cfg': cfg, 'epoch': epoch + 1, 'data_split': None,
+                'label_split': None, 'model_dict': model.state_dict(),
+                'optimizer_dict': optimizer.state_dict(),
+                'scheduler_dict': scheduler.state_dict(), 'logger': logger}
+            save(save_result, './output/model/{}_checkpoint.pt'.format(cfg['model_tag']))
+            cur = logger.mean['test/{}'.format(cfg['pivot_metric'])]
+            better = cur > cfg['pivot'] if pivot_sign > 0 else cur < cfg['pivot']
+            if better:
+                cfg['pivot'] = cur
+                shutil.copy('./output/model/{}_checkpoint.pt'.format(cfg['model_tag']),
+                            './output/model/{}_best.pt'.format(cfg['model_tag']))
+            logger.reset()
+        logger.safe(False)
+
+
 def run_fed_eval(cfg, metric_name, result_key='test'):
     """Evaluation entry (reference: src/test_classifier_fed.py:41-60): load
     {tag}_best.pt, re-run sBN stats (vision), evaluate, save

@@ -89,3 +89,32 @@ def test_scale_mask_flags():
 def test_model_mode_parse():
     rates, props = parse_model_mode('a5-b10-e1')
     assert rates == [1, 0.5, 0.0625] and props == [5, 10, 1]
+
+
+def test_profiler_summary():
+    """Analytic cost profiler: full-rate resnet18 on CIFAR shape is ~11.2M
+    params (reference poster quotes 9.6M trainable-equivalent scale)."""
+    from tests.conftest import make_cfg
+    from heterofl_amd.config import default_config
+    from heterofl_amd.profiler import summarize_level
+    cfg = default_config()
+    cfg['device'] = 'cpu'
+    cfg = make_cfg(cfg, '1_10_0.1_iid_fix_a1-e1_bn_1_1',
+                   data_name='CIFAR10', model_name='resnet18')
+    s_full = summarize_level(cfg, 1.0)
+    s_e = summarize_level(cfg, 0.0625)
+    assert 9e6 < s_full['num_params'] < 13e6
+    assert s_e['num_params'] < s_full['num_params'] / 100
+    assert s_full['num_flops'] > 10 * s_e['num_flops']
+
+
+def test_process_make_stats(tmp_path):
+    from process import make_stats
+    from heterofl_amd.utils import save
+    for lv, p in [('a', 100), ('e', 2)]:
+        save({'num_params': p, 'num_flops': 10 * p, 'space': p / 4},
+             str(tmp_path / 'CIFAR10_resnet18_{}.pt'.format(lv)))
+    st = make_stats('CIFAR10', 'resnet18', 'a1-e1', {'a': 1, 'e': 0.0625},
+                    result_dir=str(tmp_path))
+    assert abs(st['Params'] - 51) < 1e-6
+    assert abs(st['Ratio'] - 0.51) < 1e-6
