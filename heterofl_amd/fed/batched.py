@@ -516,8 +516,29 @@ class BatchedClientTrainer:
         pack_states(model, locals_list)
         model.train(True)
         params = [p for p in model.parameters() if p.requires_grad]
-        opt = torch.optim.SGD(params, lr=lr, momentum=cfg['momentum'],
-                              weight_decay=cfg['weight_decay'])
+        native = native_ops.use_native(device)
+        if native:
+            # identical kernels to the graphed path: backward into stable
+            # .grad buffers + fused per-client clip+momentum-SGD
+            if not hasattr(self, '_opt_cache'):
+                self._opt_cache = {}
+            cached = self._opt_cache.get((rate, R))
+            if cached is None or cached[0] is not model:
+                from ..ops.fused import FusedClipSGD
+                for p in params:
+                    p.grad = torch.zeros_like(p)
+                bufs = [torch.zeros_like(p) for p in params]
+                cached = (model, FusedClipSGD(params,
+                                              [p.grad for p in params],
+                                              bufs, R, device))
+                self._opt_cache[(rate, R)] = cached
+            fopt = cached[1]
+            for b in fopt._refs[2]:
+                b.zero_()
+            opt = None
+        else:
+            opt = torch.optim.SGD(params, lr=lr, momentum=cfg['momentum'],
+                                  weight_decay=cfg['weight_decay'])
         # label masks (R, classes)
         masks = None
         if cfg['mask']:
@@ -543,13 +564,19 @@ class BatchedClientTrainer:
                 xb = x_all[off:off + bs]
                 yb = y_all[off:off + bs]
                 off += bs
-                opt.zero_grad(set_to_none=True)
+                if native:
+                    torch._foreach_zero_([p.grad for p in params])
+                else:
+                    opt.zero_grad(set_to_none=True)
                 with torch.autocast('cuda', torch.bfloat16, enabled=self._amp):
                     scores = model(xb)
-                    losses = batched_masked_ce(scores.float(), yb, masks)
+                losses = batched_masked_ce(scores.float(), yb, masks)
                 losses.sum().backward()
-                per_client_clip_(params, R, 1.0)
-                opt.step()
+                if native:
+                    fopt.step(1.0, lr, cfg['momentum'], cfg['weight_decay'])
+                else:
+                    per_client_clip_(params, R, 1.0)
+                    opt.step()
                 if logger is not None:
                     with torch.no_grad():
                         pred = scores.argmax(dim=2)

@@ -1,0 +1,83 @@
+"""Multi-process CPU tests (gloo, world_size 2): the padded all-reduce
+combine must equal the sequential combine on the union of both ranks'
+clients (SURVEY §7 step 6 verification contract)."""
+import os
+
+import pytest
+import torch
+import torch.multiprocessing as mp
+
+from tests.conftest import make_cfg
+
+
+def _worker(rank, world, port, tmpdir):
+    os.environ['MASTER_ADDR'] = '127.0.0.1'
+    os.environ['MASTER_PORT'] = str(port)
+    os.environ['WORLD_SIZE'] = str(world)
+    os.environ['RANK'] = str(rank)
+    os.environ['LOCAL_RANK'] = str(rank)
+    torch.distributed.init_process_group('gloo', rank=rank, world_size=world)
+    from heterofl_amd.config import default_config
+    from heterofl_amd.data import fetch_dataset, split_dataset
+    from heterofl_amd.fed import FedRunner
+    from heterofl_amd.models import make_model
+    from heterofl_amd.parallel import init_distributed
+    from heterofl_amd.utils import process_dataset, make_optimizer
+
+    cfg = default_config()
+    cfg['device'] = 'cpu'
+    cfg['engine'] = 'sequential'
+    cfg['metric_name'] = {'train': {'Local': ['Local-Loss']},
+                          'test': {'Global': ['Global-Loss']}}
+    cfg = make_cfg(cfg, '1_6_1_iid_fix_a1-e1_bn_1_1',
+                   data_name='MNIST', model_name='conv')
+    cfg['num_epochs'] = {'global': 1, 'local': 1}
+    cfg['world_size'] = world
+
+    torch.manual_seed(0)
+    ds = fetch_dataset('MNIST', synthetic=True, synthetic_size=60)
+    process_dataset(ds, cfg)
+    torch.manual_seed(7)
+    data_split, label_split = split_dataset(ds, 6, 'iid', cfg['classes_size'])
+    torch.manual_seed(1)
+    model = make_model(cfg)
+    opt = make_optimizer(model, cfg['lr'], cfg)
+    ctx = init_distributed(backend='gloo')
+    runner = FedRunner(cfg, ds, data_split, label_split, model, opt,
+                       dist_ctx=ctx)
+    runner.train_round(1)
+    gp = {k: v.clone() for k, v in runner.federation.global_parameters.items()}
+
+    # sequential oracle: same round on one rank, all clients
+    cfg2 = dict(cfg)
+    cfg2['world_size'] = 1
+    torch.manual_seed(1)
+    model2 = make_model(cfg2)
+    opt2 = make_optimizer(model2, cfg2['lr'], cfg2)
+    runner2 = FedRunner(cfg2, ds, data_split, label_split, model2, opt2)
+    # force the same seeded sampling path the distributed round used
+    runner2.cfg['world_size'] = 2
+    runner2.dist_ctx = None
+    g = runner2._round_generator(1)
+    from heterofl_amd.fed.runner import sample_active_users
+    user_idx = sample_active_users(cfg2, 1, generator=g)
+    runner2.federation.make_model_rate(generator=g)
+    local_parameters, param_idx = runner2.federation.distribute(
+        user_idx, resample=False)
+    trained = dict(runner2.trainer.train_clients(
+        list(range(len(user_idx))), user_idx, local_parameters,
+        runner2.federation.model_rate, runner2._make_loader,
+        label_split, cfg2['lr']))
+    ordered = [trained[m] for m in range(len(user_idx))]
+    runner2.federation.combine(ordered, param_idx, user_idx)
+
+    for k, v in runner2.federation.global_parameters.items():
+        if v.is_floating_point():
+            diff = (gp[k] - v).abs().max().item()
+            assert diff < 1e-5, (rank, k, diff)
+    torch.distributed.destroy_process_group()
+
+
+def test_distributed_combine_matches_sequential(tmp_path):
+    port = 29541
+    mp.spawn(_worker, args=(2, port, str(tmp_path)), nprocs=2, join=True)

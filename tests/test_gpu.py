@@ -184,7 +184,7 @@ def test_fused_bn_relu_matches_torch(base_cfg):
         x = torch.randn(N, C, int(HW ** 0.5) if int(HW ** 0.5) ** 2 == HW else 1,
                         HW // (int(HW ** 0.5) if int(HW ** 0.5) ** 2 == HW else 1),
                         device='cuda', requires_grad=True)
-        w = torch.rand(C, device='cuda', requires_grad=True) + 0.5
+        w = (torch.rand(C, device='cuda') + 0.5).requires_grad_()
         b = torch.randn(C, device='cuda', requires_grad=True)
         y = fused_norm_relu(x, w, b, 'bn', 0)
         x2 = x.detach().clone().requires_grad_(True)
@@ -209,7 +209,7 @@ def test_fused_gn_relu_matches_torch(base_cfg):
     torch.manual_seed(0)
     for N, C, H, G in [(10, 40, 8, 20), (10, 64, 16, 4), (4, 16, 4, 16)]:
         x = torch.randn(N, C, H, H, device='cuda', requires_grad=True)
-        w = torch.rand(C, device='cuda', requires_grad=True) + 0.5
+        w = (torch.rand(C, device='cuda') + 0.5).requires_grad_()
         b = torch.randn(C, device='cuda', requires_grad=True)
         y = fused_norm_relu(x, w, b, 'gn', G)
         x2 = x.detach().clone().requires_grad_(True)
