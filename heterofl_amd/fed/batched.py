@@ -49,6 +49,10 @@ class BConv2d(nn.Module):
             self.bias = None
 
     def forward(self, x):
+        if native_ops.use_native(x) and native_ops.native_conv_enabled():
+            from ..ops.fused import grouped_conv
+            return grouped_conv(x, self.weight, self.bias, self.R,
+                                self.stride, self.padding)
         return F.conv2d(x, self.weight, self.bias, stride=self.stride,
                         padding=self.padding, groups=self.R)
 

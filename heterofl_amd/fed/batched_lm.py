@@ -1,0 +1,205 @@
+"""Batched-client transformer engine: R same-rate clients' masked-LM models
+trained as ONE model whose every parameter is the (R, *local_shape) stack of
+the clients' parameters.  LM clients own whole rows of the batchified token
+matrix (often a single row), so per-client work is tiny — batching clients
+into bmm-shaped ops is what fills an MI355X (same argument as fed/batched.py
+for vision).
+
+State-dict keys mirror models/transformer.py exactly (stacked shapes), so
+fed pack/unpack is mechanical (reference slicing rules: src/fed.py:104-156).
+Reference model semantics: src/models/transformer.py:11-174.
+"""
+import math
+
+import numpy as np
+import torch
+import torch.nn as nn
+import torch.nn.functional as F
+
+
+class SLinear(nn.Module):
+    """R stacked Linears: weight (R, out, in), x (R, B, in) -> (R, B, out)."""
+
+    def __init__(self, R, in_f, out_f):
+        super().__init__()
+        self.weight = nn.Parameter(torch.empty(R, out_f, in_f).normal_(0, 0.02))
+        self.bias = nn.Parameter(torch.zeros(R, out_f))
+
+    def forward(self, x):
+        lead = x.shape[:-1]
+        R = self.weight.size(0)
+        flat = x.reshape(R, -1, x.size(-1))
+        out = torch.baddbmm(self.bias.unsqueeze(1), flat,
+                            self.weight.transpose(1, 2))
+        return out.reshape(*lead, self.weight.size(1))
+
+
+class SLayerNorm(nn.Module):
+    """Per-client LayerNorm over the last dim: weight/bias (R, E)."""
+
+    def __init__(self, R, E):
+        super().__init__()
+        self.weight = nn.Parameter(torch.ones(R, E))
+        self.bias = nn.Parameter(torch.zeros(R, E))
+
+    def forward(self, x):
+        # x (R, ..., E)
+        mu = x.mean(-1, keepdim=True)
+        var = x.var(-1, unbiased=False, keepdim=True)
+        xhat = (x - mu) * torch.rsqrt(var + 1e-5)
+        extra = x.dim() - 2
+        w = self.weight.reshape(self.weight.size(0), *([1] * extra), -1)
+        b = self.bias.reshape(self.bias.size(0), *([1] * extra), -1)
+        return xhat * w + b
+
+
+class SScaler(nn.Module):
+    def __init__(self, rate):
+        super().__init__()
+        self.rate = rate
+
+    def forward(self, x):
+        return x / self.rate if self.training else x
+
+
+class SEmbedding(nn.Module):
+    """R stacked embedding tables (R, V, E); ids (R, B, S) -> (R, B, S, E)."""
+
+    def __init__(self, R, V, E):
+        super().__init__()
+        self.weight = nn.Parameter(torch.empty(R, V, E).normal_(0, 0.02))
+
+    def forward(self, ids):
+        R, V, E = self.weight.shape
+        flat = self.weight.reshape(R * V, E)
+        off = (torch.arange(R, device=ids.device) * V).view(R, *([1] * (ids.dim() - 1)))
+        return F.embedding(ids + off, flat)
+
+
+class STransformerEmbedding(nn.Module):
+    def __init__(self, R, num_tokens, bptt, E, dropout, rate):
+        super().__init__()
+        self.embedding = SEmbedding(R, num_tokens + 1, E)
+        self.positional_embedding = nn.ModuleDict(
+            {'positional_embedding': SEmbedding(R, bptt, E)})
+        self.norm = SLayerNorm(R, E)
+        self.dropout = nn.Dropout(dropout)
+        self.scaler = SScaler(rate)
+
+    def forward(self, src):
+        R, B, S = src.shape
+        pos = torch.arange(S, device=src.device).view(1, 1, S).expand(R, B, S)
+        x = self.scaler(self.embedding(src)) + \
+            self.scaler(self.positional_embedding['positional_embedding'](pos))
+        return self.dropout(self.norm(x))
+
+
+class SMultiheadAttention(nn.Module):
+    def __init__(self, R, E, num_heads, rate):
+        super().__init__()
+        self.num_heads = num_heads
+        self.temperature = (E // num_heads) ** 0.5
+        self.linear_q = SLinear(R, E, E)
+        self.linear_k = SLinear(R, E, E)
+        self.linear_v = SLinear(R, E, E)
+        self.linear_o = SLinear(R, E, E)
+        self.scaler = SScaler(rate)
+
+    def forward(self, x):
+        R, B, S, E = x.shape
+        H = self.num_heads
+        d = E // H
+        q = self.scaler(self.linear_q(x)).reshape(R, B, S, H, d)
+        k = self.scaler(self.linear_k(x)).reshape(R, B, S, H, d)
+        v = self.scaler(self.linear_v(x)).reshape(R, B, S, H, d)
+        q = q.permute(0, 1, 3, 2, 4).reshape(R * B * H, S, d)
+        k = k.permute(0, 1, 3, 2, 4).reshape(R * B * H, S, d)
+        v = v.permute(0, 1, 3, 2, 4).reshape(R * B * H, S, d)
+        scores = torch.bmm(q, k.transpose(1, 2)) / self.temperature
+        attn = F.softmax(scores, dim=-1)
+        out = torch.bmm(attn, v)
+        out = out.reshape(R, B, H, S, d).permute(0, 1, 3, 2, 4).reshape(R, B, S, E)
+        return self.scaler(self.linear_o(out))
+
+
+class STransformerEncoderLayer(nn.Module):
+    def __init__(self, R, E, num_heads, hidden, dropout, rate):
+        super().__init__()
+        self.mha = SMultiheadAttention(R, E, num_heads, rate)
+        self.dropout = nn.Dropout(dropout)
+        self.norm1 = SLayerNorm(R, E)
+        self.linear1 = SLinear(R, E, hidden)
+        self.dropout1 = nn.Dropout(dropout)
+        self.linear2 = SLinear(R, hidden, E)
+        self.dropout2 = nn.Dropout(dropout)
+        self.norm2 = SLayerNorm(R, E)
+        self.scaler = SScaler(rate)
+
+    def forward(self, src):
+        src = self.norm1(src + self.dropout(self.mha(src)))
+        h = self.scaler(self.linear2(self.dropout1(
+            F.gelu(self.scaler(self.linear1(src))))))
+        return self.norm2(src + self.dropout2(h))
+
+
+class SDecoder(nn.Module):
+    def __init__(self, R, num_tokens, E, rate):
+        super().__init__()
+        self.linear1 = SLinear(R, E, E)
+        self.scaler = SScaler(rate)
+        self.norm1 = SLayerNorm(R, E)
+        self.linear2 = SLinear(R, E, num_tokens)
+
+    def forward(self, x):
+        return self.linear2(self.norm1(F.gelu(self.scaler(self.linear1(x)))))
+
+
+class BatchedTransformer(nn.Module):
+    """R same-rate clients' Transformers.  Keys mirror models.transformer
+    .Transformer with an (R, ...) leading stack dim on every tensor."""
+
+    def __init__(self, R, num_tokens, bptt, E, num_heads, hidden, num_layers,
+                 dropout, rate, mask_rate):
+        super().__init__()
+        self.R = R
+        self.num_tokens = num_tokens
+        self.mask_rate = mask_rate
+        self.transformer_embedding = STransformerEmbedding(
+            R, num_tokens, bptt, E, dropout, rate)
+        self.transformer_encoder = nn.ModuleDict({'layers': nn.ModuleList(
+            [STransformerEncoderLayer(R, E, num_heads, hidden, dropout, rate)
+             for _ in range(num_layers)])})
+        self.decoder = SDecoder(R, num_tokens, E, rate)
+
+    def forward(self, tokens):
+        """tokens (R, B, S) -> logits (R, B, S, V)."""
+        mask = torch.bernoulli(
+            torch.full(tokens.shape, self.mask_rate, device=tokens.device))
+        src = tokens.masked_fill(mask == 1, self.num_tokens).detach()
+        x = self.transformer_embedding(src)
+        for layer in self.transformer_encoder['layers']:
+            x = layer(x)
+        return self.decoder(x)
+
+
+def lm_masked_ce(logits, tokens, label_masks):
+    """Per-client mean CE over all positions with optional vocab masking
+    (reference: src/models/transformer.py:156-161).  logits (R, B, S, V),
+    tokens (R, B, S), label_masks (R, V) in {0,1} or None.  Returns (R,)."""
+    R, B, S, V = logits.shape
+    if label_masks is not None:
+        logits = logits.masked_fill(
+            label_masks.view(R, 1, 1, V) == 0, 0)
+    logp = F.log_softmax(logits, dim=-1)
+    nll = -logp.gather(3, tokens.unsqueeze(3)).squeeze(3)
+    return nll.reshape(R, -1).mean(1)
+
+
+def make_batched_transformer(cfg, rate, R):
+    tcfg = cfg['transformer']
+    E = int(np.ceil(rate * tcfg['embedding_size']))
+    hidden = int(np.ceil(rate * tcfg['hidden_size']))
+    scaler_rate = rate / cfg['global_model_rate']
+    return BatchedTransformer(R, cfg['num_tokens'], cfg['bptt'], E,
+                              tcfg['num_heads'], hidden, tcfg['num_layers'],
+                              tcfg['dropout'], scaler_rate, cfg['mask_rate'])

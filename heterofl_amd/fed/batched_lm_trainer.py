@@ -1,0 +1,118 @@
+"""Trainer driving the batched transformer engine: the round's active LM
+clients grouped by (rate, rows, windows) and trained together.
+
+Reference local loop: src/train_transformer_fed.py:158-176 — each client
+iterates the bptt windows of its own rows of the batchified token matrix for
+num_epochs['local'] epochs with clip(1.0)+momentum-SGD.
+"""
+import torch
+
+from .. import ops as native_ops
+from .batched import pack_states, unpack_states, per_client_clip_
+from .batched_lm import make_batched_transformer, lm_masked_ce
+
+
+class BatchedLMClientTrainer:
+    def __init__(self, cfg):
+        self.cfg = cfg
+        self.device = torch.device(cfg['device'])
+        self.token = None
+        self.data_split = None
+        self._amp = (cfg.get('compute_dtype') == 'bfloat16'
+                     and self.device.type == 'cuda')
+        self._model_cache = {}
+        self._opt_cache = {}
+
+    def set_data(self, dataset, data_split):
+        self.token = dataset['train'].token.to(self.device)
+        self.data_split = data_split['train']
+
+    def _model(self, rate, R):
+        key = (rate, R)
+        if key not in self._model_cache:
+            self._model_cache[key] = make_batched_transformer(
+                self.cfg, rate, R).to(self.device)
+        return self._model_cache[key]
+
+    def train_clients(self, client_slots, user_idx, local_parameters,
+                      model_rate, make_loader, label_split, lr, logger=None):
+        cfg = self.cfg
+        groups = {}
+        for m in client_slots:
+            u = user_idx[m]
+            groups.setdefault((model_rate[u], len(self.data_split[u])),
+                              []).append(m)
+        out = []
+        for (rate, n_rows), slots in groups.items():
+            out.extend(self._train_group(rate, slots, user_idx,
+                                         [local_parameters[m] for m in slots],
+                                         label_split, lr, logger))
+        return out
+
+    def _train_group(self, rate, slots, user_idx, locals_list, label_split,
+                     lr, logger=None):
+        cfg = self.cfg
+        R = len(slots)
+        device = self.device
+        bptt = cfg['bptt']
+        model = self._model(rate, R)
+        pack_states(model, locals_list)
+        model.train(True)
+        params = [p for p in model.parameters() if p.requires_grad]
+        native = native_ops.use_native(device)
+        if native:
+            from ..ops.fused import FusedClipSGD
+            cached = self._opt_cache.get((rate, R))
+            if cached is None or cached[0] is not model:
+                for p in params:
+                    p.grad = torch.zeros_like(p)
+                bufs = [torch.zeros_like(p) for p in params]
+                cached = (model, FusedClipSGD(params,
+                                              [p.grad for p in params],
+                                              bufs, R, device))
+                self._opt_cache[(rate, R)] = cached
+            fopt = cached[1]
+            for b in fopt._refs[2]:
+                b.zero_()
+        else:
+            opt = torch.optim.SGD(params, lr=lr, momentum=cfg['momentum'],
+                                  weight_decay=cfg['weight_decay'])
+        masks = None
+        if cfg['mask']:
+            masks = torch.zeros(R, cfg['num_tokens'], device=device)
+            for i, m in enumerate(slots):
+                masks[i, label_split[user_idx[m]]] = 1
+        # client rows of the batchified token matrix, stacked (R, B, L)
+        rows = torch.stack([self.token[self.data_split[user_idx[m]]]
+                            for m in slots])
+        L = rows.size(2)
+        n_win = (L + bptt - 1) // bptt
+        for _ in range(cfg['num_epochs']['local']):
+            for w in range(n_win):
+                tokens = rows[:, :, w * bptt:(w + 1) * bptt]
+                if native:
+                    torch._foreach_zero_([p.grad for p in params])
+                else:
+                    opt.zero_grad(set_to_none=True)
+                with torch.autocast('cuda', torch.bfloat16,
+                                    enabled=self._amp):
+                    logits = model(tokens)
+                losses = lm_masked_ce(logits.float(), tokens, masks)
+                losses.sum().backward()
+                if native:
+                    fopt.step(1.0, lr, cfg['momentum'], cfg['weight_decay'])
+                else:
+                    per_client_clip_(params, R, 1.0)
+                    opt.step()
+                if logger is not None:
+                    with torch.no_grad():
+                        n = tokens.size(1) * tokens.size(2)
+                        for i in range(R):
+                            l = losses[i].item()
+                            logger.append({'Local-Loss': l,
+                                           'Local-Perplexity':
+                                               float(torch.exp(torch.tensor(l)))},
+                                          'train', n=n)
+        template_keys = list(locals_list[0].keys())
+        states = unpack_states(model, template_keys)
+        return list(zip(slots, [dict(st) for st in states]))

@@ -76,9 +76,12 @@ class FedRunner:
         self.federation = Federation(model.state_dict(), cfg['model_rate'],
                                      label_split, cfg)
         self.is_lm = cfg['model_name'] == 'transformer'
-        use_batched = (cfg.get('engine', 'sequential') == 'batched'
-                       and not self.is_lm)
-        if use_batched:
+        use_batched = cfg.get('engine', 'sequential') == 'batched'
+        if use_batched and self.is_lm:
+            from .batched_lm_trainer import BatchedLMClientTrainer
+            self.trainer = BatchedLMClientTrainer(cfg)
+            self.trainer.set_data(dataset, data_split)
+        elif use_batched:
             from .batched import BatchedClientTrainer
             self.trainer = BatchedClientTrainer(cfg)
             self.trainer.set_data(dataset, data_split)

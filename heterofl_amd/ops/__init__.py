@@ -58,3 +58,8 @@ def use_native(tensor_or_device=None):
     else:
         is_cuda = torch.cuda.is_available()
     return is_cuda
+
+
+def native_conv_enabled():
+    """The MFMA conv path can be toggled off for A/B benchmarking."""
+    return os.environ.get('HETEROFL_NATIVE_CONV', '1') == '1'

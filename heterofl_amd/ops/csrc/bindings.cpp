@@ -23,6 +23,13 @@ std::vector<at::Tensor> build_chunk_table(std::vector<at::Tensor> grads,
 void clip_sgd_step(at::Tensor table_blob, int64_t n_chunks, at::Tensor normsq,
                    double max_norm, double lr, double momentum,
                    double weight_decay);
+at::Tensor conv_fwd(at::Tensor x, at::Tensor w, at::Tensor bias,
+                    int64_t groups, int64_t stride, int64_t pad);
+at::Tensor conv_bwd_data(at::Tensor dy, at::Tensor w, int64_t groups,
+                         int64_t stride, int64_t pad, int64_t H, int64_t W);
+at::Tensor conv_bwd_weight(at::Tensor dy, at::Tensor x, int64_t groups,
+                           int64_t stride, int64_t pad, int64_t khw);
+at::Tensor mfma_probe(at::Tensor A, at::Tensor B);
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("bn_relu_fwd", &bn_relu_fwd, "fused sBN+ReLU forward");
@@ -34,4 +41,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("build_chunk_table", &build_chunk_table,
           "build clip+SGD chunk table");
     m.def("clip_sgd_step", &clip_sgd_step, "fused per-client clip+SGD step");
+    m.def("conv_fwd", &conv_fwd, "MFMA implicit-GEMM grouped conv forward");
+    m.def("conv_bwd_data", &conv_bwd_data, "MFMA grouped conv backward-data");
+    m.def("conv_bwd_weight", &conv_bwd_weight,
+          "MFMA grouped conv backward-weight (fp32 out)");
+    m.def("mfma_probe", &mfma_probe, "16x16x32 bf16 MFMA layout probe");
 }

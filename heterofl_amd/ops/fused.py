@@ -112,3 +112,39 @@ class FusedClipSGD:
         self.normsq.zero_()
         ext.clip_sgd_step(self.table, self.n_chunks, self.normsq, max_norm,
                           lr, momentum, weight_decay)
+
+
+class _GroupedConv(torch.autograd.Function):
+    """MFMA implicit-GEMM grouped conv (ops/csrc/conv_mfma.hip): bf16/fp32
+    activations, fp32 master weights, fp32 weight gradients."""
+
+    @staticmethod
+    def forward(ctx, x, weight, bias, groups, stride, pad):
+        ext = require_native()
+        x = x.contiguous()
+        y = ext.conv_fwd(x, weight,
+                         bias if bias is not None else torch.Tensor(),
+                         groups, stride, pad)
+        ctx.save_for_backward(x, weight)
+        ctx.meta = (groups, stride, pad, bias is not None)
+        return y
+
+    @staticmethod
+    def backward(ctx, dy):
+        ext = require_native()
+        x, w = ctx.saved_tensors
+        groups, stride, pad, has_bias = ctx.meta
+        dy = dy.contiguous()
+        dx = dw = db = None
+        if ctx.needs_input_grad[0]:
+            dx = ext.conv_bwd_data(dy, w, groups, stride, pad,
+                                   x.size(2), x.size(3))
+        if ctx.needs_input_grad[1]:
+            dw = ext.conv_bwd_weight(dy, x, groups, stride, pad, w.size(2))
+        if has_bias and ctx.needs_input_grad[2]:
+            db = dy.float().sum(dim=(0, 2, 3))
+        return dx, dw, db, None, None, None
+
+
+def grouped_conv(x, weight, bias, groups, stride, pad):
+    return _GroupedConv.apply(x, weight, bias, groups, stride, pad)

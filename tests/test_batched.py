@@ -160,3 +160,60 @@ def test_batched_engine_e2e(base_cfg):
     runner.train_round(1)
     tm = runner.stats()
     runner.test(tm, 1)
+
+
+def test_batched_lm_matches_sequential(base_cfg):
+    """BatchedTransformer (R stacked clients) == sequential per-client
+    training, with RNG-free config (mask_rate 0, dropout 0)."""
+    import torch
+    from tests.conftest import make_cfg
+    from heterofl_amd.data import fetch_dataset, split_dataset
+    from heterofl_amd.fed.batched_lm_trainer import BatchedLMClientTrainer
+    from heterofl_amd.fed.sequential import SequentialClientTrainer
+    from heterofl_amd.data import SplitDataset, BatchDataset
+    from heterofl_amd.models import make_model
+    from heterofl_amd.utils import process_dataset
+
+    cfg = make_cfg(base_cfg, '1_3_1_iid_fix_a1_bn_1_1',
+                   data_name='WikiText2', model_name='transformer')
+    cfg['num_epochs'] = {'global': 1, 'local': 2}
+    cfg['transformer']['dropout'] = 0.0
+    cfg['mask_rate'] = 0.0
+    cfg['metric_name'] = {'train': {'Local': ['Local-Loss']}}
+    torch.manual_seed(0)
+    ds = fetch_dataset('WikiText2', synthetic=True, synthetic_size=60_000)
+    # small vocab so CE is learnable-sized
+    ds['train'].vocab.itos = ds['train'].vocab.itos[:500]
+    ds['train'].token = ds['train'].token % 500
+    ds['test'].token = ds['test'].token % 500
+    process_dataset(ds, cfg)
+    cfg['num_tokens'] = 500
+    data_split, label_split = split_dataset(ds, 3, 'iid')
+    torch.manual_seed(1)
+    global_model = make_model(cfg)
+    gp = global_model.state_dict()
+    locals_ = [{k: v.clone() for k, v in gp.items()} for _ in range(3)]
+    user_idx = [0, 1, 2]
+    rates = {u: 1.0 for u in user_idx}
+
+    def make_loader(user):
+        return BatchDataset(SplitDataset(ds['train'],
+                                         data_split['train'][user]),
+                            cfg['bptt'])
+
+    seq = SequentialClientTrainer(cfg)
+    torch.manual_seed(5)
+    seq_out = dict(seq.train_clients([0, 1, 2], user_idx,
+                                     [dict(l) for l in locals_], rates,
+                                     make_loader, label_split, 0.1))
+    bt = BatchedLMClientTrainer(cfg)
+    bt.set_data(ds, data_split)
+    torch.manual_seed(5)
+    bt_out = dict(bt.train_clients([0, 1, 2], user_idx,
+                                   [dict(l) for l in locals_], rates,
+                                   make_loader, label_split, 0.1))
+    for m in range(3):
+        for k in seq_out[m]:
+            a, b = seq_out[m][k].float(), bt_out[m][k].float()
+            diff = (a - b).abs().max().item()
+            assert diff < 5e-4, (m, k, diff)

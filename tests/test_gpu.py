@@ -284,3 +284,70 @@ def test_fused_clip_sgd_matches_torch(base_cfg):
                 assert (params[i] - ref_p[i]).abs().max().item() < 1e-5, i
                 assert (bufs[i] - ref_b[i]).abs().max().item() < 1e-5, i
     torch.cuda.synchronize()
+
+
+@needs_gpu
+def test_mfma_probe_layout():
+    """Pin the 16x16x32 bf16 MFMA fragment layout empirically (asymmetric
+    operands catch transposes, cdna guide §5.4 rule 16)."""
+    from heterofl_amd.ops import require_native
+    ext = require_native()
+    torch.manual_seed(0)
+    A = torch.randn(16, 32)
+    B = torch.randn(32, 16)
+    D = ext.mfma_probe(A, B).cpu()
+    ref = (A.to(torch.bfloat16).float() @ B.to(torch.bfloat16).float())
+    assert (D - ref).abs().max().item() < 0.15, \
+        (D - ref).abs().max().item()
+
+
+@needs_gpu
+def test_mfma_conv_matches_torch():
+    """MFMA grouped conv fwd/bwd vs F.conv2d across every HeteroFL shape
+    family (3x3 s1/s2, 1x1 s2, stem, ragged channels)."""
+    import torch.nn.functional as F
+    from heterofl_amd.ops.fused import grouped_conv
+    torch.manual_seed(0)
+    shapes = [
+        # (G, N, Cin, H, Cout, k, stride, pad)
+        (5, 10, 3, 32, 64, 3, 1, 1),      # stem
+        (5, 10, 64, 32, 64, 3, 1, 1),     # layer1
+        (5, 10, 64, 32, 128, 3, 2, 1),    # downsample
+        (5, 10, 64, 32, 128, 1, 2, 0),    # 1x1 shortcut
+        (5, 10, 512, 4, 512, 3, 1, 1),    # layer4
+        (3, 10, 4, 32, 4, 3, 1, 1),       # rate-1/16 ragged
+        (2, 7, 20, 16, 36, 3, 2, 1),      # odd everything
+    ]
+    for G, N, Cin, H, Cout, k, s, p in shapes:
+        x = torch.randn(N, G * Cin, H, H, device='cuda', requires_grad=True)
+        w = torch.randn(G * Cout, Cin, k, k, device='cuda',
+                        requires_grad=True) * 0.1
+        y = grouped_conv(x, w, None, G, s, p)
+        x2 = x.detach().clone().requires_grad_(True)
+        w2 = w.detach().clone().requires_grad_(True)
+        ref = F.conv2d(x2, w2, None, stride=s, padding=p, groups=G)
+        ferr = (y - ref).abs().max().item()
+        assert ferr < 5e-4 * Cin * k, ('fwd', G, Cin, H, Cout, k, s, ferr)
+        g = torch.randn_like(y)
+        y.backward(g)
+        ref.backward(g)
+        derr = (x.grad - x2.grad).abs().max().item()
+        werr = (w.grad - w2.grad).abs().max().item()
+        assert derr < 5e-4 * Cout * k, ('bwd_data', G, Cin, H, Cout, s, derr)
+        assert werr < 5e-3 * N * H, ('bwd_w', G, Cin, H, Cout, s, werr)
+
+
+@needs_gpu
+def test_mfma_conv_bias_bf16():
+    import torch.nn.functional as F
+    from heterofl_amd.ops.fused import grouped_conv
+    torch.manual_seed(0)
+    G, N, Cin, H, Cout = 3, 10, 16, 28, 32
+    x = torch.randn(N, G * Cin, H, H, device='cuda', dtype=torch.bfloat16)
+    w = torch.randn(G * Cout, Cin, 3, 3, device='cuda') * 0.1
+    b = torch.randn(G * Cout, device='cuda')
+    y = grouped_conv(x, w, b, G, 1, 1)
+    ref = F.conv2d(x.float(), w, b, stride=1, padding=1, groups=G)
+    rel = (y.float() - ref).abs().max().item() / ref.abs().max().item()
+    assert y.dtype == torch.bfloat16
+    assert rel < 0.05, rel
