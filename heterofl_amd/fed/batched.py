@@ -48,13 +48,14 @@ class BConv2d(nn.Module):
         else:
             self.bias = None
 
-    def forward(self, x):
+    def forward(self, x, residual=None):
         if native_ops.use_native(x) and native_ops.native_conv_enabled():
             from ..ops.fused import grouped_conv
             return grouped_conv(x, self.weight, self.bias, self.R,
-                                self.stride, self.padding)
-        return F.conv2d(x, self.weight, self.bias, stride=self.stride,
-                        padding=self.padding, groups=self.R)
+                                self.stride, self.padding, residual=residual)
+        y = F.conv2d(x, self.weight, self.bias, stride=self.stride,
+                     padding=self.padding, groups=self.R)
+        return y if residual is None else y + residual
 
 
 class BBatchNorm2d(nn.Module):
@@ -179,9 +180,8 @@ class BBlock(nn.Module):
         out = self.n1(x)
         shortcut = self.shortcut(out) if hasattr(self, 'shortcut') else x
         out = self.conv1(out)
-        out = self.conv2(self.n2(out))
-        out += shortcut
-        return out
+        # residual add fused into conv2's epilogue on the native path
+        return self.conv2(self.n2(out), residual=shortcut)
 
 
 class BatchedResNet(nn.Module):

@@ -24,7 +24,8 @@ void clip_sgd_step(at::Tensor table_blob, int64_t n_chunks, at::Tensor normsq,
                    double max_norm, double lr, double momentum,
                    double weight_decay);
 at::Tensor conv_fwd(at::Tensor x, at::Tensor w, at::Tensor bias,
-                    int64_t groups, int64_t stride, int64_t pad);
+                    at::Tensor residual, int64_t groups, int64_t stride,
+                    int64_t pad);
 at::Tensor conv_bwd_data(at::Tensor dy, at::Tensor w, int64_t groups,
                          int64_t stride, int64_t pad, int64_t H, int64_t W);
 at::Tensor conv_bwd_weight(at::Tensor dy, at::Tensor x, int64_t groups,

@@ -119,11 +119,13 @@ class _GroupedConv(torch.autograd.Function):
     activations, fp32 master weights, fp32 weight gradients."""
 
     @staticmethod
-    def forward(ctx, x, weight, bias, groups, stride, pad):
+    def forward(ctx, x, weight, bias, residual, groups, stride, pad):
         ext = require_native()
         x = x.contiguous()
         y = ext.conv_fwd(x, weight,
                          bias if bias is not None else torch.Tensor(),
+                         residual.contiguous() if residual is not None
+                         else torch.Tensor(),
                          groups, stride, pad)
         ctx.save_for_backward(x, weight)
         ctx.meta = (groups, stride, pad, bias is not None)
@@ -143,8 +145,10 @@ class _GroupedConv(torch.autograd.Function):
             dw = ext.conv_bwd_weight(dy, x, groups, stride, pad, w.size(2))
         if has_bias and ctx.needs_input_grad[2]:
             db = dy.float().sum(dim=(0, 2, 3))
-        return dx, dw, db, None, None, None
+        # residual gradient is dy itself (identity add in the epilogue)
+        dres = dy if ctx.needs_input_grad[3] else None
+        return dx, dw, db, dres, None, None, None
 
 
-def grouped_conv(x, weight, bias, groups, stride, pad):
-    return _GroupedConv.apply(x, weight, bias, groups, stride, pad)
+def grouped_conv(x, weight, bias, groups, stride, pad, residual=None):
+    return _GroupedConv.apply(x, weight, bias, residual, groups, stride, pad)

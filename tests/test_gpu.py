@@ -320,8 +320,8 @@ def test_mfma_conv_matches_torch():
     ]
     for G, N, Cin, H, Cout, k, s, p in shapes:
         x = torch.randn(N, G * Cin, H, H, device='cuda', requires_grad=True)
-        w = torch.randn(G * Cout, Cin, k, k, device='cuda',
-                        requires_grad=True) * 0.1
+        w = (torch.randn(G * Cout, Cin, k, k, device='cuda') * 0.1
+             ).requires_grad_()
         y = grouped_conv(x, w, None, G, s, p)
         x2 = x.detach().clone().requires_grad_(True)
         w2 = w.detach().clone().requires_grad_(True)
@@ -351,3 +351,25 @@ def test_mfma_conv_bias_bf16():
     rel = (y.float() - ref).abs().max().item() / ref.abs().max().item()
     assert y.dtype == torch.bfloat16
     assert rel < 0.05, rel
+
+
+@needs_gpu
+def test_mfma_conv_residual_fusion():
+    import torch.nn.functional as F
+    from heterofl_amd.ops.fused import grouped_conv
+    torch.manual_seed(0)
+    G, N, C, H = 4, 10, 32, 16
+    x = torch.randn(N, G * C, H, H, device='cuda', requires_grad=True)
+    w = (torch.randn(G * C, C, 3, 3, device='cuda') * 0.1).requires_grad_()
+    res = torch.randn(N, G * C, H, H, device='cuda', requires_grad=True)
+    y = grouped_conv(x, w, None, G, 1, 1, residual=res)
+    x2 = x.detach().clone().requires_grad_(True)
+    w2 = w.detach().clone().requires_grad_(True)
+    r2 = res.detach().clone().requires_grad_(True)
+    ref = F.conv2d(x2, w2, None, 1, 1, groups=G) + r2
+    assert (y - ref).abs().max().item() < 0.05
+    g = torch.randn_like(y)
+    y.backward(g)
+    ref.backward(g)
+    assert (res.grad - r2.grad).abs().max().item() < 1e-6
+    assert (x.grad - x2.grad).abs().max().item() < 0.05
