@@ -20,9 +20,10 @@ std::vector<at::Tensor> build_chunk_table(std::vector<at::Tensor> grads,
                                           std::vector<at::Tensor> params,
                                           std::vector<at::Tensor> bufs,
                                           int64_t R, int64_t chunk_elems);
-void clip_sgd_step(at::Tensor table_blob, int64_t n_chunks, at::Tensor normsq,
-                   double max_norm, double lr, double momentum,
-                   double weight_decay);
+void clip_sgd_step(at::Tensor table_blob, int64_t n_chunks,
+                   at::Tensor chunk_client, at::Tensor partials,
+                   at::Tensor normsq, double max_norm, double lr,
+                   double momentum, double weight_decay);
 at::Tensor conv_fwd(at::Tensor x, at::Tensor w, at::Tensor bias,
                     at::Tensor residual, int64_t groups, int64_t stride,
                     int64_t pad);

@@ -21,8 +21,11 @@ struct Chunk {             // one contiguous piece of one client's slice
     int client;
 };
 
+// Per-chunk partial square-sums (no atomics); sqnorm_reduce_kernel sums
+// them per client in fixed chunk order so replays are bit-identical.
 __global__ void __launch_bounds__(256)
-sqnorm_kernel(const Chunk* __restrict__ chunks, float* __restrict__ normsq) {
+sqnorm_kernel(const Chunk* __restrict__ chunks,
+              float* __restrict__ partials) {
     const Chunk ck = chunks[blockIdx.x];
     float s = 0.f;
     for (int i = threadIdx.x; i < ck.len; i += blockDim.x) {
@@ -37,8 +40,22 @@ sqnorm_kernel(const Chunk* __restrict__ chunks, float* __restrict__ normsq) {
     if (threadIdx.x == 0) {
         float t = 0.f;
         for (int w = 0; w < blockDim.x / WAVE; ++w) t += scratch[w];
-        atomicAdd(normsq + ck.client, t);
+        partials[blockIdx.x] = t;
     }
+}
+
+__global__ void __launch_bounds__(64)
+sqnorm_reduce_kernel(const float* __restrict__ partials,
+                     const int* __restrict__ chunk_client, int n_chunks,
+                     float* __restrict__ normsq) {
+    const int r = blockIdx.x;
+    float s = 0.f;
+    // fixed serial order within each lane's stripe, then one wave reduce:
+    // the stripe pattern is deterministic for a fixed chunk table
+    for (int i = threadIdx.x; i < n_chunks; i += 64)
+        if (chunk_client[i] == r) s += partials[i];
+    for (int off = WAVE / 2; off > 0; off >>= 1) s += __shfl_down(s, off, WAVE);
+    if (threadIdx.x == 0) normsq[r] = s;
 }
 
 __global__ void __launch_bounds__(256)
@@ -96,18 +113,29 @@ std::vector<at::Tensor> build_chunk_table(std::vector<at::Tensor> grads,
                               at::TensorOptions().dtype(at::kByte))
                     .clone()
                     .to(grads[0].device(), /*non_blocking=*/false);
+    std::vector<int> clients(host.size());
+    for (size_t i = 0; i < host.size(); ++i) clients[i] = host[i].client;
+    auto cl = at::from_blob(clients.data(), {(long)clients.size()},
+                            at::TensorOptions().dtype(at::kInt))
+                  .clone()
+                  .to(grads[0].device(), /*non_blocking=*/false);
     auto n = at::scalar_tensor((long)host.size(), at::kLong);
-    return {blob, n};
+    return {blob, n, cl};
 }
 
-void clip_sgd_step(at::Tensor table_blob, int64_t n_chunks, at::Tensor normsq,
-                   double max_norm, double lr, double momentum,
-                   double weight_decay) {
+void clip_sgd_step(at::Tensor table_blob, int64_t n_chunks,
+                   at::Tensor chunk_client, at::Tensor partials,
+                   at::Tensor normsq, double max_norm, double lr,
+                   double momentum, double weight_decay) {
     auto stream = at::hip::getCurrentHIPStream();
     const Chunk* chunks = (const Chunk*)table_blob.data_ptr();
-    // normsq must be zeroed by the caller (inside the graph: normsq.zero_())
+    const int R = (int)normsq.numel();
     hipLaunchKernelGGL(sqnorm_kernel, dim3((int)n_chunks), dim3(256), 0,
-                       stream, chunks, normsq.data_ptr<float>());
+                       stream, chunks, partials.data_ptr<float>());
+    hipLaunchKernelGGL(sqnorm_reduce_kernel, dim3(R), dim3(64), 0, stream,
+                       partials.data_ptr<float>(),
+                       chunk_client.data_ptr<int>(), (int)n_chunks,
+                       normsq.data_ptr<float>());
     hipLaunchKernelGGL(clip_sgd_step_kernel, dim3((int)n_chunks), dim3(256), 0,
                        stream, chunks, normsq.data_ptr<float>(),
                        (float)max_norm, (float)lr, (float)momentum,

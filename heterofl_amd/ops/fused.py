@@ -101,17 +101,20 @@ class FusedClipSGD:
     def __init__(self, params, grads, bufs, R, device):
         ext = require_native()
         self._refs = (list(params), list(grads), list(bufs))
-        blob, n = ext.build_chunk_table(self._refs[1], self._refs[0],
-                                        self._refs[2], R, self.CHUNK)
+        blob, n, clients = ext.build_chunk_table(
+            self._refs[1], self._refs[0], self._refs[2], R, self.CHUNK)
         self.table = blob
         self.n_chunks = int(n.item())
+        self.chunk_client = clients
+        self.partials = torch.zeros(self.n_chunks, dtype=torch.float32,
+                                    device=device)
         self.normsq = torch.zeros(R, dtype=torch.float32, device=device)
 
     def step(self, max_norm, lr, momentum, weight_decay):
         ext = require_native()
-        self.normsq.zero_()
-        ext.clip_sgd_step(self.table, self.n_chunks, self.normsq, max_norm,
-                          lr, momentum, weight_decay)
+        ext.clip_sgd_step(self.table, self.n_chunks, self.chunk_client,
+                          self.partials, self.normsq, max_norm, lr, momentum,
+                          weight_decay)
 
 
 class _GroupedConv(torch.autograd.Function):
