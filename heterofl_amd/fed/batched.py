@@ -219,6 +219,10 @@ class BatchedResNet(nn.Module):
         out = self.layer3(out)
         out = self.layer4(out)
         out = self.n4(out)
+        if native_ops.use_native(out):
+            from ..ops.fused import fused_head
+            return fused_head(out, self.linear.weight, self.linear.bias,
+                              self.R)
         out = F.adaptive_avg_pool2d(out, 1)
         out = out.view(out.size(0), self.R, -1)
         return self.linear(out)
@@ -251,6 +255,9 @@ class BatchedConv(nn.Module):
 
     def forward(self, x):
         out = self.blocks(x)
+        if native_ops.use_native(out):
+            from ..ops.fused import fused_head
+            return fused_head(out, self.head.weight, self.head.bias, self.R)
         out = F.adaptive_avg_pool2d(out, 1).view(out.size(0), self.R, -1)
         return self.head(out)
 

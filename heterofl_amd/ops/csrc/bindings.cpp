@@ -32,6 +32,11 @@ at::Tensor conv_bwd_data(at::Tensor dy, at::Tensor w, int64_t groups,
 at::Tensor conv_bwd_weight(at::Tensor dy, at::Tensor x, int64_t groups,
                            int64_t stride, int64_t pad, int64_t khw);
 at::Tensor mfma_probe(at::Tensor A, at::Tensor B);
+std::vector<at::Tensor> head_fwd(at::Tensor feat, at::Tensor w, at::Tensor b,
+                                 int64_t R);
+std::vector<at::Tensor> head_bwd(at::Tensor dscores, at::Tensor pooled,
+                                 at::Tensor w, int64_t R, int64_t H,
+                                 int64_t W, bool bf16_feat, bool want_db);
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("bn_relu_fwd", &bn_relu_fwd, "fused sBN+ReLU forward");
@@ -48,4 +53,6 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("conv_bwd_weight", &conv_bwd_weight,
           "MFMA grouped conv backward-weight (fp32 out)");
     m.def("mfma_probe", &mfma_probe, "16x16x32 bf16 MFMA layout probe");
+    m.def("head_fwd", &head_fwd, "fused avgpool+per-client-linear forward");
+    m.def("head_bwd", &head_bwd, "fused head backward");
 }

@@ -78,16 +78,20 @@ struct ConvGeom {
 // -------------------------------------------------------------- fwd kernel
 // grid: (ceil(M/BM), ceil(P/BP), G).  Optional residual is added in the
 // epilogue (the ResNet block's `out += shortcut`, src/models/resnet.py:49).
+// splitk > 1: each split writes fp32 partial slabs (sp, N*G*Cout*OHW);
+// conv_out_reduce_kernel sums them in fixed order and applies bias/residual.
 template <typename T>
 __global__ void __launch_bounds__(256)
 conv_fwd_kernel(const T* __restrict__ x, const float* __restrict__ w,
                 const float* __restrict__ bias, const T* __restrict__ residual,
-                T* __restrict__ y, ConvGeom gm) {
+                T* __restrict__ y, float* __restrict__ partial, ConvGeom gm,
+                int splitk) {
     __shared__ T a_lds[BM][LDK];
     __shared__ T b_lds[BP][LDK];
     __shared__ int t_ihb[BP], t_iwb[BP];
     __shared__ long t_xbase[BP], t_ybase[BP];
-    const int g = blockIdx.z;
+    const int g = blockIdx.z % gm.G;
+    const int sp = blockIdx.z / gm.G;
     const int m0 = blockIdx.x * BM;
     const int p0 = blockIdx.y * BP;
     const int kk2 = gm.khw * gm.khw;
@@ -101,6 +105,9 @@ conv_fwd_kernel(const T* __restrict__ x, const float* __restrict__ w,
     const int wave = tid / WAVE;
     const int wm = (wave >> 1) * 32;
     const int wp = (wave & 1) * 32;
+    const int nkc = ((K + BK - 1) / BK + splitk - 1) / splitk;  // BK-chunks
+    const int ks = sp * nkc * BK;
+    const int ke = min(K, ks + nkc * BK);
 
     // per-block pixel decomposition tables
     if (tid < BP) {
@@ -123,7 +130,7 @@ conv_fwd_kernel(const T* __restrict__ x, const float* __restrict__ w,
     __syncthreads();
 
     f32x4 acc[2][2] = {};
-    for (int k0 = 0; k0 < K; k0 += BK) {
+    for (int k0 = ks; k0 < ke; k0 += BK) {
         for (int e = tid; e < BM * BK; e += 256) {
             const int kk = e & (BK - 1), mm = e >> 5;
             const int m = m0 + mm, k = k0 + kk;
@@ -155,6 +162,7 @@ conv_fwd_kernel(const T* __restrict__ x, const float* __restrict__ w,
                     &b_lds[wp + fp * 16 + (l & 15)][0], acc[fm][fp]);
         __syncthreads();
     }
+    const long slab = (long)sp * gm.N * gm.G * gm.Cout * OHW;
 #pragma unroll
     for (int fm = 0; fm < 2; ++fm)
 #pragma unroll
@@ -167,24 +175,46 @@ conv_fwd_kernel(const T* __restrict__ x, const float* __restrict__ w,
                 if (m < M && yb >= 0) {
                     const long off = yb + (long)m * OHW;
                     float v = acc[fm][fp][r];
-                    if (bias) v += bias[g * gm.Cout + m];
-                    if (residual) v += ld_f32(residual + off);
-                    st_f32(y + off, v);
+                    if (splitk == 1) {
+                        if (bias) v += bias[g * gm.Cout + m];
+                        if (residual) v += ld_f32(residual + off);
+                        st_f32(y + off, v);
+                    } else {
+                        partial[slab + off] = v;
+                    }
                 }
             }
         }
+}
+
+// sum split-K partials in fixed order; add bias/residual; cast to T
+template <typename T>
+__global__ void __launch_bounds__(256)
+conv_out_reduce_kernel(const float* __restrict__ partial,
+                       const float* __restrict__ bias,
+                       const T* __restrict__ residual, T* __restrict__ out,
+                       long total, long chanstride, int nchan, int splitk) {
+    const long i = (long)blockIdx.x * blockDim.x + threadIdx.x;
+    if (i >= total) return;
+    float v = 0.f;
+    for (int sp = 0; sp < splitk; ++sp) v += partial[(long)sp * total + i];
+    if (bias) v += bias[(i / chanstride) % nchan];
+    if (residual) v += ld_f32(residual + i);
+    st_f32(out + i, v);
 }
 
 // -------------------------------------------------------- bwd-data kernel
 template <typename T>
 __global__ void __launch_bounds__(256)
 conv_bwd_data_kernel(const T* __restrict__ dy, const float* __restrict__ w,
-                     T* __restrict__ dx, ConvGeom gm) {
+                     T* __restrict__ dx, float* __restrict__ partial,
+                     ConvGeom gm, int splitk) {
     __shared__ T a_lds[BM][LDK];
     __shared__ T b_lds[BP][LDK];
     __shared__ int t_oh[BP], t_ow[BP];  // ih+pad, iw+pad (pre-division)
     __shared__ long t_dybase[BP], t_xbase[BP];
-    const int g = blockIdx.z;
+    const int g = blockIdx.z % gm.G;
+    const int sp = blockIdx.z / gm.G;
     const int c0 = blockIdx.x * BM;
     const int q0 = blockIdx.y * BP;
     const int kk2 = gm.khw * gm.khw;
@@ -219,8 +249,11 @@ conv_bwd_data_kernel(const T* __restrict__ dy, const float* __restrict__ w,
     }
     __syncthreads();
 
+    const int njc = ((J + BK - 1) / BK + splitk - 1) / splitk;
+    const int js = sp * njc * BK;
+    const int je = min(J, js + njc * BK);
     f32x4 acc[2][2] = {};
-    for (int j0 = 0; j0 < J; j0 += BK) {
+    for (int j0 = js; j0 < je; j0 += BK) {
         for (int e = tid; e < BM * BK; e += 256) {
             const int jj = e & (BK - 1), cc = e >> 5;
             const int c = c0 + cc, j = j0 + jj;
@@ -264,6 +297,7 @@ conv_bwd_data_kernel(const T* __restrict__ dy, const float* __restrict__ w,
                     &b_lds[wp + fp * 16 + (l & 15)][0], acc[fm][fp]);
         __syncthreads();
     }
+    const long slab = (long)sp * gm.N * gm.G * gm.Cin * HW;
 #pragma unroll
     for (int fm = 0; fm < 2; ++fm)
 #pragma unroll
@@ -273,8 +307,12 @@ conv_bwd_data_kernel(const T* __restrict__ dy, const float* __restrict__ w,
 #pragma unroll
             for (int r = 0; r < 4; ++r) {
                 const int c = c0 + wm + fm * 16 + (l >> 4) * 4 + r;
-                if (c < gm.Cin && xb >= 0)
-                    st_f32(dx + xb + (long)c * HW, acc[fm][fp][r]);
+                if (c < gm.Cin && xb >= 0) {
+                    if (splitk == 1)
+                        st_f32(dx + xb + (long)c * HW, acc[fm][fp][r]);
+                    else
+                        partial[slab + xb + (long)c * HW] = acc[fm][fp][r];
+                }
             }
         }
 }
@@ -289,6 +327,9 @@ conv_bwd_weight_kernel(const T* __restrict__ dy, const T* __restrict__ x,
                        float* __restrict__ out, ConvGeom gm, int splitp) {
     __shared__ T a_lds[BM][LDK];
     __shared__ T b_lds[BP][LDK];
+    __shared__ int t_kcin[BP], t_kkh[BP], t_kkw[BP];   // per-block k tables
+    __shared__ int t_ihb[BK], t_iwb[BK];               // per-step p tables
+    __shared__ long t_xb[BK], t_dyb[BK];
     const int g = blockIdx.z % gm.G;
     const int sp = blockIdx.z / gm.G;
     const int m0 = blockIdx.x * BM;
@@ -310,32 +351,57 @@ conv_bwd_weight_kernel(const T* __restrict__ dy, const T* __restrict__ x,
     const int wp = (wave & 1) * 32;
     f32x4 acc[2][2] = {};
 
+    if (tid < BP) {
+        const int k = k0 + tid;
+        if (k < K) {
+            const int cin = k / kk2, r = k - cin * kk2;
+            t_kcin[tid] = cin;
+            t_kkh[tid] = r / gm.khw;
+            t_kkw[tid] = r - (r / gm.khw) * gm.khw;
+        } else {
+            t_kcin[tid] = 0;
+            t_kkh[tid] = 1 << 28;
+            t_kkw[tid] = 1 << 28;
+        }
+    }
+
     for (int p0 = pstart; p0 < pend; p0 += BK) {
+        __syncthreads();
+        if (tid < BK) {
+            const int p = p0 + tid;
+            if (p < pend) {
+                const int n = p / OHW, hw = p - n * OHW;
+                const int oh = hw / gm.OW, ow = hw - oh * gm.OW;
+                t_ihb[tid] = oh * gm.stride - gm.pad;
+                t_iwb[tid] = ow * gm.stride - gm.pad;
+                t_xb[tid] = ((long)n * gm.G * gm.Cin + (long)g * gm.Cin) * HW;
+                t_dyb[tid] = ((long)n * gm.G * gm.Cout
+                              + (long)g * gm.Cout) * OHW + hw;
+            } else {
+                t_ihb[tid] = 1 << 28;
+                t_iwb[tid] = 1 << 28;
+                t_xb[tid] = 0;
+                t_dyb[tid] = -1;
+            }
+        }
+        __syncthreads();
         for (int e = tid; e < BM * BK; e += 256) {
             const int pp = e & (BK - 1), mm = e >> 5;
-            const int m = m0 + mm, p = p0 + pp;
+            const int m = m0 + mm;
             float v = 0.f;
-            if (m < M && p < pend) {
-                const int n = p / OHW, hw = p - n * OHW;
-                v = ld_f32(dy + ((long)n * gm.G * gm.Cout
-                                 + (long)g * gm.Cout + m) * OHW + hw);
-            }
+            if (m < M && t_dyb[pp] >= 0)
+                v = ld_f32(dy + t_dyb[pp] + (long)m * OHW);
             a_lds[mm][pp] = (T)v;
         }
         for (int e = tid; e < BK * BP; e += 256) {
             const int pp = e & (BK - 1), kk = e >> 5;
-            const int k = k0 + kk, p = p0 + pp;
+            const int k = k0 + kk;
             float v = 0.f;
-            if (k < K && p < pend) {
-                const int cin = k / kk2, r = k - cin * kk2;
-                const int kh = r / gm.khw, kw = r - kh * gm.khw;
-                const int n = p / OHW, hw = p - n * OHW;
-                const int oh = hw / gm.OW, ow = hw - oh * gm.OW;
-                const int ih = oh * gm.stride + kh - gm.pad;
-                const int iw = ow * gm.stride + kw - gm.pad;
+            if (k < K) {
+                const int ih = t_ihb[pp] + t_kkh[kk];
+                const int iw = t_iwb[pp] + t_kkw[kk];
                 if (ih >= 0 && ih < gm.H && iw >= 0 && iw < gm.W)
-                    v = ld_f32(x + ((long)n * gm.G * gm.Cin
-                                    + (long)g * gm.Cin + cin) * HW
+                    v = ld_f32(x + t_xb[pp] + (long)t_kcin[kk] * HW
                                + ih * gm.W + iw);
             }
             b_lds[kk][pp] = (T)v;
@@ -402,8 +468,17 @@ at::Tensor conv_fwd(at::Tensor x, at::Tensor w, at::Tensor bias,
     gm.OW = (gm.W + 2 * gm.pad - gm.khw) / gm.stride + 1;
     auto y = at::empty({gm.N, gm.G * gm.Cout, gm.OH, gm.OW}, x.options());
     const int M = gm.Cout, P = gm.N * gm.OH * gm.OW;
-    dim3 grid((M + BM - 1) / BM, (P + BP - 1) / BP, gm.G);
+    const int K = gm.Cin * gm.khw * gm.khw;
+    const int kchunks = (K + BK - 1) / BK;
+    const int tiles = ((M + BM - 1) / BM) * ((P + BP - 1) / BP) * gm.G;
+    int splitk = 1;
+    while (tiles * splitk < 512 && splitk * 2 <= kchunks / 2) splitk *= 2;
+    dim3 grid((M + BM - 1) / BM, (P + BP - 1) / BP, gm.G * splitk);
     auto stream = at::hip::getCurrentHIPStream();
+    at::Tensor partial;
+    if (splitk > 1)
+        partial = at::empty({(long)splitk * y.numel()},
+                            x.options().dtype(at::kFloat));
     DISPATCH_CONV_FT(x.scalar_type(), {
         hipLaunchKernelGGL(conv_fwd_kernel<scalar_t>, grid, dim3(256), 0,
                            stream, (const scalar_t*)x.data_ptr(),
@@ -412,7 +487,23 @@ at::Tensor conv_fwd(at::Tensor x, at::Tensor w, at::Tensor bias,
                            residual.defined()
                                ? (const scalar_t*)residual.data_ptr()
                                : nullptr,
-                           (scalar_t*)y.data_ptr(), gm);
+                           (scalar_t*)y.data_ptr(),
+                           splitk > 1 ? partial.data_ptr<float>() : nullptr,
+                           gm, splitk);
+        if (splitk > 1) {
+            const long total = y.numel();
+            const int blocks = (int)((total + 255) / 256);
+            hipLaunchKernelGGL(conv_out_reduce_kernel<scalar_t>, dim3(blocks),
+                               dim3(256), 0, stream,
+                               partial.data_ptr<float>(),
+                               bias.defined() ? bias.data_ptr<float>()
+                                              : nullptr,
+                               residual.defined()
+                                   ? (const scalar_t*)residual.data_ptr()
+                                   : nullptr,
+                               (scalar_t*)y.data_ptr(), total,
+                               (long)gm.OH * gm.OW, gm.G * gm.Cout, splitk);
+        }
     });
     return y;
 }
@@ -436,12 +527,32 @@ at::Tensor conv_bwd_data(at::Tensor dy, at::Tensor w, int64_t groups,
     gm.OW = dy.size(3);
     auto dx = at::empty({gm.N, gm.G * gm.Cin, gm.H, gm.W}, dy.options());
     const int Q = gm.N * gm.H * gm.W;
-    dim3 grid((gm.Cin + BM - 1) / BM, (Q + BP - 1) / BP, gm.G);
+    const int J = gm.Cout * gm.khw * gm.khw;
+    const int jchunks = (J + BK - 1) / BK;
+    const int tiles = ((gm.Cin + BM - 1) / BM) * ((Q + BP - 1) / BP) * gm.G;
+    int splitk = 1;
+    while (tiles * splitk < 512 && splitk * 2 <= jchunks / 2) splitk *= 2;
+    dim3 grid((gm.Cin + BM - 1) / BM, (Q + BP - 1) / BP, gm.G * splitk);
     auto stream = at::hip::getCurrentHIPStream();
+    at::Tensor partial;
+    if (splitk > 1)
+        partial = at::empty({(long)splitk * dx.numel()},
+                            dy.options().dtype(at::kFloat));
     DISPATCH_CONV_FT(dy.scalar_type(), {
         hipLaunchKernelGGL(conv_bwd_data_kernel<scalar_t>, grid, dim3(256), 0,
                            stream, (const scalar_t*)dyc.data_ptr(),
-                           w.data_ptr<float>(), (scalar_t*)dx.data_ptr(), gm);
+                           w.data_ptr<float>(), (scalar_t*)dx.data_ptr(),
+                           splitk > 1 ? partial.data_ptr<float>() : nullptr,
+                           gm, splitk);
+        if (splitk > 1) {
+            const long total = dx.numel();
+            const int blocks = (int)((total + 255) / 256);
+            hipLaunchKernelGGL(conv_out_reduce_kernel<scalar_t>, dim3(blocks),
+                               dim3(256), 0, stream,
+                               partial.data_ptr<float>(), nullptr, nullptr,
+                               (scalar_t*)dx.data_ptr(), total, 1L, 1,
+                               splitk);
+        }
     });
     return dx;
 }

@@ -155,3 +155,33 @@ class _GroupedConv(torch.autograd.Function):
 
 def grouped_conv(x, weight, bias, groups, stride, pad, residual=None):
     return _GroupedConv.apply(x, weight, bias, residual, groups, stride, pad)
+
+
+class _FusedHead(torch.autograd.Function):
+    """AdaptiveAvgPool(1) + flatten + per-client Linear in one kernel each
+    way (ops/csrc/head.hip).  scores come back fp32 (they feed masked-CE)."""
+
+    @staticmethod
+    def forward(ctx, feat, weight, bias, R):
+        ext = require_native()
+        feat = feat.contiguous()
+        scores, pooled = ext.head_fwd(feat, weight,
+                                      bias if bias is not None
+                                      else torch.Tensor(), R)
+        ctx.save_for_backward(pooled, weight)
+        ctx.meta = (R, feat.size(2), feat.size(3),
+                    feat.dtype == torch.bfloat16, bias is not None)
+        return scores
+
+    @staticmethod
+    def backward(ctx, dscores):
+        ext = require_native()
+        pooled, weight = ctx.saved_tensors
+        R, H, W, bf16_feat, has_bias = ctx.meta
+        dw, db, dfeat = ext.head_bwd(dscores.contiguous().float(), pooled,
+                                     weight, R, H, W, bf16_feat, has_bias)
+        return dfeat, dw, (db if has_bias else None), None
+
+
+def fused_head(feat, weight, bias, R):
+    return _FusedHead.apply(feat, weight, bias, R)

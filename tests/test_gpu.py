@@ -373,3 +373,27 @@ def test_mfma_conv_residual_fusion():
     ref.backward(g)
     assert (res.grad - r2.grad).abs().max().item() < 1e-6
     assert (x.grad - x2.grad).abs().max().item() < 0.05
+
+
+@needs_gpu
+def test_fused_head_matches_torch():
+    import torch.nn.functional as F
+    from heterofl_amd.ops.fused import fused_head
+    torch.manual_seed(0)
+    N, R, C, H, J = 10, 5, 512, 4, 10
+    feat = torch.randn(N, R * C, H, H, device='cuda', requires_grad=True)
+    w = (torch.randn(R, J, C, device='cuda') * 0.05).requires_grad_()
+    b = torch.randn(R, J, device='cuda', requires_grad=True)
+    scores = fused_head(feat, w, b, R)
+    f2 = feat.detach().clone().requires_grad_(True)
+    w2 = w.detach().clone().requires_grad_(True)
+    b2 = b.detach().clone().requires_grad_(True)
+    pooled = F.adaptive_avg_pool2d(f2, 1).view(N, R, C)
+    ref = torch.einsum('nri,roi->nro', pooled, w2) + b2
+    assert (scores - ref).abs().max().item() < 1e-4
+    g = torch.randn_like(scores)
+    scores.backward(g)
+    ref.backward(g)
+    assert (feat.grad - f2.grad).abs().max().item() < 1e-5
+    assert (w.grad - w2.grad).abs().max().item() < 1e-4
+    assert (b.grad - b2.grad).abs().max().item() < 1e-5
