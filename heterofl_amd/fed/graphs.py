@@ -53,14 +53,12 @@ class GraphedGroupStep:
         self.graph = None
         self._arange = torch.arange(batch, device=device)
         self._native = (device.type == 'cuda' and native_ops.use_native(device))
+        self._capturing = False
+        self._captured_grads = None
         if self._native:
             native_ops.require_native()
-            for p in self.params:
-                p.grad = torch.zeros_like(p)
-            from ..ops.fused import FusedClipSGD
-            self.opt = FusedClipSGD(self.params,
-                                    [p.grad for p in self.params],
-                                    self.bufs, R, device)
+            from ..ops.fused import GraphClipSGD
+            self.opt = GraphClipSGD(self.params, self.bufs, R, device)
 
     def _input_shape(self):
         m = self.model
@@ -74,14 +72,16 @@ class GraphedGroupStep:
         xb = self.x_all.index_select(0, idx)
         yb = self.y_all.index_select(0, idx)
         if self._native:
-            with torch.no_grad():
-                torch._foreach_zero_([p.grad for p in self.params])
             with torch.autocast('cuda', torch.bfloat16, enabled=self.amp):
                 scores = self.model(xb)
             losses = batched_masked_ce(scores.float(), yb, self.masks,
                                        metrics=self.metrics)
-            losses.sum().backward()
-            self.opt.step(1.0, self.lr, self.momentum, self.weight_decay)
+            raw = torch.autograd.grad(losses.sum(), self.params)
+            grads = [t if t.is_contiguous() else t.contiguous() for t in raw]
+            self._captured_grads = grads
+            if not self._capturing:
+                self.opt.bind(grads)  # eager warmup: per-iteration pointers
+            self.opt.launch(1.0, self.lr, self.momentum, self.weight_decay)
             with torch.no_grad():
                 self.counter += 1
             return
@@ -130,8 +130,13 @@ class GraphedGroupStep:
         torch.cuda.current_stream().wait_stream(s)
         self.counter.zero_()
         self.graph = torch.cuda.CUDAGraph()
+        self._capturing = True
         with torch.cuda.graph(self.graph):
             self._one_step()
+        self._capturing = False
+        if self._native:
+            # bind the captured grad-pool addresses (stable across replays)
+            self.opt.bind(self._captured_grads)
         with torch.no_grad():
             for p, sp in zip(self.params, saved_p):
                 p.copy_(sp)

@@ -20,6 +20,10 @@ std::vector<at::Tensor> build_chunk_table(std::vector<at::Tensor> grads,
                                           std::vector<at::Tensor> params,
                                           std::vector<at::Tensor> bufs,
                                           int64_t R, int64_t chunk_elems);
+void fill_chunk_table(at::Tensor blob, std::vector<at::Tensor> grads,
+                      std::vector<at::Tensor> params,
+                      std::vector<at::Tensor> bufs, int64_t R,
+                      int64_t chunk_elems);
 void clip_sgd_step(at::Tensor table_blob, int64_t n_chunks,
                    at::Tensor chunk_client, at::Tensor partials,
                    at::Tensor normsq, double max_norm, double lr,
@@ -48,6 +52,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("build_chunk_table", &build_chunk_table,
           "build clip+SGD chunk table");
     m.def("clip_sgd_step", &clip_sgd_step, "fused per-client clip+SGD step");
+    m.def("fill_chunk_table", &fill_chunk_table,
+          "fill a preallocated chunk table post-capture");
     m.def("conv_fwd", &conv_fwd, "MFMA implicit-GEMM grouped conv forward");
     m.def("conv_bwd_data", &conv_bwd_data, "MFMA grouped conv backward-data");
     m.def("conv_bwd_weight", &conv_bwd_weight,

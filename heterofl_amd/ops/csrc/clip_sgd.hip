@@ -123,6 +123,39 @@ std::vector<at::Tensor> build_chunk_table(std::vector<at::Tensor> grads,
     return {blob, n, cl};
 }
 
+// Fill an already-allocated device chunk table from (grads, params, bufs)
+// pointer lists.  Used by the hipGraph path: the kernels are RECORDED
+// against the blob's device address during capture, and the contents are
+// written here once afterwards (grad pool addresses are stable per graph).
+void fill_chunk_table(at::Tensor blob, std::vector<at::Tensor> grads,
+                      std::vector<at::Tensor> params,
+                      std::vector<at::Tensor> bufs, int64_t R,
+                      int64_t chunk_elems) {
+    std::vector<Chunk> host;
+    for (size_t t = 0; t < grads.size(); ++t) {
+        const long numel = grads[t].numel();
+        const long per = numel / R;
+        const float* g = grads[t].data_ptr<float>();
+        float* p = params[t].data_ptr<float>();
+        float* b = bufs[t].data_ptr<float>();
+        for (long r = 0; r < R; ++r)
+            for (long off = 0; off < per; off += chunk_elems) {
+                Chunk ck;
+                ck.grad = g + r * per + off;
+                ck.param = p + r * per + off;
+                ck.buf = b + r * per + off;
+                ck.len = (int)std::min((long)chunk_elems, per - off);
+                ck.client = (int)r;
+                host.push_back(ck);
+            }
+    }
+    TORCH_CHECK((long)(host.size() * sizeof(Chunk)) == blob.numel(),
+                "chunk table size changed");
+    auto src = at::from_blob(host.data(), {(long)(host.size() * sizeof(Chunk))},
+                             at::TensorOptions().dtype(at::kByte)).clone();
+    blob.copy_(src, /*non_blocking=*/false);
+}
+
 void clip_sgd_step(at::Tensor table_blob, int64_t n_chunks,
                    at::Tensor chunk_client, at::Tensor partials,
                    at::Tensor normsq, double max_norm, double lr,

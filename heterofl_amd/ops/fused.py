@@ -185,3 +185,44 @@ class _FusedHead(torch.autograd.Function):
 
 def fused_head(feat, weight, bias, R):
     return _FusedHead.apply(feat, weight, bias, R)
+
+
+class GraphClipSGD:
+    """hipGraph-capture variant of FusedClipSGD: the chunk LAYOUT is fixed by
+    parameter shapes, so the device table is preallocated and the kernels are
+    recorded against it; the grad POINTERS (graph-pool addresses captured by
+    autograd.grad) are filled in afterwards with fill_chunk_table."""
+
+    CHUNK = 65536
+    CHUNK_BYTES = 32  # sizeof(Chunk): 3 pointers + 2 ints
+
+    def __init__(self, params, bufs, R, device):
+        require_native()
+        self.params = list(params)
+        self.bufs = list(bufs)
+        self.R = R
+        clients = []
+        for p in self.params:
+            per = p.numel() // R
+            nch = (per + self.CHUNK - 1) // self.CHUNK
+            for r in range(R):
+                clients.extend([r] * nch)
+        self.n_chunks = len(clients)
+        self.table = torch.zeros(self.n_chunks * self.CHUNK_BYTES,
+                                 dtype=torch.uint8, device=device)
+        self.chunk_client = torch.tensor(clients, dtype=torch.int32).to(device)
+        self.partials = torch.zeros(self.n_chunks, dtype=torch.float32,
+                                    device=device)
+        self.normsq = torch.zeros(R, dtype=torch.float32, device=device)
+
+    def bind(self, grads):
+        """Write (grad, param, buf) pointers into the device table."""
+        ext = require_native()
+        ext.fill_chunk_table(self.table, list(grads), self.params, self.bufs,
+                             self.R, self.CHUNK)
+
+    def launch(self, max_norm, lr, momentum, weight_decay):
+        ext = require_native()
+        ext.clip_sgd_step(self.table, self.n_chunks, self.chunk_client,
+                          self.partials, self.normsq, max_norm, lr, momentum,
+                          weight_decay)
