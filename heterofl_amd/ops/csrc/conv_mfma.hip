@@ -578,7 +578,7 @@ at::Tensor conv_bwd_weight(at::Tensor dy, at::Tensor x, int64_t groups,
     const int mk_tiles = ((gm.Cout + BM - 1) / BM) * ((K + BP - 1) / BP)
                          * gm.G;
     int splitp = 1;
-    while (mk_tiles * splitp < 256 && splitp * BK * 4 < P) splitp *= 2;
+    while (mk_tiles * splitp < 512 && splitp * BK * 4 < P) splitp *= 2;
     auto dw = at::empty({(long)gm.G * gm.Cout, gm.Cin, gm.khw, gm.khw},
                         x.options().dtype(at::kFloat));
     auto stream = at::hip::getCurrentHIPStream();
