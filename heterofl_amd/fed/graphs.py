@@ -9,11 +9,12 @@ round's pre-staged epoch buffer via the counter, so an entire local-training
 run (5 epochs x 50 steps) is 250 back-to-back graph replays with no host
 work in between.
 
-With the native extension the step is: fused norm+relu kernels in the model,
-fused masked-CE (which also accumulates device-side metrics), backward into
-preallocated .grad buffers, then the two-kernel fused per-client
-clip(1.0)+momentum-SGD (ops/csrc/clip_sgd.hip) over a chunk table built once
-at capture.  Without it (CPU/debug) the pure-torch step runs.
+With the native extension the step is: MFMA grouped-conv + fused norm+relu
++ fused head kernels in the model, fused masked-CE (which also accumulates
+device-side metrics), autograd.grad into graph-pool buffers, then the fused
+per-client clip(1.0)+momentum-SGD (ops/csrc/clip_sgd.hip) over a chunk table
+whose pointers are bound after capture (GraphClipSGD).  Without the
+extension (CPU/debug) the pure-torch step runs.
 
 Graphs are cached per (rate, R, batch, lr, n_steps_capacity); weights,
 momentum and the label masks live in stable buffers the graph reads, so a
