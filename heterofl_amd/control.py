@@ -88,7 +88,7 @@ def _dataset_hyperparameters(cfg):
     (reference: src/utils.py:150-212)."""
     name = cfg['data_name']
     split = cfg['data_split_mode']
-    if name in ('MNIST', 'FashionMNIST'):
+    if name in ('MNIST', 'FashionMNIST', 'EMNIST'):
         cfg['data_shape'] = [1, 28, 28]
         cfg.update(optimizer_name='SGD', lr=1e-2, momentum=0.9,
                    weight_decay=5e-4, scheduler_name='MultiStepLR', factor=0.1)

@@ -34,6 +34,7 @@ _NORM = {
 
 _REAL_SIZE = {  # (train, test) sample counts of the real datasets
     'MNIST': (60000, 10000), 'FashionMNIST': (60000, 10000),
+    'EMNIST': (112800, 18800),  # balanced split
     'CIFAR10': (50000, 10000), 'CIFAR100': (50000, 10000),
 }
 
@@ -178,7 +179,7 @@ def _load_wikitext_raw(root, split):
 # --------------------------------------------------------------- synthetic
 def _synthetic_vision(data_name, n, classes_size, seed):
     g = torch.Generator().manual_seed(seed)
-    if data_name in ('MNIST', 'FashionMNIST'):
+    if data_name in ('MNIST', 'FashionMNIST', 'EMNIST'):
         img = torch.randint(0, 256, (n, 28, 28), dtype=torch.uint8, generator=g)
     else:
         img = torch.randint(0, 256, (n, 32, 32, 3), dtype=torch.uint8, generator=g)
@@ -196,12 +197,12 @@ def fetch_dataset(data_name, subset='label', synthetic=False,
     root (default ./data/<name>)."""
     root = root or os.path.join('.', 'data', data_name)
     dataset = {}
-    if data_name in ('MNIST', 'FashionMNIST', 'CIFAR10', 'CIFAR100'):
-        classes_size = 100 if data_name == 'CIFAR100' else 10
+    if data_name in ('MNIST', 'FashionMNIST', 'EMNIST', 'CIFAR10', 'CIFAR100'):
+        classes_size = {'CIFAR100': 100, 'EMNIST': 47}.get(data_name, 10)
         for split in ('train', 'test'):
             raw = None
             if not synthetic:
-                if data_name in ('MNIST', 'FashionMNIST'):
+                if data_name in ('MNIST', 'FashionMNIST', 'EMNIST'):
                     raw = _load_mnist_raw(root, split)
                 else:
                     raw = _load_cifar_raw(root, split, data_name)

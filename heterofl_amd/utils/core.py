@@ -68,7 +68,8 @@ def batchify(dataset, batch_size):
 def process_dataset(dataset, cfg):
     """Record classes_size / vocab on cfg and batchify LM streams
     (reference: src/utils.py:100-110)."""
-    if cfg['data_name'] in ('MNIST', 'FashionMNIST', 'CIFAR10', 'CIFAR100'):
+    if cfg['data_name'] in ('MNIST', 'FashionMNIST', 'EMNIST', 'CIFAR10',
+                            'CIFAR100'):
         cfg['classes_size'] = dataset['train'].classes_size
     elif cfg['data_name'] in ('PennTreebank', 'WikiText2', 'WikiText103'):
         cfg['vocab'] = dataset['train'].vocab
