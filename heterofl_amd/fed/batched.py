@@ -569,6 +569,8 @@ class BatchedClientTrainer:
                 ys.append(lab[perm])
             x_all = torch.stack(xs, 1)   # (n, R, C, H, W)
             x_all = x_all.reshape(n, -1, x_all.size(-2), x_all.size(-1))
+            if self._amp:
+                x_all = x_all.to(torch.bfloat16)
             y_all = torch.stack(ys, 1)   # (n, R)
             off = 0
             for bs in sched:
@@ -634,6 +636,8 @@ class BatchedClientTrainer:
                 xs_ep.append(x_all.reshape(n, -1, x_all.size(-2), x_all.size(-1)))
                 ys_ep.append(torch.stack(ys, 1))
             x_cat = torch.cat(xs_ep, 0)
+            if self._amp:
+                x_cat = x_cat.to(torch.bfloat16)
             y_cat = torch.cat(ys_ep, 0)
         with _phase_timer('2d.replay'):
             gs.run_epochs(x_cat, y_cat, len(sched))

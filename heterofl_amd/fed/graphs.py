@@ -45,7 +45,10 @@ class GraphedGroupStep:
         self.bufs = [torch.zeros_like(p) for p in self.params]
         self.capacity = capacity  # max samples in the epoch buffer
         shape = self._input_shape()
-        self.x_all = torch.zeros(capacity, *shape, device=device)
+        # bf16 local-training: inputs staged in bf16 so the native kernels
+        # (which bypass torch autocast) really compute in bf16
+        xdt = torch.bfloat16 if amp else torch.float32
+        self.x_all = torch.zeros(capacity, *shape, device=device, dtype=xdt)
         self.y_all = torch.zeros(capacity, R, dtype=torch.long, device=device)
         self.masks = torch.ones(R, classes, device=device)
         self.counter = torch.zeros(1, dtype=torch.long, device=device)
