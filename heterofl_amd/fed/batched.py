@@ -488,10 +488,50 @@ class BatchedClientTrainer:
             sched = tuple([B] * (n // B) + ([n % B] if n % B else []))
             groups.setdefault((model_rate[u], sched), []).append(m)
         out = []
+        # graph-eligible groups are prepared on the default stream, then
+        # their replay trains run CONCURRENTLY on per-group HIP streams (the
+        # rate-e group's tiny latency-bound steps hide under rate-a's)
+        graphed = []
         for (rate, sched), slots in groups.items():
-            out.extend(self._train_group(rate, sched, slots, user_idx,
-                                         local_parameters, label_split, lr,
-                                         logger))
+            use_graph = (self.device.type == 'cuda'
+                         and cfg.get('hip_graphs', True)
+                         and len(set(sched)) == 1)
+            if use_graph:
+                graphed.append((rate, sched, slots))
+            else:
+                out.extend(self._train_group(rate, sched, slots, user_idx,
+                                             local_parameters, label_split,
+                                             lr, logger))
+        if graphed:
+            preps = [self._prepare_graphed(rate, sched, slots, user_idx,
+                                           [local_parameters[m]
+                                            for m in slots],
+                                           label_split, lr)
+                     for rate, sched, slots in graphed]
+            from .runner import _phase_timer
+            with _phase_timer('2d.replay'):
+                if len(preps) > 1:
+                    if not hasattr(self, '_streams') or                             len(self._streams) < len(preps):
+                        self._streams = [torch.cuda.Stream()
+                                         for _ in range(len(preps))]
+                    cur = torch.cuda.current_stream()
+                    events = []
+                    for (gs, x_cat, y_cat, nspe), st in zip(preps,
+                                                            self._streams):
+                        st.wait_stream(cur)
+                        with torch.cuda.stream(st):
+                            gs.run_epochs(x_cat, y_cat, nspe)
+                        ev = torch.cuda.Event()
+                        ev.record(st)
+                        events.append(ev)
+                    for ev in events:
+                        cur.wait_event(ev)
+                else:
+                    gs, x_cat, y_cat, nspe = preps[0]
+                    gs.run_epochs(x_cat, y_cat, nspe)
+            for (rate, sched, slots), (gs, _, _, _) in zip(graphed, preps):
+                out.extend(self._finish_graphed(
+                    gs, slots, [local_parameters[m] for m in slots], logger))
         return out
 
     def _graph_step(self, rate, sched, R, lr):
@@ -603,16 +643,18 @@ class BatchedClientTrainer:
         cpu_or_dev = [{k: v for k, v in st.items()} for st in states]
         return list(zip(slots, cpu_or_dev))
 
-    def _train_group_graphed(self, rate, sched, slots, user_idx, locals_list,
-                             label_split, lr, logger=None):
-        """hipGraph path: pack -> stage all epochs' augmented data -> replay
-        the captured step n_steps times -> unpack + device-side metrics."""
+    def _prepare_graphed(self, rate, sched, slots, user_idx, locals_list,
+                         label_split, lr):
+        """Default-stream phase of the hipGraph path: build/capture, pack,
+        masks, stage all epochs' augmented data.  Returns (gs, x, y, nspe)."""
         from .runner import _phase_timer
         cfg = self.cfg
         R = len(slots)
         device = self.device
         with _phase_timer('2a.graph_build'):
             gs = self._graph_step(rate, sched, R, lr)
+            if gs.graph is None:
+                gs.ensure_captured()
         with _phase_timer('2b.pack'):
             pack_states(gs.model, locals_list)
         masks = None
@@ -639,8 +681,11 @@ class BatchedClientTrainer:
             if self._amp:
                 x_cat = x_cat.to(torch.bfloat16)
             y_cat = torch.cat(ys_ep, 0)
-        with _phase_timer('2d.replay'):
-            gs.run_epochs(x_cat, y_cat, len(sched))
+        return gs, x_cat, y_cat, len(sched)
+
+    def _finish_graphed(self, gs, slots, locals_list, logger=None):
+        from .runner import _phase_timer
+        R = len(slots)
         if logger is not None:
             m = gs.metrics.detach().cpu()
             for i in range(R):

@@ -175,3 +175,7 @@ class GraphedGroupStep:
             self.capture()
         for _ in range(n_steps):
             self.graph.replay()
+
+    def ensure_captured(self):
+        if self.graph is None:
+            self.capture()

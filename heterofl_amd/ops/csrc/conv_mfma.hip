@@ -66,6 +66,20 @@ __device__ __forceinline__ f32x4 mfma_tile<float>(const float* a_row,
     return acc;
 }
 
+
+// store 8 fp32 values as one contiguous 16B (bf16) / 32B (f32) LDS write
+__device__ __forceinline__ void st8_lds(__hip_bfloat16* dst, const float* v) {
+    bf16x8 t;
+#pragma unroll
+    for (int j = 0; j < 8; ++j) t[j] = (__bf16)v[j];
+    *(bf16x8*)dst = t;
+}
+__device__ __forceinline__ void st8_lds(float* dst, const float* v) {
+    typedef __attribute__((ext_vector_type(4))) float f4;
+    *(f4*)dst = *(const f4*)v;
+    *(f4*)(dst + 4) = *(const f4*)(v + 4);
+}
+
 struct ConvGeom {
     int G;
     int N, H, W;
@@ -131,26 +145,35 @@ conv_fwd_kernel(const T* __restrict__ x, const float* __restrict__ w,
 
     f32x4 acc[2][2] = {};
     for (int k0 = ks; k0 < ke; k0 += BK) {
-        for (int e = tid; e < BM * BK; e += 256) {
-            const int kk = e & (BK - 1), mm = e >> 5;
-            const int m = m0 + mm, k = k0 + kk;
-            float v = 0.f;
-            if (m < M && k < K) v = w[(long)(g * gm.Cout + m) * K + k];
-            a_lds[mm][kk] = (T)v;
-        }
-        for (int e = tid; e < BK * BP; e += 256) {
-            const int pp = e & (BP - 1), kk = e >> 6;
-            const int k = k0 + kk;
-            float v = 0.f;
-            if (k < K) {
-                const int cin = k / kk2, r = k - cin * kk2;
-                const int kh = r / gm.khw, kw = r - kh * gm.khw;
-                const int ih = t_ihb[pp] + kh, iw = t_iwb[pp] + kw;
-                if (ih >= 0 && ih < gm.H && iw >= 0 && iw < gm.W)
-                    v = ld_f32(x + t_xbase[pp] + (long)cin * HW
-                               + ih * gm.W + iw);
+        {   // A tile: thread owns 8 contiguous k of one row -> one b128
+            const int mm = tid >> 2, kkb = (tid & 3) * 8;
+            const int m = m0 + mm;
+            float v[8];
+            const float* wrow = w + (long)(g * gm.Cout + m) * K + k0 + kkb;
+#pragma unroll
+            for (int j = 0; j < 8; ++j) {
+                const int k = k0 + kkb + j;
+                v[j] = (m < M && k < K) ? wrow[j] : 0.f;
             }
-            b_lds[pp][kk] = (T)v;
+            st8_lds(&a_lds[mm][kkb], v);
+        }
+        {   // B tile: thread owns 8 contiguous k of one pixel -> one b128
+            const int pp = tid >> 2, kkb = (tid & 3) * 8;
+            float v[8];
+#pragma unroll
+            for (int j = 0; j < 8; ++j) {
+                const int k = k0 + kkb + j;
+                v[j] = 0.f;
+                if (k < K) {
+                    const int cin = k / kk2, r = k - cin * kk2;
+                    const int kh = r / gm.khw, kw = r - kh * gm.khw;
+                    const int ih = t_ihb[pp] + kh, iw = t_iwb[pp] + kw;
+                    if (ih >= 0 && ih < gm.H && iw >= 0 && iw < gm.W)
+                        v[j] = ld_f32(x + t_xbase[pp] + (long)cin * HW
+                                      + ih * gm.W + iw);
+                }
+            }
+            st8_lds(&b_lds[pp][kkb], v);
         }
         __syncthreads();
 #pragma unroll
@@ -254,38 +277,49 @@ conv_bwd_data_kernel(const T* __restrict__ dy, const float* __restrict__ w,
     const int je = min(J, js + njc * BK);
     f32x4 acc[2][2] = {};
     for (int j0 = js; j0 < je; j0 += BK) {
-        for (int e = tid; e < BM * BK; e += 256) {
-            const int jj = e & (BK - 1), cc = e >> 5;
-            const int c = c0 + cc, j = j0 + jj;
-            float v = 0.f;
-            if (c < gm.Cin && j < J) {
-                const int cout = j / kk2, r = j - cout * kk2;
-                v = w[(long)(g * gm.Cout + cout) * K + c * kk2 + r];
+        {
+            const int cc = tid >> 2, jjb = (tid & 3) * 8;
+            const int c = c0 + cc;
+            float v[8];
+#pragma unroll
+            for (int j8 = 0; j8 < 8; ++j8) {
+                const int j = j0 + jjb + j8;
+                v[j8] = 0.f;
+                if (c < gm.Cin && j < J) {
+                    const int cout = j / kk2, r = j - cout * kk2;
+                    v[j8] = w[(long)(g * gm.Cout + cout) * K + c * kk2 + r];
+                }
             }
-            a_lds[cc][jj] = (T)v;
+            st8_lds(&a_lds[cc][jjb], v);
         }
-        for (int e = tid; e < BK * BP; e += 256) {
-            const int qq = e & (BP - 1), jj = e >> 6;
-            const int j = j0 + jj;
-            float v = 0.f;
-            if (j < J) {
-                const int cout = j / kk2, r = j - cout * kk2;
-                const int kh = r / gm.khw, kw = r - kh * gm.khw;
-                const int ohs = t_oh[qq] - kh, ows = t_ow[qq] - kw;
-                if (ohs >= 0 && ows >= 0) {
-                    if (gm.stride == 1) {
-                        if (ohs < gm.OH && ows < gm.OW)
-                            v = ld_f32(dy + t_dybase[qq] + (long)cout * OHW
-                                       + ohs * gm.OW + ows);
-                    } else if ((ohs & 1) == 0 && (ows & 1) == 0) {
-                        const int oh = ohs >> 1, ow = ows >> 1;
-                        if (oh < gm.OH && ow < gm.OW)
-                            v = ld_f32(dy + t_dybase[qq] + (long)cout * OHW
-                                       + oh * gm.OW + ow);
+        {
+            const int qq = tid >> 2, jjb = (tid & 3) * 8;
+            float v[8];
+#pragma unroll
+            for (int j8 = 0; j8 < 8; ++j8) {
+                const int j = j0 + jjb + j8;
+                v[j8] = 0.f;
+                if (j < J) {
+                    const int cout = j / kk2, r = j - cout * kk2;
+                    const int kh = r / gm.khw, kw = r - kh * gm.khw;
+                    const int ohs = t_oh[qq] - kh, ows = t_ow[qq] - kw;
+                    if (ohs >= 0 && ows >= 0) {
+                        if (gm.stride == 1) {
+                            if (ohs < gm.OH && ows < gm.OW)
+                                v[j8] = ld_f32(dy + t_dybase[qq]
+                                               + (long)cout * OHW
+                                               + ohs * gm.OW + ows);
+                        } else if ((ohs & 1) == 0 && (ows & 1) == 0) {
+                            const int oh = ohs >> 1, ow = ows >> 1;
+                            if (oh < gm.OH && ow < gm.OW)
+                                v[j8] = ld_f32(dy + t_dybase[qq]
+                                               + (long)cout * OHW
+                                               + oh * gm.OW + ow);
+                        }
                     }
                 }
             }
-            b_lds[qq][jj] = (T)v;
+            st8_lds(&b_lds[qq][jjb], v);
         }
         __syncthreads();
 #pragma unroll
@@ -385,26 +419,35 @@ conv_bwd_weight_kernel(const T* __restrict__ dy, const T* __restrict__ x,
             }
         }
         __syncthreads();
-        for (int e = tid; e < BM * BK; e += 256) {
-            const int pp = e & (BK - 1), mm = e >> 5;
+        {   // a tile rows: 64 rows x 32 p; thread owns 8 contiguous p
+            const int mm = (tid >> 2) & 63, ppb = (tid & 3) * 8;
             const int m = m0 + mm;
-            float v = 0.f;
-            if (m < M && t_dyb[pp] >= 0)
-                v = ld_f32(dy + t_dyb[pp] + (long)m * OHW);
-            a_lds[mm][pp] = (T)v;
-        }
-        for (int e = tid; e < BK * BP; e += 256) {
-            const int pp = e & (BK - 1), kk = e >> 5;
-            const int k = k0 + kk;
-            float v = 0.f;
-            if (k < K) {
-                const int ih = t_ihb[pp] + t_kkh[kk];
-                const int iw = t_iwb[pp] + t_kkw[kk];
-                if (ih >= 0 && ih < gm.H && iw >= 0 && iw < gm.W)
-                    v = ld_f32(x + t_xb[pp] + (long)t_kcin[kk] * HW
-                               + ih * gm.W + iw);
+            float v[8];
+#pragma unroll
+            for (int j = 0; j < 8; ++j) {
+                const int pp = ppb + j;
+                v[j] = (m < M && t_dyb[pp] >= 0)
+                           ? ld_f32(dy + t_dyb[pp] + (long)m * OHW) : 0.f;
             }
-            b_lds[kk][pp] = (T)v;
+            st8_lds(&a_lds[mm][ppb], v);
+        }
+        {
+            const int kk = (tid >> 2) & 63, ppb = (tid & 3) * 8;
+            const int k = k0 + kk;
+            float v[8];
+#pragma unroll
+            for (int j = 0; j < 8; ++j) {
+                const int pp = ppb + j;
+                v[j] = 0.f;
+                if (k < K) {
+                    const int ih = t_ihb[pp] + t_kkh[kk];
+                    const int iw = t_iwb[pp] + t_kkw[kk];
+                    if (ih >= 0 && ih < gm.H && iw >= 0 && iw < gm.W)
+                        v[j] = ld_f32(x + t_xb[pp] + (long)t_kcin[kk] * HW
+                                      + ih * gm.W + iw);
+                }
+            }
+            st8_lds(&b_lds[kk][ppb], v);
         }
         __syncthreads();
 #pragma unroll
