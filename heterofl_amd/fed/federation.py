@@ -258,16 +258,21 @@ class Federation:
             return v.narrow(0, 0, out.n).clone()
         return v.index_select(0, out.idx.to(v.device)).clone()
 
-    def distribute(self, user_idx, resample=True, generator=None):
+    def distribute(self, user_idx, resample=True, generator=None,
+                   slots=None):
         """reference: src/fed.py:161-178 — resample rates, build index maps,
         and hand each client its parameter slices.  Set resample=False when
         the engine already resampled rates (e.g. with a shared seeded
-        generator for multi-rank determinism)."""
+        generator for multi-rank determinism).  ``slots`` restricts the
+        tensor slicing to this rank's clients (index maps are still built
+        for every slot — combine needs only the local ones, but they are
+        cheap); other slots get None."""
         if resample:
             self.make_model_rate(generator)
         param_idx = self.split_model(user_idx)
-        local_parameters = []
-        for m in range(len(user_idx)):
+        want = range(len(user_idx)) if slots is None else slots
+        local_parameters = [None] * len(user_idx)
+        for m in want:
             lp = OrderedDict()
             for k, v in self.global_parameters.items():
                 ptype = k.split('.')[-1]
@@ -275,7 +280,7 @@ class Federation:
                     lp[k] = self._slice_value(v, param_idx[m][k])
                 else:
                     lp[k] = v.clone()
-            local_parameters.append(lp)
+            local_parameters[m] = lp
         return local_parameters, param_idx
 
     # --------------------------------------------------------------- combine

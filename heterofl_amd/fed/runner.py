@@ -125,14 +125,14 @@ class FedRunner:
         else:
             user_idx = sample_active_users(cfg, epoch)
             resample = True
-        with _phase_timer('1.distribute'):
-            local_parameters, param_idx = self.federation.distribute(
-                user_idx, resample=resample)
-
         # shard clients across ranks (identity shard when single rank)
         rank, world = (0, 1) if self.dist_ctx is None else \
             (self.dist_ctx.rank, self.dist_ctx.world_size)
         my_clients = list(range(rank, len(user_idx), world))
+        with _phase_timer('1.distribute'):
+            local_parameters, param_idx = self.federation.distribute(
+                user_idx, resample=resample,
+                slots=my_clients if world > 1 else None)
 
         with _phase_timer('2.local_train'):
             trained = dict(self.trainer.train_clients(
