@@ -351,6 +351,148 @@ conv_bwd_data_kernel(const T* __restrict__ dy, const float* __restrict__ w,
         }
 }
 
+
+// ------------------------------------------- stride-2 bwd-data (parity)
+// For stride 2 only 1/4 of the (kh, kw) taps can contribute to a given
+// input pixel (ohs/ows parity).  Pixels are tiled by parity class
+// (ph, pw): each class is a dense (H/2, W/2) sub-image with a compressed
+// tap list, so no MFMA work is spent on dead taps.  grid.y = 4 * tiles.
+template <typename T>
+__global__ void __launch_bounds__(256)
+conv_bwd_data_s2_kernel(const T* __restrict__ dy, const float* __restrict__ w,
+                        T* __restrict__ dx, float* __restrict__ partial,
+                        ConvGeom gm, int splitk, int nyp) {
+    __shared__ T a_lds[BM][LDK];
+    __shared__ T b_lds[BP][LDK];
+    __shared__ int t_oh2[BP], t_ow2[BP];
+    __shared__ long t_dybase[BP], t_xbase[BP];
+    const int g = blockIdx.z % gm.G;
+    const int sp = blockIdx.z / gm.G;
+    const int c0 = blockIdx.x * BM;
+    const int cls = blockIdx.y / nyp;
+    const int q0 = (blockIdx.y - cls * nyp) * BP;
+    const int ph = cls >> 1, pw = cls & 1;
+    const int kk2 = gm.khw * gm.khw;
+    const int K = gm.Cin * kk2;
+    const int H2 = gm.H >> 1, W2 = gm.W >> 1;
+    const int HW2 = H2 * W2;
+    const int Qp = gm.N * HW2;
+    const int HW = gm.H * gm.W;
+    const int OHW = gm.OH * gm.OW;
+    const int tid = threadIdx.x;
+    const int l = tid & (WAVE - 1);
+    const int wave = tid / WAVE;
+    const int wm = (wave >> 1) * 32;
+    const int wp = (wave & 1) * 32;
+    // compressed tap lists for this parity class (khw <= 3)
+    int lkh[3], lkw[3], skh[3], skw[3];
+    int nkh = 0, nkw = 0;
+    for (int kh = 0; kh < gm.khw; ++kh)
+        if (((ph + gm.pad - kh) & 1) == 0) {
+            lkh[nkh] = kh;
+            skh[nkh] = (ph + gm.pad - kh) >> 1;
+            ++nkh;
+        }
+    for (int kw = 0; kw < gm.khw; ++kw)
+        if (((pw + gm.pad - kw) & 1) == 0) {
+            lkw[nkw] = kw;
+            skw[nkw] = (pw + gm.pad - kw) >> 1;
+            ++nkw;
+        }
+    const int tap2 = nkh * nkw;
+    const int J = gm.Cout * tap2;
+
+    if (tid < BP) {
+        const int q = q0 + tid;
+        if (q < Qp) {
+            const int n = q / HW2, hw2 = q - n * HW2;
+            const int ih2 = hw2 / W2, iw2 = hw2 - ih2 * W2;
+            t_oh2[tid] = ih2;
+            t_ow2[tid] = iw2;
+            t_dybase[tid] = ((long)n * gm.G * gm.Cout + (long)g * gm.Cout)
+                            * OHW;
+            t_xbase[tid] = ((long)n * gm.G * gm.Cin + (long)g * gm.Cin) * HW
+                           + (2 * ih2 + ph) * gm.W + (2 * iw2 + pw);
+        } else {
+            t_oh2[tid] = -(1 << 28);
+            t_ow2[tid] = -(1 << 28);
+            t_dybase[tid] = 0;
+            t_xbase[tid] = -1;
+        }
+    }
+    __syncthreads();
+
+    const int njc = ((J + BK - 1) / BK + splitk - 1) / splitk;
+    const int js = sp * njc * BK;
+    const int je = min(J, js + njc * BK);
+    f32x4 acc[2][2] = {};
+    for (int j0 = js; j0 < je; j0 += BK) {
+        {
+            const int cc = tid >> 2, jjb = (tid & 3) * 8;
+            const int c = c0 + cc;
+            float v[8];
+#pragma unroll
+            for (int j8 = 0; j8 < 8; ++j8) {
+                const int j = j0 + jjb + j8;
+                v[j8] = 0.f;
+                if (c < gm.Cin && j < J) {
+                    const int cout = j / tap2, rr = j - cout * tap2;
+                    const int kh = lkh[rr / nkw], kw = lkw[rr - (rr / nkw) * nkw];
+                    v[j8] = w[(long)(g * gm.Cout + cout) * K + c * kk2
+                              + kh * gm.khw + kw];
+                }
+            }
+            st8_lds(&a_lds[cc][jjb], v);
+        }
+        {
+            const int qq = tid >> 2, jjb = (tid & 3) * 8;
+            float v[8];
+#pragma unroll
+            for (int j8 = 0; j8 < 8; ++j8) {
+                const int j = j0 + jjb + j8;
+                v[j8] = 0.f;
+                if (j < J) {
+                    const int cout = j / tap2, rr = j - cout * tap2;
+                    const int a = rr / nkw, b = rr - a * nkw;
+                    const int oh = t_oh2[qq] + skh[a];
+                    const int ow = t_ow2[qq] + skw[b];
+                    if (oh >= 0 && oh < gm.OH && ow >= 0 && ow < gm.OW)
+                        v[j8] = ld_f32(dy + t_dybase[qq] + (long)cout * OHW
+                                       + oh * gm.OW + ow);
+                }
+            }
+            st8_lds(&b_lds[qq][jjb], v);
+        }
+        __syncthreads();
+#pragma unroll
+        for (int fm = 0; fm < 2; ++fm)
+#pragma unroll
+            for (int fp = 0; fp < 2; ++fp)
+                acc[fm][fp] = mfma_tile<T>(
+                    &a_lds[wm + fm * 16 + (l & 15)][0],
+                    &b_lds[wp + fp * 16 + (l & 15)][0], acc[fm][fp]);
+        __syncthreads();
+    }
+    const long slab = (long)sp * gm.N * gm.G * gm.Cin * HW;
+#pragma unroll
+    for (int fm = 0; fm < 2; ++fm)
+#pragma unroll
+        for (int fp = 0; fp < 2; ++fp) {
+            const int qq = wp + fp * 16 + (l & 15);
+            const long xb = t_xbase[qq];
+#pragma unroll
+            for (int r = 0; r < 4; ++r) {
+                const int c = c0 + wm + fm * 16 + (l >> 4) * 4 + r;
+                if (c < gm.Cin && xb >= 0) {
+                    if (splitk == 1)
+                        st_f32(dx + xb + (long)c * HW, acc[fm][fp][r]);
+                    else
+                        partial[slab + xb + (long)c * HW] = acc[fm][fp][r];
+                }
+            }
+        }
+}
+
 // ------------------------------------------------------ bwd-weight kernel
 // Writes per-split partials (splitp, G*Cout*K) when splitp > 1; a
 // deterministic reduce kernel sums them in fixed order (no atomics, so the
@@ -570,23 +712,39 @@ at::Tensor conv_bwd_data(at::Tensor dy, at::Tensor w, int64_t groups,
     gm.OW = dy.size(3);
     auto dx = at::empty({gm.N, gm.G * gm.Cin, gm.H, gm.W}, dy.options());
     const int Q = gm.N * gm.H * gm.W;
-    const int J = gm.Cout * gm.khw * gm.khw;
+    const bool s2 = (gm.stride == 2 && gm.H % 2 == 0 && gm.W % 2 == 0);
+    // parity path: 4 classes of Q/4 pixels, ~1/4 the taps each
+    const int Jfull = gm.Cout * gm.khw * gm.khw;
+    const int J = s2 ? (Jfull + 3) / 4 : Jfull;
     const int jchunks = (J + BK - 1) / BK;
-    const int tiles = ((gm.Cin + BM - 1) / BM) * ((Q + BP - 1) / BP) * gm.G;
+    const int qtile = s2 ? Q / 4 : Q;
+    const int nyp = (qtile + BP - 1) / BP;
+    const int tiles = ((gm.Cin + BM - 1) / BM) * nyp * (s2 ? 4 : 1) * gm.G;
     int splitk = 1;
     while (tiles * splitk < 512 && splitk * 2 <= jchunks / 2) splitk *= 2;
-    dim3 grid((gm.Cin + BM - 1) / BM, (Q + BP - 1) / BP, gm.G * splitk);
+    dim3 grid((gm.Cin + BM - 1) / BM, nyp * (s2 ? 4 : 1), gm.G * splitk);
     auto stream = at::hip::getCurrentHIPStream();
     at::Tensor partial;
     if (splitk > 1)
         partial = at::empty({(long)splitk * dx.numel()},
                             dy.options().dtype(at::kFloat));
     DISPATCH_CONV_FT(dy.scalar_type(), {
-        hipLaunchKernelGGL(conv_bwd_data_kernel<scalar_t>, grid, dim3(256), 0,
-                           stream, (const scalar_t*)dyc.data_ptr(),
-                           w.data_ptr<float>(), (scalar_t*)dx.data_ptr(),
-                           splitk > 1 ? partial.data_ptr<float>() : nullptr,
-                           gm, splitk);
+        if (s2)
+            hipLaunchKernelGGL(conv_bwd_data_s2_kernel<scalar_t>, grid,
+                               dim3(256), 0, stream,
+                               (const scalar_t*)dyc.data_ptr(),
+                               w.data_ptr<float>(), (scalar_t*)dx.data_ptr(),
+                               splitk > 1 ? partial.data_ptr<float>()
+                                          : nullptr,
+                               gm, splitk, nyp);
+        else
+            hipLaunchKernelGGL(conv_bwd_data_kernel<scalar_t>, grid,
+                               dim3(256), 0, stream,
+                               (const scalar_t*)dyc.data_ptr(),
+                               w.data_ptr<float>(), (scalar_t*)dx.data_ptr(),
+                               splitk > 1 ? partial.data_ptr<float>()
+                                          : nullptr,
+                               gm, splitk);
         if (splitk > 1) {
             const long total = dx.numel();
             const int blocks = (int)((total + 255) / 256);
