@@ -55,6 +55,10 @@ class BatchedLMClientTrainer:
         R = len(slots)
         device = self.device
         bptt = cfg['bptt']
+        if device.type == 'cuda' and cfg.get('hip_graphs', True):
+            return self._train_group_graphed(rate, slots, user_idx,
+                                             locals_list, label_split, lr,
+                                             logger)
         model = self._model(rate, R)
         pack_states(model, locals_list)
         model.train(True)
@@ -115,4 +119,53 @@ class BatchedLMClientTrainer:
                                           'train', n=n)
         template_keys = list(locals_list[0].keys())
         states = unpack_states(model, template_keys)
+        return list(zip(slots, [dict(st) for st in states]))
+
+
+    def _train_group_graphed(self, rate, slots, user_idx, locals_list,
+                             label_split, lr, logger=None):
+        """hipGraph path: stage token windows once, replay the captured step
+        for every full window per epoch; a ragged last window runs eagerly."""
+        from .graphs import LMGraphedStep
+        cfg = self.cfg
+        R = len(slots)
+        device = self.device
+        bptt = cfg['bptt']
+        rows = torch.stack([self.token[self.data_split[user_idx[m]]]
+                            for m in slots])
+        L = rows.size(2)
+        n_full = L // bptt
+        key = ('g', rate, R, rows.size(1), n_full, lr)
+        if not hasattr(self, '_graph_cache'):
+            self._graph_cache = {}
+        if key not in self._graph_cache:
+            model = make_batched_transformer(cfg, rate, R).to(device)
+            model.train(True)
+            self._graph_cache[key] = LMGraphedStep(
+                model, R, rows.size(1), bptt, n_full, lr, cfg['momentum'],
+                cfg['weight_decay'], cfg['num_tokens'], self._amp, device)
+        gs = self._graph_cache[key]
+        pack_states(gs.model, locals_list)
+        masks = None
+        if cfg['mask']:
+            masks = torch.zeros(R, cfg['num_tokens'], device=device)
+            for i, m in enumerate(slots):
+                masks[i, label_split[user_idx[m]]] = 1
+        gs.begin_round(rows, masks)
+        tail = rows[:, :, n_full * bptt:] if L % bptt else None
+        for _ in range(cfg['num_epochs']['local']):
+            gs.run_window_pass()
+            if tail is not None and tail.size(2) > 0:
+                gs.tail_step(tail)
+        if logger is not None:
+            m = gs.metrics.detach().cpu()
+            for i in range(R):
+                cnt = max(int(m[i, 2].item()), 1)
+                l = m[i, 0].item() / cnt
+                import math as _math
+                logger.append({'Local-Loss': l,
+                               'Local-Perplexity': float(_math.exp(min(l, 30)))},
+                              'train', n=cnt)
+        template_keys = list(locals_list[0].keys())
+        states = unpack_states(gs.model, template_keys)
         return list(zip(slots, [dict(st) for st in states]))

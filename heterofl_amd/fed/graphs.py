@@ -179,3 +179,137 @@ class GraphedGroupStep:
     def ensure_captured(self):
         if self.graph is None:
             self.capture()
+
+
+class LMGraphedStep:
+    """hipGraph-captured masked-LM training step for a (rate, R) client
+    group (reference local loop: src/train_transformer_fed.py:163-176).
+
+    LM client data is static across local epochs (ordered bptt windows of
+    the client's token rows, no augmentation), so the round stages the
+    full-window tensor (n_win, R, B, S) once and the graph gathers window
+    counter % n_win with a device-side counter.  torch's graph-private
+    philox state makes the in-graph Bernoulli masking / dropout draw fresh
+    randomness each replay."""
+
+    def __init__(self, model, R, n_rows, bptt, n_win, lr, momentum,
+                 weight_decay, num_tokens, amp, device):
+        self.model = model
+        self.R = R
+        self.bptt = bptt
+        self.n_win = n_win
+        self.lr = lr
+        self.momentum = momentum
+        self.weight_decay = weight_decay
+        self.amp = amp
+        self.device = device
+        self.params = [p for p in model.parameters() if p.requires_grad]
+        self.bufs = [torch.zeros_like(p) for p in self.params]
+        self.windows = torch.zeros(n_win, R, n_rows, bptt, dtype=torch.long,
+                                   device=device)
+        self.masks = torch.ones(R, num_tokens, device=device)
+        self.counter = torch.zeros(1, dtype=torch.long, device=device)
+        self.metrics = torch.zeros(R, 3, device=device)
+        self.graph = None
+        self._capturing = False
+        self._captured_grads = None
+        self._native = (device.type == 'cuda' and native_ops.use_native(device))
+        if self._native:
+            from ..ops.fused import GraphClipSGD
+            self.opt = GraphClipSGD(self.params, self.bufs, R, device)
+
+    def _one_step(self):
+        w = self.counter - (self.counter // self.n_win) * self.n_win
+        tokens = self.windows.index_select(0, w).squeeze(0)
+        self._step_on(tokens)
+        with torch.no_grad():
+            self.counter += 1
+
+    def tail_step(self, tokens):
+        """Eager step for a ragged last window (shorter S than the graph)."""
+        self._step_on(tokens)
+
+    def _step_on(self, tokens):
+        from .batched_lm import lm_masked_ce
+        with torch.autocast('cuda', torch.bfloat16, enabled=self.amp):
+            logits = self.model(tokens)
+        losses = lm_masked_ce(logits.float(), tokens, self.masks)
+        with torch.no_grad():
+            n = tokens.size(1) * tokens.size(2)
+            self.metrics[:, 0] += losses.detach() * n
+            self.metrics[:, 2] += n
+        if self._native:
+            raw = torch.autograd.grad(losses.sum(), self.params)
+            grads = [t if t.is_contiguous() else t.contiguous() for t in raw]
+            self._captured_grads = grads
+            if not self._capturing:
+                self.opt.bind(grads)
+            self.opt.launch(1.0, self.lr, self.momentum, self.weight_decay)
+        else:
+            raw = torch.autograd.grad(losses.sum(), self.params)
+            with torch.no_grad():
+                grads = [g.contiguous() for g in raw]
+                sq = None
+                views = [g.view(self.R, -1) for g in grads]
+                for v in views:
+                    s = (v.float() ** 2).sum(dim=1)
+                    sq = s if sq is None else sq + s
+                scale = (1.0 / (sq.sqrt() + 1e-6)).clamp(max=1.0)
+                for v in views:
+                    v.mul_(scale.unsqueeze(1).to(v.dtype))
+                torch._foreach_add_(grads, self.params,
+                                    alpha=self.weight_decay)
+                torch._foreach_mul_(self.bufs, self.momentum)
+                torch._foreach_add_(self.bufs, grads)
+                torch._foreach_add_(self.params, self.bufs, alpha=-self.lr)
+
+    def capture(self):
+        saved_p = [p.detach().clone() for p in self.params]
+        saved_b = [b.clone() for b in self.bufs]
+        saved_m = self.metrics.clone()
+        torch.cuda.synchronize()
+        s = torch.cuda.Stream()
+        s.wait_stream(torch.cuda.current_stream())
+        with torch.cuda.stream(s):
+            for _ in range(3):
+                self.counter.zero_()
+                self._one_step()
+        torch.cuda.current_stream().wait_stream(s)
+        self.counter.zero_()
+        self.graph = torch.cuda.CUDAGraph()
+        self._capturing = True
+        with torch.cuda.graph(self.graph):
+            self._one_step()
+        self._capturing = False
+        if self._native:
+            self.opt.bind(self._captured_grads)
+        with torch.no_grad():
+            for p, sp in zip(self.params, saved_p):
+                p.copy_(sp)
+            for b, sb in zip(self.bufs, saved_b):
+                b.copy_(sb)
+            self.metrics.copy_(saved_m)
+            self.counter.zero_()
+        torch.cuda.synchronize()
+
+    def begin_round(self, rows, masks):
+        """rows (R, B, L): client token rows; stages the window tensor."""
+        with torch.no_grad():
+            for w in range(self.n_win):
+                self.windows[w].copy_(
+                    rows[:, :, w * self.bptt:(w + 1) * self.bptt])
+            if masks is not None:
+                self.masks.copy_(masks)
+            else:
+                self.masks.fill_(1)
+            for b in self.bufs:
+                b.zero_()
+            self.metrics.zero_()
+            self.counter.zero_()
+
+    def run_window_pass(self):
+        """One epoch's worth of full windows (counter wraps modulo n_win)."""
+        if self.graph is None:
+            self.capture()
+        for _ in range(self.n_win):
+            self.graph.replay()

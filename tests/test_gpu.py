@@ -397,3 +397,51 @@ def test_fused_head_matches_torch():
     assert (feat.grad - f2.grad).abs().max().item() < 1e-5
     assert (w.grad - w2.grad).abs().max().item() < 1e-4
     assert (b.grad - b2.grad).abs().max().item() < 1e-5
+
+
+@needs_gpu
+def test_lm_graph_matches_eager(base_cfg):
+    """LM hipGraph path == eager batched path (RNG-free config)."""
+    from heterofl_amd.data import fetch_dataset, split_dataset
+    from heterofl_amd.fed.batched_lm_trainer import BatchedLMClientTrainer
+    from heterofl_amd.models import make_model
+    from heterofl_amd.utils import process_dataset
+    cfg = make_cfg(base_cfg, '1_3_1_iid_fix_a1_bn_1_1',
+                   data_name='WikiText2', model_name='transformer')
+    cfg['device'] = 'cuda:0'
+    cfg['compute_dtype'] = 'float32'
+    cfg['num_epochs'] = {'global': 1, 'local': 2}
+    cfg['transformer']['dropout'] = 0.0
+    cfg['mask_rate'] = 0.0
+    torch.manual_seed(0)
+    ds = fetch_dataset('WikiText2', synthetic=True, synthetic_size=60_000)
+    ds['train'].vocab.itos = ds['train'].vocab.itos[:500]
+    ds['train'].token = ds['train'].token % 500
+    ds['test'].token = ds['test'].token % 500
+    process_dataset(ds, cfg)
+    cfg['num_tokens'] = 500
+    data_split, label_split = split_dataset(ds, 3, 'iid')
+    torch.manual_seed(1)
+    gm = make_model(cfg).to('cuda:0')
+    gp = gm.state_dict()
+    locals_ = [{k: v.clone() for k, v in gp.items()} for _ in range(3)]
+    user_idx = [0, 1, 2]
+    rates = {u: 1.0 for u in user_idx}
+    results = {}
+    for use_graph in (False, True):
+        cfg2 = dict(cfg)
+        cfg2['hip_graphs'] = use_graph
+        tr = BatchedLMClientTrainer(cfg2)
+        tr.set_data(ds, data_split)
+        torch.manual_seed(3)
+        torch.cuda.manual_seed_all(3)
+        out = tr.train_clients([0, 1, 2], user_idx,
+                               [dict(l) for l in locals_], rates, None,
+                               label_split, 0.1)
+        results[use_graph] = dict(out)
+    for m in range(3):
+        for k in results[False][m]:
+            a = results[False][m][k].float().cpu()
+            b = results[True][m][k].float().cpu()
+            diff = (a - b).abs().max().item()
+            assert diff < 1e-4, (m, k, diff)
