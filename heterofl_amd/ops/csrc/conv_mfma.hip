@@ -144,38 +144,39 @@ conv_fwd_kernel(const T* __restrict__ x, const float* __restrict__ w,
     __syncthreads();
 
     f32x4 acc[2][2] = {};
+    const int mm_a = tid >> 2, kkb = (tid & 3) * 8;
+    const int pp_b = tid >> 2;
+    float va[8], vb[8];
+    // loader for one K-step into registers (2-phase: issue next tile's
+    // loads before this tile's MFMA so HBM/L2 latency hides under it)
+    auto load_regs = [&](int k0) {
+        const int m = m0 + mm_a;
+        const float* wrow = w + (long)(g * gm.Cout + m) * K + k0 + kkb;
+#pragma unroll
+        for (int j = 0; j < 8; ++j) {
+            const int k = k0 + kkb + j;
+            va[j] = (m < M && k < K) ? wrow[j] : 0.f;
+        }
+#pragma unroll
+        for (int j = 0; j < 8; ++j) {
+            const int k = k0 + kkb + j;
+            vb[j] = 0.f;
+            if (k < K) {
+                const int cin = k / kk2, r = k - cin * kk2;
+                const int kh = r / gm.khw, kw = r - kh * gm.khw;
+                const int ih = t_ihb[pp_b] + kh, iw = t_iwb[pp_b] + kw;
+                if (ih >= 0 && ih < gm.H && iw >= 0 && iw < gm.W)
+                    vb[j] = ld_f32(x + t_xbase[pp_b] + (long)cin * HW
+                                   + ih * gm.W + iw);
+            }
+        }
+    };
+    load_regs(ks);
     for (int k0 = ks; k0 < ke; k0 += BK) {
-        {   // A tile: thread owns 8 contiguous k of one row -> one b128
-            const int mm = tid >> 2, kkb = (tid & 3) * 8;
-            const int m = m0 + mm;
-            float v[8];
-            const float* wrow = w + (long)(g * gm.Cout + m) * K + k0 + kkb;
-#pragma unroll
-            for (int j = 0; j < 8; ++j) {
-                const int k = k0 + kkb + j;
-                v[j] = (m < M && k < K) ? wrow[j] : 0.f;
-            }
-            st8_lds(&a_lds[mm][kkb], v);
-        }
-        {   // B tile: thread owns 8 contiguous k of one pixel -> one b128
-            const int pp = tid >> 2, kkb = (tid & 3) * 8;
-            float v[8];
-#pragma unroll
-            for (int j = 0; j < 8; ++j) {
-                const int k = k0 + kkb + j;
-                v[j] = 0.f;
-                if (k < K) {
-                    const int cin = k / kk2, r = k - cin * kk2;
-                    const int kh = r / gm.khw, kw = r - kh * gm.khw;
-                    const int ih = t_ihb[pp] + kh, iw = t_iwb[pp] + kw;
-                    if (ih >= 0 && ih < gm.H && iw >= 0 && iw < gm.W)
-                        v[j] = ld_f32(x + t_xbase[pp] + (long)cin * HW
-                                      + ih * gm.W + iw);
-                }
-            }
-            st8_lds(&b_lds[pp][kkb], v);
-        }
+        st8_lds(&a_lds[mm_a][kkb], va);
+        st8_lds(&b_lds[pp_b][kkb], vb);
         __syncthreads();
+        if (k0 + BK < ke) load_regs(k0 + BK);
 #pragma unroll
         for (int fm = 0; fm < 2; ++fm)
 #pragma unroll
@@ -276,52 +277,50 @@ conv_bwd_data_kernel(const T* __restrict__ dy, const float* __restrict__ w,
     const int js = sp * njc * BK;
     const int je = min(J, js + njc * BK);
     f32x4 acc[2][2] = {};
-    for (int j0 = js; j0 < je; j0 += BK) {
-        {
-            const int cc = tid >> 2, jjb = (tid & 3) * 8;
-            const int c = c0 + cc;
-            float v[8];
+    const int cc_a = tid >> 2, jjb = (tid & 3) * 8;
+    float va[8], vb[8];
+    auto load_regs = [&](int j0) {
+        const int c = c0 + cc_a;
 #pragma unroll
-            for (int j8 = 0; j8 < 8; ++j8) {
-                const int j = j0 + jjb + j8;
-                v[j8] = 0.f;
-                if (c < gm.Cin && j < J) {
-                    const int cout = j / kk2, r = j - cout * kk2;
-                    v[j8] = w[(long)(g * gm.Cout + cout) * K + c * kk2 + r];
-                }
+        for (int j8 = 0; j8 < 8; ++j8) {
+            const int j = j0 + jjb + j8;
+            va[j8] = 0.f;
+            if (c < gm.Cin && j < J) {
+                const int cout = j / kk2, r = j - cout * kk2;
+                va[j8] = w[(long)(g * gm.Cout + cout) * K + c * kk2 + r];
             }
-            st8_lds(&a_lds[cc][jjb], v);
         }
-        {
-            const int qq = tid >> 2, jjb = (tid & 3) * 8;
-            float v[8];
 #pragma unroll
-            for (int j8 = 0; j8 < 8; ++j8) {
-                const int j = j0 + jjb + j8;
-                v[j8] = 0.f;
-                if (j < J) {
-                    const int cout = j / kk2, r = j - cout * kk2;
-                    const int kh = r / gm.khw, kw = r - kh * gm.khw;
-                    const int ohs = t_oh[qq] - kh, ows = t_ow[qq] - kw;
-                    if (ohs >= 0 && ows >= 0) {
-                        if (gm.stride == 1) {
-                            if (ohs < gm.OH && ows < gm.OW)
-                                v[j8] = ld_f32(dy + t_dybase[qq]
-                                               + (long)cout * OHW
-                                               + ohs * gm.OW + ows);
-                        } else if ((ohs & 1) == 0 && (ows & 1) == 0) {
-                            const int oh = ohs >> 1, ow = ows >> 1;
-                            if (oh < gm.OH && ow < gm.OW)
-                                v[j8] = ld_f32(dy + t_dybase[qq]
-                                               + (long)cout * OHW
-                                               + oh * gm.OW + ow);
-                        }
+        for (int j8 = 0; j8 < 8; ++j8) {
+            const int j = j0 + jjb + j8;
+            vb[j8] = 0.f;
+            if (j < J) {
+                const int cout = j / kk2, r = j - cout * kk2;
+                const int kh = r / gm.khw, kw = r - kh * gm.khw;
+                const int ohs = t_oh[cc_a] - kh, ows = t_ow[cc_a] - kw;
+                if (ohs >= 0 && ows >= 0) {
+                    if (gm.stride == 1) {
+                        if (ohs < gm.OH && ows < gm.OW)
+                            vb[j8] = ld_f32(dy + t_dybase[cc_a]
+                                            + (long)cout * OHW
+                                            + ohs * gm.OW + ows);
+                    } else if ((ohs & 1) == 0 && (ows & 1) == 0) {
+                        const int oh = ohs >> 1, ow = ows >> 1;
+                        if (oh < gm.OH && ow < gm.OW)
+                            vb[j8] = ld_f32(dy + t_dybase[cc_a]
+                                            + (long)cout * OHW
+                                            + oh * gm.OW + ow);
                     }
                 }
             }
-            st8_lds(&b_lds[qq][jjb], v);
         }
+    };
+    if (js < je) load_regs(js);
+    for (int j0 = js; j0 < je; j0 += BK) {
+        st8_lds(&a_lds[cc_a][jjb], va);
+        st8_lds(&b_lds[cc_a][jjb], vb);
         __syncthreads();
+        if (j0 + BK < je) load_regs(j0 + BK);
 #pragma unroll
         for (int fm = 0; fm < 2; ++fm)
 #pragma unroll
@@ -426,44 +425,42 @@ conv_bwd_data_s2_kernel(const T* __restrict__ dy, const float* __restrict__ w,
     const int js = sp * njc * BK;
     const int je = min(J, js + njc * BK);
     f32x4 acc[2][2] = {};
+    const int cc_a = tid >> 2, jjb = (tid & 3) * 8;
+    float va[8], vb[8];
+    auto load_regs = [&](int j0) {
+        const int c = c0 + cc_a;
+#pragma unroll
+        for (int j8 = 0; j8 < 8; ++j8) {
+            const int j = j0 + jjb + j8;
+            va[j8] = 0.f;
+            if (c < gm.Cin && j < J) {
+                const int cout = j / tap2, rr = j - cout * tap2;
+                const int kh = lkh[rr / nkw], kw = lkw[rr - (rr / nkw) * nkw];
+                va[j8] = w[(long)(g * gm.Cout + cout) * K + c * kk2
+                           + kh * gm.khw + kw];
+            }
+        }
+#pragma unroll
+        for (int j8 = 0; j8 < 8; ++j8) {
+            const int j = j0 + jjb + j8;
+            vb[j8] = 0.f;
+            if (j < J) {
+                const int cout = j / tap2, rr = j - cout * tap2;
+                const int a = rr / nkw, b = rr - a * nkw;
+                const int oh = t_oh2[cc_a] + skh[a];
+                const int ow = t_ow2[cc_a] + skw[b];
+                if (oh >= 0 && oh < gm.OH && ow >= 0 && ow < gm.OW)
+                    vb[j8] = ld_f32(dy + t_dybase[cc_a] + (long)cout * OHW
+                                    + oh * gm.OW + ow);
+            }
+        }
+    };
+    if (js < je) load_regs(js);
     for (int j0 = js; j0 < je; j0 += BK) {
-        {
-            const int cc = tid >> 2, jjb = (tid & 3) * 8;
-            const int c = c0 + cc;
-            float v[8];
-#pragma unroll
-            for (int j8 = 0; j8 < 8; ++j8) {
-                const int j = j0 + jjb + j8;
-                v[j8] = 0.f;
-                if (c < gm.Cin && j < J) {
-                    const int cout = j / tap2, rr = j - cout * tap2;
-                    const int kh = lkh[rr / nkw], kw = lkw[rr - (rr / nkw) * nkw];
-                    v[j8] = w[(long)(g * gm.Cout + cout) * K + c * kk2
-                              + kh * gm.khw + kw];
-                }
-            }
-            st8_lds(&a_lds[cc][jjb], v);
-        }
-        {
-            const int qq = tid >> 2, jjb = (tid & 3) * 8;
-            float v[8];
-#pragma unroll
-            for (int j8 = 0; j8 < 8; ++j8) {
-                const int j = j0 + jjb + j8;
-                v[j8] = 0.f;
-                if (j < J) {
-                    const int cout = j / tap2, rr = j - cout * tap2;
-                    const int a = rr / nkw, b = rr - a * nkw;
-                    const int oh = t_oh2[qq] + skh[a];
-                    const int ow = t_ow2[qq] + skw[b];
-                    if (oh >= 0 && oh < gm.OH && ow >= 0 && ow < gm.OW)
-                        v[j8] = ld_f32(dy + t_dybase[qq] + (long)cout * OHW
-                                       + oh * gm.OW + ow);
-                }
-            }
-            st8_lds(&b_lds[qq][jjb], v);
-        }
+        st8_lds(&a_lds[cc_a][jjb], va);
+        st8_lds(&b_lds[cc_a][jjb], vb);
         __syncthreads();
+        if (j0 + BK < je) load_regs(j0 + BK);
 #pragma unroll
         for (int fm = 0; fm < 2; ++fm)
 #pragma unroll
