@@ -226,8 +226,13 @@ class LMGraphedStep:
             self.counter += 1
 
     def tail_step(self, tokens):
-        """Eager step for a ragged last window (shorter S than the graph)."""
+        """Eager step for a ragged last window (shorter S than the graph).
+        The eager step rebinds the shared chunk table to its own grad
+        tensors; restore the captured binding so later replays read the
+        graph-pool grads again."""
         self._step_on(tokens)
+        if self._native and self.graph is not None:
+            self.opt.bind(self._graph_grads)
 
     def _step_on(self, tokens):
         from .batched_lm import lm_masked_ce
@@ -282,7 +287,8 @@ class LMGraphedStep:
             self._one_step()
         self._capturing = False
         if self._native:
-            self.opt.bind(self._captured_grads)
+            self._graph_grads = self._captured_grads
+            self.opt.bind(self._graph_grads)
         with torch.no_grad():
             for p, sp in zip(self.params, saved_p):
                 p.copy_(sp)
