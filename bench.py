@@ -55,7 +55,8 @@ def main():
     p.add_argument('--gpus', type=int, default=1)
     p.add_argument('--steps', type=int, default=5)
     p.add_argument('--warmup', type=int, default=2)
-    p.add_argument('--dtype', type=str, default='bfloat16')
+    p.add_argument('--dtype', type=str, default='bfloat16',
+                   choices=['float32', 'bfloat16', 'fp8'])
     p.add_argument('--engine', type=str, default='batched',
                    choices=['batched', 'sequential'])
     args = p.parse_args()

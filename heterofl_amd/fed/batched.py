@@ -424,9 +424,12 @@ class BatchedClientTrainer:
         self.dataset = None
         self.data_split = None
         self.augment = None
-        # bf16 local training (master weights fp32, compute bf16)
-        self._amp = (cfg.get('compute_dtype') == 'bfloat16'
+        # bf16/fp8 local training (master weights fp32, compute bf16, and
+        # in fp8 mode the conv GEMMs on fp8 MFMA)
+        self._amp = (cfg.get('compute_dtype') in ('bfloat16', 'fp8')
                      and self.device.type == 'cuda')
+        if self.device.type == 'cuda':
+            native_ops.set_fp8(cfg.get('compute_dtype') == 'fp8')
         self._shard_cache = {}
         self._model_cache = {}
         self._graph_cache = {}

@@ -63,3 +63,18 @@ def use_native(tensor_or_device=None):
 def native_conv_enabled():
     """The MFMA conv path can be toggled off for A/B benchmarking."""
     return os.environ.get('HETEROFL_NATIVE_CONV', '1') == '1'
+
+
+_FP8 = False
+
+
+def set_fp8(on):
+    """fp8 local-training mode (BASELINE config 5): the conv GEMMs run on
+    CDNA4 fp8 MFMA (e4m3 weights/activations x e5m2 grads), tensors stay
+    bf16 in memory."""
+    global _FP8
+    _FP8 = bool(on)
+
+
+def fp8_enabled():
+    return _FP8

@@ -30,11 +30,13 @@ void clip_sgd_step(at::Tensor table_blob, int64_t n_chunks,
                    double momentum, double weight_decay);
 at::Tensor conv_fwd(at::Tensor x, at::Tensor w, at::Tensor bias,
                     at::Tensor residual, int64_t groups, int64_t stride,
-                    int64_t pad);
+                    int64_t pad, int64_t fp8);
 at::Tensor conv_bwd_data(at::Tensor dy, at::Tensor w, int64_t groups,
-                         int64_t stride, int64_t pad, int64_t H, int64_t W);
+                         int64_t stride, int64_t pad, int64_t H, int64_t W,
+                         int64_t fp8);
 at::Tensor conv_bwd_weight(at::Tensor dy, at::Tensor x, int64_t groups,
-                           int64_t stride, int64_t pad, int64_t khw);
+                           int64_t stride, int64_t pad, int64_t khw,
+                           int64_t fp8);
 at::Tensor mfma_probe(at::Tensor A, at::Tensor B);
 std::vector<at::Tensor> head_fwd(at::Tensor feat, at::Tensor w, at::Tensor b,
                                  int64_t R);

@@ -80,6 +80,68 @@ __device__ __forceinline__ void st8_lds(float* dst, const float* v) {
     *(f4*)(dst + 4) = *(const f4*)(v + 4);
 }
 
+// fp8 LDS tile element tags (OCP formats; gfx950 fp8 MFMA is e4m3/e5m2 —
+// cdna_hip_programming.md §4).  e4m3 carries weights/activations, e5m2
+// gradients; the mixed mfma_f32_16x16x32_{fp8,bf8}_{fp8,bf8} forms multiply
+// them directly (CDNA4 fp8 MFMA, BASELINE config 5).
+#include <hip/hip_fp8.h>
+struct e4m3 { unsigned char x; };
+struct e5m2 { unsigned char x; };
+typedef long long i64;
+
+__device__ __forceinline__ void st8_lds(e4m3* dst, const float* v) {
+    unsigned char t[8];
+#pragma unroll
+    for (int j = 0; j < 8; ++j)
+        t[j] = __hip_cvt_float_to_fp8(v[j], __HIP_SATFINITE, __HIP_E4M3);
+    *(i64*)dst = *(const i64*)t;
+}
+__device__ __forceinline__ void st8_lds(e5m2* dst, const float* v) {
+    unsigned char t[8];
+#pragma unroll
+    for (int j = 0; j < 8; ++j)
+        t[j] = __hip_cvt_float_to_fp8(v[j], __HIP_SATFINITE, __HIP_E5M2);
+    *(i64*)dst = *(const i64*)t;
+}
+
+// mixed-type MFMA: a_row/b_col point at the lane's LDS rows; the k-offset
+// (l/16)*8 selects the lane's 8-element fragment (layout as above).
+template <typename TA, typename TB>
+__device__ __forceinline__ f32x4 mfma_tile2(const TA* a_row, const TB* b_col,
+                                            f32x4 acc) {
+    return mfma_tile<TA>(a_row, b_col, acc);  // TA == TB fast path
+}
+template <>
+__device__ __forceinline__ f32x4 mfma_tile2<e4m3, e4m3>(const e4m3* a_row,
+                                                        const e4m3* b_col,
+                                                        f32x4 acc) {
+    const int l = threadIdx.x & (WAVE - 1);
+    const int kb = (l >> 4) * 8;
+    i64 a = *(const i64*)(a_row + kb);
+    i64 b = *(const i64*)(b_col + kb);
+    return __builtin_amdgcn_mfma_f32_16x16x32_fp8_fp8(a, b, acc, 0, 0, 0);
+}
+template <>
+__device__ __forceinline__ f32x4 mfma_tile2<e4m3, e5m2>(const e4m3* a_row,
+                                                        const e5m2* b_col,
+                                                        f32x4 acc) {
+    const int l = threadIdx.x & (WAVE - 1);
+    const int kb = (l >> 4) * 8;
+    i64 a = *(const i64*)(a_row + kb);
+    i64 b = *(const i64*)(b_col + kb);
+    return __builtin_amdgcn_mfma_f32_16x16x32_fp8_bf8(a, b, acc, 0, 0, 0);
+}
+template <>
+__device__ __forceinline__ f32x4 mfma_tile2<e5m2, e4m3>(const e5m2* a_row,
+                                                        const e4m3* b_col,
+                                                        f32x4 acc) {
+    const int l = threadIdx.x & (WAVE - 1);
+    const int kb = (l >> 4) * 8;
+    i64 a = *(const i64*)(a_row + kb);
+    i64 b = *(const i64*)(b_col + kb);
+    return __builtin_amdgcn_mfma_f32_16x16x32_bf8_fp8(a, b, acc, 0, 0, 0);
+}
+
 struct ConvGeom {
     int G;
     int N, H, W;
@@ -94,14 +156,14 @@ struct ConvGeom {
 // epilogue (the ResNet block's `out += shortcut`, src/models/resnet.py:49).
 // splitk > 1: each split writes fp32 partial slabs (sp, N*G*Cout*OHW);
 // conv_out_reduce_kernel sums them in fixed order and applies bias/residual.
-template <typename T>
+template <typename T, typename TA, typename TB>
 __global__ void __launch_bounds__(256)
 conv_fwd_kernel(const T* __restrict__ x, const float* __restrict__ w,
                 const float* __restrict__ bias, const T* __restrict__ residual,
                 T* __restrict__ y, float* __restrict__ partial, ConvGeom gm,
                 int splitk) {
-    __shared__ T a_lds[BM][LDK];
-    __shared__ T b_lds[BP][LDK];
+    __shared__ TA a_lds[BM][LDK];
+    __shared__ TB b_lds[BP][LDK];
     __shared__ int t_ihb[BP], t_iwb[BP];
     __shared__ long t_xbase[BP], t_ybase[BP];
     const int g = blockIdx.z % gm.G;
@@ -181,7 +243,7 @@ conv_fwd_kernel(const T* __restrict__ x, const float* __restrict__ w,
         for (int fm = 0; fm < 2; ++fm)
 #pragma unroll
             for (int fp = 0; fp < 2; ++fp)
-                acc[fm][fp] = mfma_tile<T>(
+                acc[fm][fp] = mfma_tile2<TA, TB>(
                     &a_lds[wm + fm * 16 + (l & 15)][0],
                     &b_lds[wp + fp * 16 + (l & 15)][0], acc[fm][fp]);
         __syncthreads();
@@ -228,13 +290,13 @@ conv_out_reduce_kernel(const float* __restrict__ partial,
 }
 
 // -------------------------------------------------------- bwd-data kernel
-template <typename T>
+template <typename T, typename TA, typename TB>
 __global__ void __launch_bounds__(256)
 conv_bwd_data_kernel(const T* __restrict__ dy, const float* __restrict__ w,
                      T* __restrict__ dx, float* __restrict__ partial,
                      ConvGeom gm, int splitk) {
-    __shared__ T a_lds[BM][LDK];
-    __shared__ T b_lds[BP][LDK];
+    __shared__ TA a_lds[BM][LDK];
+    __shared__ TB b_lds[BP][LDK];
     __shared__ int t_oh[BP], t_ow[BP];  // ih+pad, iw+pad (pre-division)
     __shared__ long t_dybase[BP], t_xbase[BP];
     const int g = blockIdx.z % gm.G;
@@ -325,7 +387,7 @@ conv_bwd_data_kernel(const T* __restrict__ dy, const float* __restrict__ w,
         for (int fm = 0; fm < 2; ++fm)
 #pragma unroll
             for (int fp = 0; fp < 2; ++fp)
-                acc[fm][fp] = mfma_tile<T>(
+                acc[fm][fp] = mfma_tile2<TA, TB>(
                     &a_lds[wm + fm * 16 + (l & 15)][0],
                     &b_lds[wp + fp * 16 + (l & 15)][0], acc[fm][fp]);
         __syncthreads();
@@ -356,13 +418,13 @@ conv_bwd_data_kernel(const T* __restrict__ dy, const float* __restrict__ w,
 // input pixel (ohs/ows parity).  Pixels are tiled by parity class
 // (ph, pw): each class is a dense (H/2, W/2) sub-image with a compressed
 // tap list, so no MFMA work is spent on dead taps.  grid.y = 4 * tiles.
-template <typename T>
+template <typename T, typename TA, typename TB>
 __global__ void __launch_bounds__(256)
 conv_bwd_data_s2_kernel(const T* __restrict__ dy, const float* __restrict__ w,
                         T* __restrict__ dx, float* __restrict__ partial,
                         ConvGeom gm, int splitk, int nyp) {
-    __shared__ T a_lds[BM][LDK];
-    __shared__ T b_lds[BP][LDK];
+    __shared__ TA a_lds[BM][LDK];
+    __shared__ TB b_lds[BP][LDK];
     __shared__ int t_oh2[BP], t_ow2[BP];
     __shared__ long t_dybase[BP], t_xbase[BP];
     const int g = blockIdx.z % gm.G;
@@ -465,7 +527,7 @@ conv_bwd_data_s2_kernel(const T* __restrict__ dy, const float* __restrict__ w,
         for (int fm = 0; fm < 2; ++fm)
 #pragma unroll
             for (int fp = 0; fp < 2; ++fp)
-                acc[fm][fp] = mfma_tile<T>(
+                acc[fm][fp] = mfma_tile2<TA, TB>(
                     &a_lds[wm + fm * 16 + (l & 15)][0],
                     &b_lds[wp + fp * 16 + (l & 15)][0], acc[fm][fp]);
         __syncthreads();
@@ -494,12 +556,12 @@ conv_bwd_data_s2_kernel(const T* __restrict__ dy, const float* __restrict__ w,
 // Writes per-split partials (splitp, G*Cout*K) when splitp > 1; a
 // deterministic reduce kernel sums them in fixed order (no atomics, so the
 // hipGraph-captured step replays bit-identically run to run).
-template <typename T>
+template <typename T, typename TA, typename TB>
 __global__ void __launch_bounds__(256)
 conv_bwd_weight_kernel(const T* __restrict__ dy, const T* __restrict__ x,
                        float* __restrict__ out, ConvGeom gm, int splitp) {
-    __shared__ T a_lds[BM][LDK];
-    __shared__ T b_lds[BP][LDK];
+    __shared__ TA a_lds[BM][LDK];
+    __shared__ TB b_lds[BP][LDK];
     __shared__ int t_kcin[BP], t_kkh[BP], t_kkw[BP];   // per-block k tables
     __shared__ int t_ihb[BK], t_iwb[BK];               // per-step p tables
     __shared__ long t_xb[BK], t_dyb[BK];
@@ -593,7 +655,7 @@ conv_bwd_weight_kernel(const T* __restrict__ dy, const T* __restrict__ x,
         for (int fm = 0; fm < 2; ++fm)
 #pragma unroll
             for (int fp = 0; fp < 2; ++fp)
-                acc[fm][fp] = mfma_tile<T>(
+                acc[fm][fp] = mfma_tile2<TA, TB>(
                     &a_lds[wm + fm * 16 + (l & 15)][0],
                     &b_lds[wp + fp * 16 + (l & 15)][0], acc[fm][fp]);
         __syncthreads();
@@ -633,7 +695,7 @@ splitp_reduce_kernel(const float* __restrict__ partials,
 
 at::Tensor conv_fwd(at::Tensor x, at::Tensor w, at::Tensor bias,
                     at::Tensor residual, int64_t groups, int64_t stride,
-                    int64_t pad) {
+                    int64_t pad, int64_t fp8) {
     TORCH_CHECK(x.is_cuda() && x.is_contiguous() && w.is_contiguous());
     TORCH_CHECK(w.scalar_type() == at::kFloat, "weights must be fp32 master");
     ConvGeom gm;
@@ -662,16 +724,34 @@ at::Tensor conv_fwd(at::Tensor x, at::Tensor w, at::Tensor bias,
         partial = at::empty({(long)splitk * y.numel()},
                             x.options().dtype(at::kFloat));
     DISPATCH_CONV_FT(x.scalar_type(), {
-        hipLaunchKernelGGL(conv_fwd_kernel<scalar_t>, grid, dim3(256), 0,
-                           stream, (const scalar_t*)x.data_ptr(),
-                           w.data_ptr<float>(),
-                           bias.defined() ? bias.data_ptr<float>() : nullptr,
-                           residual.defined()
-                               ? (const scalar_t*)residual.data_ptr()
-                               : nullptr,
-                           (scalar_t*)y.data_ptr(),
-                           splitk > 1 ? partial.data_ptr<float>() : nullptr,
-                           gm, splitk);
+        if (fp8 && x.scalar_type() == at::kBFloat16)
+            hipLaunchKernelGGL((conv_fwd_kernel<scalar_t, e4m3, e4m3>), grid,
+                               dim3(256), 0, stream,
+                               (const scalar_t*)x.data_ptr(),
+                               w.data_ptr<float>(),
+                               bias.defined() ? bias.data_ptr<float>()
+                                              : nullptr,
+                               residual.defined()
+                                   ? (const scalar_t*)residual.data_ptr()
+                                   : nullptr,
+                               (scalar_t*)y.data_ptr(),
+                               splitk > 1 ? partial.data_ptr<float>()
+                                          : nullptr,
+                               gm, splitk);
+        else
+            hipLaunchKernelGGL((conv_fwd_kernel<scalar_t, scalar_t, scalar_t>),
+                               grid, dim3(256), 0, stream,
+                               (const scalar_t*)x.data_ptr(),
+                               w.data_ptr<float>(),
+                               bias.defined() ? bias.data_ptr<float>()
+                                              : nullptr,
+                               residual.defined()
+                                   ? (const scalar_t*)residual.data_ptr()
+                                   : nullptr,
+                               (scalar_t*)y.data_ptr(),
+                               splitk > 1 ? partial.data_ptr<float>()
+                                          : nullptr,
+                               gm, splitk);
         if (splitk > 1) {
             const long total = y.numel();
             const int blocks = (int)((total + 255) / 256);
@@ -691,7 +771,8 @@ at::Tensor conv_fwd(at::Tensor x, at::Tensor w, at::Tensor bias,
 }
 
 at::Tensor conv_bwd_data(at::Tensor dy, at::Tensor w, int64_t groups,
-                         int64_t stride, int64_t pad, int64_t H, int64_t W) {
+                         int64_t stride, int64_t pad, int64_t H, int64_t W,
+                         int64_t fp8) {
     TORCH_CHECK(dy.is_cuda() && w.is_contiguous());
     TORCH_CHECK(stride == 1 || stride == 2, "stride must be 1 or 2");
     auto dyc = dy.contiguous();
@@ -726,22 +807,50 @@ at::Tensor conv_bwd_data(at::Tensor dy, at::Tensor w, int64_t groups,
         partial = at::empty({(long)splitk * dx.numel()},
                             dy.options().dtype(at::kFloat));
     DISPATCH_CONV_FT(dy.scalar_type(), {
-        if (s2)
-            hipLaunchKernelGGL(conv_bwd_data_s2_kernel<scalar_t>, grid,
-                               dim3(256), 0, stream,
-                               (const scalar_t*)dyc.data_ptr(),
-                               w.data_ptr<float>(), (scalar_t*)dx.data_ptr(),
-                               splitk > 1 ? partial.data_ptr<float>()
-                                          : nullptr,
-                               gm, splitk, nyp);
-        else
-            hipLaunchKernelGGL(conv_bwd_data_kernel<scalar_t>, grid,
-                               dim3(256), 0, stream,
-                               (const scalar_t*)dyc.data_ptr(),
-                               w.data_ptr<float>(), (scalar_t*)dx.data_ptr(),
-                               splitk > 1 ? partial.data_ptr<float>()
-                                          : nullptr,
-                               gm, splitk);
+        const bool q = fp8 && dy.scalar_type() == at::kBFloat16;
+        if (s2) {
+            if (q)
+                hipLaunchKernelGGL((conv_bwd_data_s2_kernel<scalar_t, e4m3,
+                                                            e5m2>),
+                                   grid, dim3(256), 0, stream,
+                                   (const scalar_t*)dyc.data_ptr(),
+                                   w.data_ptr<float>(),
+                                   (scalar_t*)dx.data_ptr(),
+                                   splitk > 1 ? partial.data_ptr<float>()
+                                              : nullptr,
+                                   gm, splitk, nyp);
+            else
+                hipLaunchKernelGGL((conv_bwd_data_s2_kernel<scalar_t, scalar_t,
+                                                            scalar_t>),
+                                   grid, dim3(256), 0, stream,
+                                   (const scalar_t*)dyc.data_ptr(),
+                                   w.data_ptr<float>(),
+                                   (scalar_t*)dx.data_ptr(),
+                                   splitk > 1 ? partial.data_ptr<float>()
+                                              : nullptr,
+                                   gm, splitk, nyp);
+        } else {
+            if (q)
+                hipLaunchKernelGGL((conv_bwd_data_kernel<scalar_t, e4m3,
+                                                         e5m2>),
+                                   grid, dim3(256), 0, stream,
+                                   (const scalar_t*)dyc.data_ptr(),
+                                   w.data_ptr<float>(),
+                                   (scalar_t*)dx.data_ptr(),
+                                   splitk > 1 ? partial.data_ptr<float>()
+                                              : nullptr,
+                                   gm, splitk);
+            else
+                hipLaunchKernelGGL((conv_bwd_data_kernel<scalar_t, scalar_t,
+                                                         scalar_t>),
+                                   grid, dim3(256), 0, stream,
+                                   (const scalar_t*)dyc.data_ptr(),
+                                   w.data_ptr<float>(),
+                                   (scalar_t*)dx.data_ptr(),
+                                   splitk > 1 ? partial.data_ptr<float>()
+                                              : nullptr,
+                                   gm, splitk);
+        }
         if (splitk > 1) {
             const long total = dx.numel();
             const int blocks = (int)((total + 255) / 256);
@@ -756,7 +865,8 @@ at::Tensor conv_bwd_data(at::Tensor dy, at::Tensor w, int64_t groups,
 }
 
 at::Tensor conv_bwd_weight(at::Tensor dy, at::Tensor x, int64_t groups,
-                           int64_t stride, int64_t pad, int64_t khw) {
+                           int64_t stride, int64_t pad, int64_t khw,
+                           int64_t fp8) {
     auto dyc = dy.contiguous();
     ConvGeom gm;
     gm.G = (int)groups;
@@ -781,22 +891,41 @@ at::Tensor conv_bwd_weight(at::Tensor dy, at::Tensor x, int64_t groups,
                         x.options().dtype(at::kFloat));
     auto stream = at::hip::getCurrentHIPStream();
     dim3 grid((gm.Cout + BM - 1) / BM, (K + BP - 1) / BP, gm.G * splitp);
+    const bool q = fp8 && x.scalar_type() == at::kBFloat16;
     if (splitp == 1) {
         DISPATCH_CONV_FT(x.scalar_type(), {
-            hipLaunchKernelGGL(conv_bwd_weight_kernel<scalar_t>, grid,
-                               dim3(256), 0, stream,
-                               (const scalar_t*)dyc.data_ptr(),
-                               (const scalar_t*)x.data_ptr(),
-                               dw.data_ptr<float>(), gm, 1);
+            if (q)
+                hipLaunchKernelGGL((conv_bwd_weight_kernel<scalar_t, e5m2,
+                                                           e4m3>),
+                                   grid, dim3(256), 0, stream,
+                                   (const scalar_t*)dyc.data_ptr(),
+                                   (const scalar_t*)x.data_ptr(),
+                                   dw.data_ptr<float>(), gm, 1);
+            else
+                hipLaunchKernelGGL((conv_bwd_weight_kernel<scalar_t, scalar_t,
+                                                           scalar_t>),
+                                   grid, dim3(256), 0, stream,
+                                   (const scalar_t*)dyc.data_ptr(),
+                                   (const scalar_t*)x.data_ptr(),
+                                   dw.data_ptr<float>(), gm, 1);
         });
         return dw;
     }
     auto partials = at::empty({splitp, GK}, x.options().dtype(at::kFloat));
     DISPATCH_CONV_FT(x.scalar_type(), {
-        hipLaunchKernelGGL(conv_bwd_weight_kernel<scalar_t>, grid, dim3(256),
-                           0, stream, (const scalar_t*)dyc.data_ptr(),
-                           (const scalar_t*)x.data_ptr(),
-                           partials.data_ptr<float>(), gm, splitp);
+        if (q)
+            hipLaunchKernelGGL((conv_bwd_weight_kernel<scalar_t, e5m2, e4m3>),
+                               grid, dim3(256), 0, stream,
+                               (const scalar_t*)dyc.data_ptr(),
+                               (const scalar_t*)x.data_ptr(),
+                               partials.data_ptr<float>(), gm, splitp);
+        else
+            hipLaunchKernelGGL((conv_bwd_weight_kernel<scalar_t, scalar_t,
+                                                       scalar_t>),
+                               grid, dim3(256), 0, stream,
+                               (const scalar_t*)dyc.data_ptr(),
+                               (const scalar_t*)x.data_ptr(),
+                               partials.data_ptr<float>(), gm, splitp);
     });
     const int threads = 256;
     const int blocks = (int)((GK + threads - 1) / threads);

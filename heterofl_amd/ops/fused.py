@@ -123,29 +123,32 @@ class _GroupedConv(torch.autograd.Function):
 
     @staticmethod
     def forward(ctx, x, weight, bias, residual, groups, stride, pad):
+        from . import fp8_enabled
         ext = require_native()
+        fp8 = 1 if fp8_enabled() else 0
         x = x.contiguous()
         y = ext.conv_fwd(x, weight,
                          bias if bias is not None else torch.Tensor(),
                          residual.contiguous() if residual is not None
                          else torch.Tensor(),
-                         groups, stride, pad)
+                         groups, stride, pad, fp8)
         ctx.save_for_backward(x, weight)
-        ctx.meta = (groups, stride, pad, bias is not None)
+        ctx.meta = (groups, stride, pad, bias is not None, fp8)
         return y
 
     @staticmethod
     def backward(ctx, dy):
         ext = require_native()
         x, w = ctx.saved_tensors
-        groups, stride, pad, has_bias = ctx.meta
+        groups, stride, pad, has_bias, fp8 = ctx.meta
         dy = dy.contiguous()
         dx = dw = db = None
         if ctx.needs_input_grad[0]:
             dx = ext.conv_bwd_data(dy, w, groups, stride, pad,
-                                   x.size(2), x.size(3))
+                                   x.size(2), x.size(3), fp8)
         if ctx.needs_input_grad[1]:
-            dw = ext.conv_bwd_weight(dy, x, groups, stride, pad, w.size(2))
+            dw = ext.conv_bwd_weight(dy, x, groups, stride, pad, w.size(2),
+                                     fp8)
         if has_bias and ctx.needs_input_grad[2]:
             db = dy.float().sum(dim=(0, 2, 3))
         # residual gradient is dy itself (identity add in the epilogue)

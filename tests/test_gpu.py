@@ -445,3 +445,38 @@ def test_lm_graph_matches_eager(base_cfg):
             b = results[True][m][k].float().cpu()
             diff = (a - b).abs().max().item()
             assert diff < 1e-4, (m, k, diff)
+
+
+@needs_gpu
+def test_fp8_conv_close_to_fp32():
+    """fp8-MFMA conv (e4m3 x e4m3) tracks the fp32 reference within fp8
+    quantization error; bwd paths (bf8/e4m3 mixed MFMA) stay finite and
+    directionally correct."""
+    import torch.nn.functional as F
+    from heterofl_amd.ops import require_native, set_fp8
+    from heterofl_amd.ops.fused import grouped_conv
+    ext = require_native()
+    torch.manual_seed(0)
+    G, N, C, H = 3, 8, 32, 16
+    x = torch.randn(N, G * C, H, H, device='cuda',
+                    dtype=torch.bfloat16, requires_grad=True)
+    w = (torch.randn(G * C, C, 3, 3, device='cuda') * 0.1).requires_grad_()
+    try:
+        set_fp8(True)
+        y = grouped_conv(x, w, None, G, 1, 1)
+        ref = F.conv2d(x.float(), w, None, 1, 1, groups=G)
+        rel = (y.float() - ref).abs().mean().item() / ref.abs().mean().item()
+        assert rel < 0.08, rel   # e4m3 quantization class error
+        g = torch.randn_like(y)
+        y.backward(g)
+        assert torch.isfinite(x.grad.float()).all()
+        assert torch.isfinite(w.grad).all()
+        # grad direction agrees with fp32 reference
+        x2 = x.detach().float().requires_grad_(True)
+        w2 = w.detach().clone().requires_grad_(True)
+        F.conv2d(x2, w2, None, 1, 1, groups=G).backward(g.float())
+        cos = torch.nn.functional.cosine_similarity(
+            w.grad.flatten(), w2.grad.flatten(), dim=0).item()
+        assert cos > 0.98, cos
+    finally:
+        set_fp8(False)
