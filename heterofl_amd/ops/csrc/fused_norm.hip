@@ -15,7 +15,10 @@
 
 // ------------------------------------------------------------------ forward
 // y = relu((x - mean_c) * invstd_c * g_c + b_c), per channel c over (N,H,W).
-template <typename T>
+// One workgroup per channel; the channel's N*HW elements are staged in LDS
+// during the stats pass (HeteroFL shapes fit: <= 10240 elems), so the
+// normalize pass reads LDS instead of a second global pass.
+template <typename T, bool STAGE>
 __global__ void __launch_bounds__(256)
 bn_relu_fwd_kernel(const T* __restrict__ x, const float* __restrict__ gamma,
                    const float* __restrict__ beta, T* __restrict__ y,
@@ -23,12 +26,16 @@ bn_relu_fwd_kernel(const T* __restrict__ x, const float* __restrict__ gamma,
                    int N, int C, int HW, float eps) {
     const int c = blockIdx.x;
     __shared__ float scratch[2 * 256 / WAVE];
+    extern __shared__ __attribute__((aligned(16))) char smem[];
+    T* stage = (T*)smem;
     float s1 = 0.f, s2 = 0.f;
     const long chan_off = (long)c * HW;
     const long samp_stride = (long)C * HW;
     for (int i = threadIdx.x; i < N * HW; i += blockDim.x) {
         const int n = i / HW, hw = i - n * HW;
-        const float v = ld_f32(x + n * samp_stride + chan_off + hw);
+        const T raw = x[n * samp_stride + chan_off + hw];
+        if (STAGE) stage[i] = raw;
+        const float v = (float)raw;
         s1 += v;
         s2 += v * v;
     }
@@ -48,7 +55,7 @@ bn_relu_fwd_kernel(const T* __restrict__ x, const float* __restrict__ gamma,
     for (int i = threadIdx.x; i < N * HW; i += blockDim.x) {
         const int n = i / HW, hw = i - n * HW;
         const long off = n * samp_stride + chan_off + hw;
-        const float v = ld_f32(x + off);
+        const float v = STAGE ? (float)stage[i] : ld_f32(x + off);
         st_f32(y + off, fmaxf(v * scale + shift, 0.f));
     }
 }
@@ -200,14 +207,32 @@ std::vector<at::Tensor> bn_relu_fwd(at::Tensor x, at::Tensor gamma,
     auto mean = at::empty({C}, opts);
     auto invstd = at::empty({C}, opts);
     auto stream = at::hip::getCurrentHIPStream();
+    const long stage_bytes = (long)N * HW * x.element_size();
     DISPATCH_FT(x.scalar_type(), {
-        hipLaunchKernelGGL(bn_relu_fwd_kernel<scalar_t>, dim3(C), dim3(256), 0,
-                           stream,
-                           (const scalar_t*)x.data_ptr(),
-                           gamma.defined() ? gamma.data_ptr<float>() : nullptr,
-                           beta.defined() ? beta.data_ptr<float>() : nullptr,
-                           (scalar_t*)y.data_ptr(), mean.data_ptr<float>(),
-                           invstd.data_ptr<float>(), N, C, HW, (float)eps);
+        if (stage_bytes <= 64 * 1024)
+            hipLaunchKernelGGL((bn_relu_fwd_kernel<scalar_t, true>), dim3(C),
+                               dim3(256), (int)stage_bytes, stream,
+                               (const scalar_t*)x.data_ptr(),
+                               gamma.defined() ? gamma.data_ptr<float>()
+                                               : nullptr,
+                               beta.defined() ? beta.data_ptr<float>()
+                                              : nullptr,
+                               (scalar_t*)y.data_ptr(),
+                               mean.data_ptr<float>(),
+                               invstd.data_ptr<float>(), N, C, HW,
+                               (float)eps);
+        else
+            hipLaunchKernelGGL((bn_relu_fwd_kernel<scalar_t, false>), dim3(C),
+                               dim3(256), 0, stream,
+                               (const scalar_t*)x.data_ptr(),
+                               gamma.defined() ? gamma.data_ptr<float>()
+                                               : nullptr,
+                               beta.defined() ? beta.data_ptr<float>()
+                                              : nullptr,
+                               (scalar_t*)y.data_ptr(),
+                               mean.data_ptr<float>(),
+                               invstd.data_ptr<float>(), N, C, HW,
+                               (float)eps);
     });
     return {y, mean, invstd};
 }
