@@ -16,6 +16,8 @@ import torch
 import torch.nn as nn
 import torch.nn.functional as F
 
+from .. import ops as native_ops
+
 
 class SLinear(nn.Module):
     """R stacked Linears: weight (R, out, in), x (R, B, in) -> (R, B, out)."""
@@ -115,9 +117,13 @@ class SMultiheadAttention(nn.Module):
         q = q.permute(0, 1, 3, 2, 4).reshape(R * B * H, S, d)
         k = k.permute(0, 1, 3, 2, 4).reshape(R * B * H, S, d)
         v = v.permute(0, 1, 3, 2, 4).reshape(R * B * H, S, d)
-        scores = torch.bmm(q, k.transpose(1, 2)) / self.temperature
-        attn = F.softmax(scores, dim=-1)
-        out = torch.bmm(attn, v)
+        if native_ops.use_native(q) and S <= 64 and d <= 32:
+            from ..ops.fused import fused_attention
+            out = fused_attention(q, k, v, self.temperature)
+        else:
+            scores = torch.bmm(q, k.transpose(1, 2)) / self.temperature
+            attn = F.softmax(scores, dim=-1)
+            out = torch.bmm(attn, v)
         out = out.reshape(R, B, H, S, d).permute(0, 1, 3, 2, 4).reshape(R, B, S, E)
         return self.scaler(self.linear_o(out))
 

@@ -40,6 +40,11 @@ at::Tensor conv_bwd_weight(at::Tensor dy, at::Tensor x, int64_t groups,
 at::Tensor mfma_probe(at::Tensor A, at::Tensor B);
 std::vector<at::Tensor> head_fwd(at::Tensor feat, at::Tensor w, at::Tensor b,
                                  int64_t R);
+std::vector<at::Tensor> attn_fwd(at::Tensor q, at::Tensor k, at::Tensor v,
+                                 double temperature);
+std::vector<at::Tensor> attn_bwd(at::Tensor dout, at::Tensor q, at::Tensor k,
+                                 at::Tensor v, at::Tensor p,
+                                 double temperature);
 std::vector<at::Tensor> head_bwd(at::Tensor dscores, at::Tensor pooled,
                                  at::Tensor w, int64_t R, int64_t H,
                                  int64_t W, bool bf16_feat, bool want_db);
@@ -63,4 +68,6 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("mfma_probe", &mfma_probe, "16x16x32 bf16 MFMA layout probe");
     m.def("head_fwd", &head_fwd, "fused avgpool+per-client-linear forward");
     m.def("head_bwd", &head_bwd, "fused head backward");
+    m.def("attn_fwd", &attn_fwd, "fused small-S attention forward");
+    m.def("attn_bwd", &attn_bwd, "fused small-S attention backward");
 }

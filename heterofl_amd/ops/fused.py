@@ -229,3 +229,29 @@ class GraphClipSGD:
         ext.clip_sgd_step(self.table, self.n_chunks, self.chunk_client,
                           self.partials, self.normsq, max_norm, lr, momentum,
                           weight_decay)
+
+
+class _FusedAttention(torch.autograd.Function):
+    """Whole-(batch*head) fused attention for bptt<=64, head_dim<=32
+    (ops/csrc/attention.hip): QK^T/temp -> softmax -> PV in one kernel;
+    saves the bf16 softmax matrix for the one-kernel backward."""
+
+    @staticmethod
+    def forward(ctx, q, k, v, temperature):
+        ext = require_native()
+        q, k, v = q.contiguous(), k.contiguous(), v.contiguous()
+        out, p = ext.attn_fwd(q, k, v, temperature)
+        ctx.save_for_backward(q, k, v, p)
+        ctx.temperature = temperature
+        return out
+
+    @staticmethod
+    def backward(ctx, dout):
+        ext = require_native()
+        q, k, v, p = ctx.saved_tensors
+        dq, dk, dv = ext.attn_bwd(dout, q, k, v, p, ctx.temperature)
+        return dq, dk, dv, None
+
+
+def fused_attention(q, k, v, temperature):
+    return _FusedAttention.apply(q, k, v, temperature)

@@ -480,3 +480,28 @@ def test_fp8_conv_close_to_fp32():
         assert cos > 0.98, cos
     finally:
         set_fp8(False)
+
+
+@needs_gpu
+def test_fused_attention_matches_torch():
+    import torch.nn.functional as F
+    from heterofl_amd.ops.fused import fused_attention
+    torch.manual_seed(0)
+    for B, S, d in [(40, 64, 32), (12, 64, 16), (8, 24, 32), (6, 64, 2)]:
+        q = torch.randn(B, S, d, device='cuda', requires_grad=True)
+        k = torch.randn(B, S, d, device='cuda', requires_grad=True)
+        v = torch.randn(B, S, d, device='cuda', requires_grad=True)
+        temp = d ** 0.5
+        out = fused_attention(q, k, v, temp)
+        q2 = q.detach().clone().requires_grad_(True)
+        k2 = k.detach().clone().requires_grad_(True)
+        v2 = v.detach().clone().requires_grad_(True)
+        ref = torch.bmm(F.softmax(torch.bmm(q2, k2.transpose(1, 2)) / temp,
+                                  dim=-1), v2)
+        assert (out - ref).abs().max().item() < 0.03, (B, S, d)
+        g = torch.randn_like(out)
+        out.backward(g)
+        ref.backward(g)
+        for a, b, name in [(q.grad, q2.grad, 'dq'), (k.grad, k2.grad, 'dk'),
+                           (v.grad, v2.grad, 'dv')]:
+            assert (a - b).abs().max().item() < 0.05, (B, S, d, name)
