@@ -46,6 +46,10 @@ class SLayerNorm(nn.Module):
 
     def forward(self, x):
         # x (R, ..., E)
+        if native_ops.use_native(x):
+            from ..ops.fused import fused_layernorm
+            return fused_layernorm(x, self.weight, self.bias,
+                                   self.weight.size(0))
         mu = x.mean(-1, keepdim=True)
         var = x.var(-1, unbiased=False, keepdim=True)
         xhat = (x - mu) * torch.rsqrt(var + 1e-5)

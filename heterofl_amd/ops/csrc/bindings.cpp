@@ -45,6 +45,10 @@ std::vector<at::Tensor> attn_fwd(at::Tensor q, at::Tensor k, at::Tensor v,
 std::vector<at::Tensor> attn_bwd(at::Tensor dout, at::Tensor q, at::Tensor k,
                                  at::Tensor v, at::Tensor p,
                                  double temperature);
+std::vector<at::Tensor> ln_fwd(at::Tensor x, at::Tensor gamma,
+                               at::Tensor beta, int64_t R, double eps);
+std::vector<at::Tensor> ln_bwd(at::Tensor dy, at::Tensor x, at::Tensor gamma,
+                               at::Tensor mean, at::Tensor invstd, int64_t R);
 std::vector<at::Tensor> head_bwd(at::Tensor dscores, at::Tensor pooled,
                                  at::Tensor w, int64_t R, int64_t H,
                                  int64_t W, bool bf16_feat, bool want_db);
@@ -70,4 +74,6 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("head_bwd", &head_bwd, "fused head backward");
     m.def("attn_fwd", &attn_fwd, "fused small-S attention forward");
     m.def("attn_bwd", &attn_bwd, "fused small-S attention backward");
+    m.def("ln_fwd", &ln_fwd, "fused per-client LayerNorm forward");
+    m.def("ln_bwd", &ln_bwd, "fused per-client LayerNorm backward");
 }

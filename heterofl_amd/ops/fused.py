@@ -255,3 +255,27 @@ class _FusedAttention(torch.autograd.Function):
 
 def fused_attention(q, k, v, temperature):
     return _FusedAttention.apply(q, k, v, temperature)
+
+
+class _FusedLayerNorm(torch.autograd.Function):
+    """Per-client LayerNorm over the last dim (ops/csrc/layernorm.hip)."""
+
+    @staticmethod
+    def forward(ctx, x, weight, bias, R, eps):
+        ext = require_native()
+        x = x.contiguous()
+        y, mean, invstd = ext.ln_fwd(x, weight, bias, R, eps)
+        ctx.save_for_backward(x, weight, mean, invstd)
+        ctx.R = R
+        return y
+
+    @staticmethod
+    def backward(ctx, dy):
+        ext = require_native()
+        x, weight, mean, invstd = ctx.saved_tensors
+        dx, dgamma, dbeta = ext.ln_bwd(dy, x, weight, mean, invstd, ctx.R)
+        return dx, dgamma, dbeta, None, None
+
+
+def fused_layernorm(x, weight, bias, R, eps=1e-5):
+    return _FusedLayerNorm.apply(x, weight, bias, R, eps)

@@ -505,3 +505,27 @@ def test_fused_attention_matches_torch():
         for a, b, name in [(q.grad, q2.grad, 'dq'), (k.grad, k2.grad, 'dk'),
                            (v.grad, v2.grad, 'dv')]:
             assert (a - b).abs().max().item() < 0.05, (B, S, d, name)
+
+
+@needs_gpu
+def test_fused_layernorm_matches_torch():
+    import torch.nn.functional as F
+    from heterofl_amd.ops.fused import fused_layernorm
+    torch.manual_seed(0)
+    R, B, S, E = 5, 3, 64, 256
+    x = torch.randn(R, B, S, E, device='cuda', requires_grad=True)
+    w = (torch.rand(R, E, device='cuda') + 0.5).requires_grad_()
+    b = torch.randn(R, E, device='cuda', requires_grad=True)
+    y = fused_layernorm(x, w, b, R)
+    x2 = x.detach().clone().requires_grad_(True)
+    w2 = w.detach().clone().requires_grad_(True)
+    b2 = b.detach().clone().requires_grad_(True)
+    xhat = F.layer_norm(x2, (E,), None, None, 1e-5)
+    ref = xhat * w2.view(R, 1, 1, E) + b2.view(R, 1, 1, E)
+    assert (y - ref).abs().max().item() < 1e-4
+    g = torch.randn_like(y)
+    y.backward(g)
+    ref.backward(g)
+    assert (x.grad - x2.grad).abs().max().item() < 1e-4
+    assert (w.grad - w2.grad).abs().max().item() < 2e-3
+    assert (b.grad - b2.grad).abs().max().item() < 2e-3
