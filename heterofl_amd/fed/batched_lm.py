@@ -195,8 +195,12 @@ class BatchedTransformer(nn.Module):
 def lm_masked_ce(logits, tokens, label_masks):
     """Per-client mean CE over all positions with optional vocab masking
     (reference: src/models/transformer.py:156-161).  logits (R, B, S, V),
-    tokens (R, B, S), label_masks (R, V) in {0,1} or None.  Returns (R,)."""
+    tokens (R, B, S), label_masks (R, V) in {0,1} or None.  Returns (R,).
+    On GPU the fused large-vocab kernel runs (logits stay bf16)."""
     R, B, S, V = logits.shape
+    if native_ops.use_native(logits):
+        from ..ops.fused import fused_lm_ce
+        return fused_lm_ce(logits, tokens, label_masks, R)
     if label_masks is not None:
         logits = logits.masked_fill(
             label_masks.view(R, 1, 1, V) == 0, 0)

@@ -101,7 +101,7 @@ class BatchedLMClientTrainer:
                 with torch.autocast('cuda', torch.bfloat16,
                                     enabled=self._amp):
                     logits = model(tokens)
-                losses = lm_masked_ce(logits.float(), tokens, masks)
+                losses = lm_masked_ce(logits, tokens, masks)
                 losses.sum().backward()
                 if native:
                     fopt.step(1.0, lr, cfg['momentum'], cfg['weight_decay'])

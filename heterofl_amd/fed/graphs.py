@@ -238,7 +238,7 @@ class LMGraphedStep:
         from .batched_lm import lm_masked_ce
         with torch.autocast('cuda', torch.bfloat16, enabled=self.amp):
             logits = self.model(tokens)
-        losses = lm_masked_ce(logits.float(), tokens, self.masks)
+        losses = lm_masked_ce(logits, tokens, self.masks)
         with torch.no_grad():
             n = tokens.size(1) * tokens.size(2)
             self.metrics[:, 0] += losses.detach() * n

@@ -49,6 +49,10 @@ std::vector<at::Tensor> ln_fwd(at::Tensor x, at::Tensor gamma,
                                at::Tensor beta, int64_t R, double eps);
 std::vector<at::Tensor> ln_bwd(at::Tensor dy, at::Tensor x, at::Tensor gamma,
                                at::Tensor mean, at::Tensor invstd, int64_t R);
+std::vector<at::Tensor> lm_ce_fwd(at::Tensor logits, at::Tensor labels,
+                                  at::Tensor mask, int64_t R);
+at::Tensor lm_ce_bwd(at::Tensor logits, at::Tensor labels, at::Tensor mask,
+                     at::Tensor row_lse, at::Tensor up, int64_t R);
 std::vector<at::Tensor> head_bwd(at::Tensor dscores, at::Tensor pooled,
                                  at::Tensor w, int64_t R, int64_t H,
                                  int64_t W, bool bf16_feat, bool want_db);
@@ -76,4 +80,6 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("attn_bwd", &attn_bwd, "fused small-S attention backward");
     m.def("ln_fwd", &ln_fwd, "fused per-client LayerNorm forward");
     m.def("ln_bwd", &ln_bwd, "fused per-client LayerNorm backward");
+    m.def("lm_ce_fwd", &lm_ce_fwd, "fused vocab-masked LM CE forward");
+    m.def("lm_ce_bwd", &lm_ce_bwd, "fused vocab-masked LM CE backward");
 }

@@ -279,3 +279,34 @@ class _FusedLayerNorm(torch.autograd.Function):
 
 def fused_layernorm(x, weight, bias, R, eps=1e-5):
     return _FusedLayerNorm.apply(x, weight, bias, R, eps)
+
+
+class _FusedLMCE(torch.autograd.Function):
+    """Vocab-masked LM cross-entropy over (R, B, S, V) bf16/fp32 logits;
+    per-client mean NLL over every position (ops/csrc/lm_ce.hip)."""
+
+    @staticmethod
+    def forward(ctx, logits, tokens, mask, R):
+        ext = require_native()
+        logits = logits.contiguous()
+        labels = tokens.reshape(-1)
+        losses, row_lse = ext.lm_ce_fwd(
+            logits, labels, mask if mask is not None else torch.Tensor(), R)
+        ctx.save_for_backward(logits, labels, row_lse,
+                              mask if mask is not None else torch.Tensor())
+        ctx.R = R
+        ctx.has_mask = mask is not None
+        return losses
+
+    @staticmethod
+    def backward(ctx, up):
+        ext = require_native()
+        logits, labels, row_lse, mask = ctx.saved_tensors
+        d = ext.lm_ce_bwd(logits, labels,
+                          mask if ctx.has_mask else torch.Tensor(), row_lse,
+                          up, ctx.R)
+        return d, None, None, None
+
+
+def fused_lm_ce(logits, tokens, mask, R):
+    return _FusedLMCE.apply(logits, tokens, mask, R)

@@ -529,3 +529,25 @@ def test_fused_layernorm_matches_torch():
     assert (x.grad - x2.grad).abs().max().item() < 1e-4
     assert (w.grad - w2.grad).abs().max().item() < 2e-3
     assert (b.grad - b2.grad).abs().max().item() < 2e-3
+
+
+@needs_gpu
+def test_fused_lm_ce_matches_torch():
+    import torch.nn.functional as F
+    from heterofl_amd.ops.fused import fused_lm_ce
+    torch.manual_seed(0)
+    R, B, S, V = 4, 2, 16, 1000
+    logits = torch.randn(R, B, S, V, device='cuda', requires_grad=True)
+    tokens = torch.randint(0, V, (R, B, S), device='cuda')
+    mask = (torch.rand(R, V, device='cuda') > 0.2).float()
+    for r in range(R):
+        mask[r].scatter_(0, tokens[r].reshape(-1), 1.0)
+    losses = fused_lm_ce(logits, tokens, mask, R)
+    l2 = logits.detach().clone().requires_grad_(True)
+    masked = l2.masked_fill(mask.view(R, 1, 1, V) == 0, 0)
+    logp = F.log_softmax(masked, dim=-1)
+    ref = -logp.gather(3, tokens.unsqueeze(3)).squeeze(3).reshape(R, -1).mean(1)
+    assert (losses - ref).abs().max().item() < 1e-5
+    losses.sum().backward()
+    ref.sum().backward()
+    assert (logits.grad - l2.grad).abs().max().item() < 1e-5
