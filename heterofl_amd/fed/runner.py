@@ -165,7 +165,15 @@ class FedRunner:
                                     track=True).to(cfg['device'])
             test_model.load_state_dict(self.global_model.state_dict(), strict=False)
             test_model.train(True)
-            loader = make_data_loader({'train': self.dataset['train']}, cfg)['train']
+            # BN momentum=None is a cumulative average over batches, which is
+            # batch-size invariant for equal-size batches — so the stats pass
+            # can use large batches on GPU instead of the train batch of 10
+            # (5000 launches -> ~100 per round)
+            scfg = dict(cfg)
+            if torch.cuda.is_available():
+                scfg['batch_size'] = dict(cfg['batch_size'])
+                scfg['batch_size']['train'] = 500
+            loader = make_data_loader({'train': self.dataset['train']}, scfg)['train']
             for input in loader:
                 input = collate(input)
                 input = to_device(input, cfg['device'])
