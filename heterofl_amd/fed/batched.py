@@ -491,11 +491,32 @@ class BatchedClientTrainer:
             sched = tuple([B] * (n // B) + ([n % B] if n % B else []))
             groups.setdefault((model_rate[u], sched), []).append(m)
         out = []
+        # optional: split each rate group into HETEROFL_GROUP_SPLIT
+        # subgroups so independent client chains replay on extra concurrent
+        # streams (fills the chip when one group's kernels underfill it)
+        import os as _os
+        gsplit = int(_os.environ.get('HETEROFL_GROUP_SPLIT', '1'))
+        if gsplit > 1 and self.device.type == 'cuda':
+            split_groups = {}
+            for (rate, sched), slots in groups.items():
+                if len(slots) >= 2 * gsplit:
+                    per = (len(slots) + gsplit - 1) // gsplit
+                    for si in range(0, len(slots), per):
+                        split_groups[(rate, sched, si)] = slots[si:si + per]
+                else:
+                    split_groups[(rate, sched, 0)] = slots
+            groups = {(r, s): v for (r, s, _), v in split_groups.items()} \
+                if len(split_groups) == len(groups) else None
+            if groups is None:
+                groups = {}
+                for (rate, sched, si), slots in split_groups.items():
+                    groups[(rate, sched, si)] = slots
         # graph-eligible groups are prepared on the default stream, then
         # their replay trains run CONCURRENTLY on per-group HIP streams (the
         # rate-e group's tiny latency-bound steps hide under rate-a's)
         graphed = []
-        for (rate, sched), slots in groups.items():
+        for key, slots in groups.items():
+            rate, sched = key[0], key[1]
             use_graph = (self.device.type == 'cuda'
                          and cfg.get('hip_graphs', True)
                          and len(set(sched)) == 1)
@@ -505,6 +526,7 @@ class BatchedClientTrainer:
                 out.extend(self._train_group(rate, sched, slots, user_idx,
                                              local_parameters, label_split,
                                              lr, logger))
+        del groups
         if graphed:
             preps = [self._prepare_graphed(rate, sched, slots, user_idx,
                                            [local_parameters[m]

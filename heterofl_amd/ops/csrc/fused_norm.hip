@@ -62,7 +62,9 @@ bn_relu_fwd_kernel(const T* __restrict__ x, const float* __restrict__ gamma,
 
 // ----------------------------------------------------------------- backward
 // relu mask from pre = xhat*g+b; dx = invstd*g*(dym - s1/M - xhat*s2/M).
-template <typename T>
+// STAGE: the channel's x and (relu-masked) dy are kept in LDS from the
+// reduction pass so the dx pass reads LDS, not global.
+template <typename T, bool STAGE>
 __global__ void __launch_bounds__(256)
 bn_relu_bwd_kernel(const T* __restrict__ dy, const T* __restrict__ x,
                    const float* __restrict__ gamma, const float* __restrict__ beta,
@@ -71,6 +73,9 @@ bn_relu_bwd_kernel(const T* __restrict__ dy, const T* __restrict__ x,
                    float* __restrict__ dbeta, int N, int C, int HW) {
     const int c = blockIdx.x;
     __shared__ float scratch[2 * 256 / WAVE];
+    extern __shared__ __attribute__((aligned(16))) char smem[];
+    T* sx = (T*)smem;
+    T* sd = sx + (STAGE ? N * HW : 0);
     const long chan_off = (long)c * HW;
     const long samp_stride = (long)C * HW;
     const float m = mean[c], is = invstd[c];
@@ -80,9 +85,14 @@ bn_relu_bwd_kernel(const T* __restrict__ dy, const T* __restrict__ x,
     for (int i = threadIdx.x; i < N * HW; i += blockDim.x) {
         const int n = i / HW, hw = i - n * HW;
         const long off = n * samp_stride + chan_off + hw;
-        const float xh = (ld_f32(x + off) - m) * is;
+        const T xr = x[off];
+        const float xh = ((float)xr - m) * is;
         const float pre = xh * g + b;
         const float d = pre > 0.f ? ld_f32(dy + off) : 0.f;
+        if (STAGE) {
+            sx[i] = xr;
+            sd[i] = (T)d;
+        }
         s1 += d;
         s2 += d * xh;
     }
@@ -97,9 +107,15 @@ bn_relu_bwd_kernel(const T* __restrict__ dy, const T* __restrict__ x,
     for (int i = threadIdx.x; i < N * HW; i += blockDim.x) {
         const int n = i / HW, hw = i - n * HW;
         const long off = n * samp_stride + chan_off + hw;
-        const float xh = (ld_f32(x + off) - m) * is;
-        const float pre = xh * g + b;
-        const float d = pre > 0.f ? ld_f32(dy + off) : 0.f;
+        float xh, d;
+        if (STAGE) {
+            xh = ((float)sx[i] - m) * is;
+            d = (float)sd[i];
+        } else {
+            xh = (ld_f32(x + off) - m) * is;
+            const float pre = xh * g + b;
+            d = pre > 0.f ? ld_f32(dy + off) : 0.f;
+        }
         st_f32(dx + off, gis * (d - k1 - xh * k2));
     }
 }
@@ -248,16 +264,36 @@ std::vector<at::Tensor> bn_relu_bwd(at::Tensor dy, at::Tensor x,
     auto dbeta = at::empty({C}, opts);
     auto dyc = dy.contiguous();
     auto stream = at::hip::getCurrentHIPStream();
+    const long stage_bytes = 2L * N * (x.numel() / (N * C)) * x.element_size();
     DISPATCH_FT(x.scalar_type(), {
-        hipLaunchKernelGGL(bn_relu_bwd_kernel<scalar_t>, dim3(C), dim3(256), 0,
-                           stream,
-                           (const scalar_t*)dyc.data_ptr(),
-                           (const scalar_t*)x.data_ptr(),
-                           gamma.defined() ? gamma.data_ptr<float>() : nullptr,
-                           beta.defined() ? beta.data_ptr<float>() : nullptr,
-                           mean.data_ptr<float>(), invstd.data_ptr<float>(),
-                           (scalar_t*)dx.data_ptr(), dgamma.data_ptr<float>(),
-                           dbeta.data_ptr<float>(), N, C, HW);
+        if (stage_bytes <= 64 * 1024)
+            hipLaunchKernelGGL((bn_relu_bwd_kernel<scalar_t, true>), dim3(C),
+                               dim3(256), (int)stage_bytes, stream,
+                               (const scalar_t*)dyc.data_ptr(),
+                               (const scalar_t*)x.data_ptr(),
+                               gamma.defined() ? gamma.data_ptr<float>()
+                                               : nullptr,
+                               beta.defined() ? beta.data_ptr<float>()
+                                              : nullptr,
+                               mean.data_ptr<float>(),
+                               invstd.data_ptr<float>(),
+                               (scalar_t*)dx.data_ptr(),
+                               dgamma.data_ptr<float>(),
+                               dbeta.data_ptr<float>(), N, C, HW);
+        else
+            hipLaunchKernelGGL((bn_relu_bwd_kernel<scalar_t, false>), dim3(C),
+                               dim3(256), 0, stream,
+                               (const scalar_t*)dyc.data_ptr(),
+                               (const scalar_t*)x.data_ptr(),
+                               gamma.defined() ? gamma.data_ptr<float>()
+                                               : nullptr,
+                               beta.defined() ? beta.data_ptr<float>()
+                                              : nullptr,
+                               mean.data_ptr<float>(),
+                               invstd.data_ptr<float>(),
+                               (scalar_t*)dx.data_ptr(),
+                               dgamma.data_ptr<float>(),
+                               dbeta.data_ptr<float>(), N, C, HW);
     });
     return {dx, dgamma, dbeta};
 }
