@@ -526,7 +526,6 @@ class BatchedClientTrainer:
                 out.extend(self._train_group(rate, sched, slots, user_idx,
                                              local_parameters, label_split,
                                              lr, logger))
-        del groups
         if graphed:
             preps = [self._prepare_graphed(rate, sched, slots, user_idx,
                                            [local_parameters[m]

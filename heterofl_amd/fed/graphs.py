@@ -20,12 +20,10 @@ Graphs are cached per (rate, R, batch, lr, n_steps_capacity); weights,
 momentum and the label masks live in stable buffers the graph reads, so a
 new round only repacks buffers and replays.
 """
-import math
-
 import torch
 
 from .. import ops as native_ops
-from .batched import batched_masked_ce, per_client_clip_
+from .batched import batched_masked_ce
 
 
 class GraphedGroupStep:
