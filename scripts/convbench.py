@@ -56,14 +56,14 @@ def main():
         rows = [
             ('fwd',
              lambda: ext.conv_fwd(x, w, torch.Tensor(), torch.Tensor(),
-                                  G, s, p),
+                                  G, s, p, 0),
              lambda: F.conv2d(x, wb, None, s, p, 1, G)),
             ('bwdD',
-             lambda: ext.conv_bwd_data(dy, w, G, s, p, H, H),
+             lambda: ext.conv_bwd_data(dy, w, G, s, p, H, H, 0),
              lambda: torch.nn.grad.conv2d_input(
                  (N, G * Cin, H, H), wb, dy, (s, s), (p, p), (1, 1), G)),
             ('bwdW',
-             lambda: ext.conv_bwd_weight(dy, x, G, s, p, k),
+             lambda: ext.conv_bwd_weight(dy, x, G, s, p, k, 0),
              lambda: torch.nn.grad.conv2d_weight(
                  x, (G * Cout, Cin, k, k), dy, (s, s), (p, p), (1, 1), G)),
         ]
