@@ -33,7 +33,7 @@ typedef __attribute__((ext_vector_type(8))) __bf16 bf16x8;
 
 constexpr int BM = 64;
 constexpr int BP = 64;
-constexpr int BK = 64;
+constexpr int BK = 32;
 constexpr int LDK = BK + 8;  // padded k-stride for [.][k]-contiguous tiles
 
 // one 16x16 MFMA accumulation over a BK=32 K-slab, element type T
@@ -142,14 +142,6 @@ __device__ __forceinline__ f32x4 mfma_tile2<e5m2, e4m3>(const e5m2* a_row,
     return __builtin_amdgcn_mfma_f32_16x16x32_bf8_fp8(a, b, acc, 0, 0, 0);
 }
 
-// accumulate a full BK=64 K-slab: two 32-deep MFMA slabs
-template <typename TA, typename TB>
-__device__ __forceinline__ f32x4 mfma_tile64(const TA* a_row, const TB* b_col,
-                                             f32x4 acc) {
-    acc = mfma_tile2<TA, TB>(a_row, b_col, acc);
-    return mfma_tile2<TA, TB>(a_row + 32, b_col + 32, acc);
-}
-
 struct ConvGeom {
     int G;
     int N, H, W;
@@ -216,50 +208,42 @@ conv_fwd_kernel(const T* __restrict__ x, const float* __restrict__ w,
     f32x4 acc[2][2] = {};
     const int mm_a = tid >> 2, kkb = (tid & 3) * 8;
     const int pp_b = tid >> 2;
-    float va[16], vb[16];
+    float va[8], vb[8];
     // loader for one K-step into registers (2-phase: issue next tile's
-    // loads before this tile's MFMA so HBM/L2 latency hides under it);
-    // each thread owns two 8-element slabs (kkb and kkb+32) of its row
+    // loads before this tile's MFMA so HBM/L2 latency hides under it)
     auto load_regs = [&](int k0) {
         const int m = m0 + mm_a;
         const float* wrow = w + (long)(g * gm.Cout + m) * K + k0 + kkb;
 #pragma unroll
-        for (int h = 0; h < 2; ++h)
+        for (int j = 0; j < 8; ++j) {
+            const int k = k0 + kkb + j;
+            va[j] = (m < M && k < K) ? wrow[j] : 0.f;
+        }
 #pragma unroll
-            for (int j = 0; j < 8; ++j) {
-                const int k = k0 + kkb + h * 32 + j;
-                va[h * 8 + j] = (m < M && k < K) ? wrow[h * 32 + j] : 0.f;
-            }
-#pragma unroll
-        for (int h = 0; h < 2; ++h)
-#pragma unroll
-            for (int j = 0; j < 8; ++j) {
-                const int k = k0 + kkb + h * 32 + j;
-                float v = 0.f;
-                if (k < K) {
-                    const int cin = k / kk2, r = k - cin * kk2;
-                    const int kh = r / gm.khw, kw = r - kh * gm.khw;
-                    const int ih = t_ihb[pp_b] + kh, iw = t_iwb[pp_b] + kw;
-                    if (ih >= 0 && ih < gm.H && iw >= 0 && iw < gm.W)
-                        v = ld_f32(x + t_xbase[pp_b] + (long)cin * HW
+        for (int j = 0; j < 8; ++j) {
+            const int k = k0 + kkb + j;
+            vb[j] = 0.f;
+            if (k < K) {
+                const int cin = k / kk2, r = k - cin * kk2;
+                const int kh = r / gm.khw, kw = r - kh * gm.khw;
+                const int ih = t_ihb[pp_b] + kh, iw = t_iwb[pp_b] + kw;
+                if (ih >= 0 && ih < gm.H && iw >= 0 && iw < gm.W)
+                    vb[j] = ld_f32(x + t_xbase[pp_b] + (long)cin * HW
                                    + ih * gm.W + iw);
-                }
-                vb[h * 8 + j] = v;
             }
+        }
     };
     load_regs(ks);
     for (int k0 = ks; k0 < ke; k0 += BK) {
         st8_lds(&a_lds[mm_a][kkb], va);
-        st8_lds(&a_lds[mm_a][kkb + 32], va + 8);
         st8_lds(&b_lds[pp_b][kkb], vb);
-        st8_lds(&b_lds[pp_b][kkb + 32], vb + 8);
         __syncthreads();
         if (k0 + BK < ke) load_regs(k0 + BK);
 #pragma unroll
         for (int fm = 0; fm < 2; ++fm)
 #pragma unroll
             for (int fp = 0; fp < 2; ++fp)
-                acc[fm][fp] = mfma_tile64<TA, TB>(
+                acc[fm][fp] = mfma_tile2<TA, TB>(
                     &a_lds[wm + fm * 16 + (l & 15)][0],
                     &b_lds[wp + fp * 16 + (l & 15)][0], acc[fm][fp]);
         __syncthreads();
@@ -356,62 +340,54 @@ conv_bwd_data_kernel(const T* __restrict__ dy, const float* __restrict__ w,
     const int je = min(J, js + njc * BK);
     f32x4 acc[2][2] = {};
     const int cc_a = tid >> 2, jjb = (tid & 3) * 8;
-    float va[16], vb[16];
+    float va[8], vb[8];
     auto load_regs = [&](int j0) {
         const int c = c0 + cc_a;
 #pragma unroll
-        for (int h = 0; h < 2; ++h)
-#pragma unroll
-            for (int j8 = 0; j8 < 8; ++j8) {
-                const int j = j0 + jjb + h * 32 + j8;
-                float v = 0.f;
-                if (c < gm.Cin && j < J) {
-                    const int cout = j / kk2, r = j - cout * kk2;
-                    v = w[(long)(g * gm.Cout + cout) * K + c * kk2 + r];
-                }
-                va[h * 8 + j8] = v;
+        for (int j8 = 0; j8 < 8; ++j8) {
+            const int j = j0 + jjb + j8;
+            va[j8] = 0.f;
+            if (c < gm.Cin && j < J) {
+                const int cout = j / kk2, r = j - cout * kk2;
+                va[j8] = w[(long)(g * gm.Cout + cout) * K + c * kk2 + r];
             }
+        }
 #pragma unroll
-        for (int h = 0; h < 2; ++h)
-#pragma unroll
-            for (int j8 = 0; j8 < 8; ++j8) {
-                const int j = j0 + jjb + h * 32 + j8;
-                float v = 0.f;
-                if (j < J) {
-                    const int cout = j / kk2, r = j - cout * kk2;
-                    const int kh = r / gm.khw, kw = r - kh * gm.khw;
-                    const int ohs = t_oh[cc_a] - kh, ows = t_ow[cc_a] - kw;
-                    if (ohs >= 0 && ows >= 0) {
-                        if (gm.stride == 1) {
-                            if (ohs < gm.OH && ows < gm.OW)
-                                v = ld_f32(dy + t_dybase[cc_a]
-                                           + (long)cout * OHW
-                                           + ohs * gm.OW + ows);
-                        } else if ((ohs & 1) == 0 && (ows & 1) == 0) {
-                            const int oh = ohs >> 1, ow = ows >> 1;
-                            if (oh < gm.OH && ow < gm.OW)
-                                v = ld_f32(dy + t_dybase[cc_a]
-                                           + (long)cout * OHW
-                                           + oh * gm.OW + ow);
-                        }
+        for (int j8 = 0; j8 < 8; ++j8) {
+            const int j = j0 + jjb + j8;
+            vb[j8] = 0.f;
+            if (j < J) {
+                const int cout = j / kk2, r = j - cout * kk2;
+                const int kh = r / gm.khw, kw = r - kh * gm.khw;
+                const int ohs = t_oh[cc_a] - kh, ows = t_ow[cc_a] - kw;
+                if (ohs >= 0 && ows >= 0) {
+                    if (gm.stride == 1) {
+                        if (ohs < gm.OH && ows < gm.OW)
+                            vb[j8] = ld_f32(dy + t_dybase[cc_a]
+                                            + (long)cout * OHW
+                                            + ohs * gm.OW + ows);
+                    } else if ((ohs & 1) == 0 && (ows & 1) == 0) {
+                        const int oh = ohs >> 1, ow = ows >> 1;
+                        if (oh < gm.OH && ow < gm.OW)
+                            vb[j8] = ld_f32(dy + t_dybase[cc_a]
+                                            + (long)cout * OHW
+                                            + oh * gm.OW + ow);
                     }
                 }
-                vb[h * 8 + j8] = v;
             }
+        }
     };
     if (js < je) load_regs(js);
     for (int j0 = js; j0 < je; j0 += BK) {
         st8_lds(&a_lds[cc_a][jjb], va);
-        st8_lds(&a_lds[cc_a][jjb + 32], va + 8);
         st8_lds(&b_lds[cc_a][jjb], vb);
-        st8_lds(&b_lds[cc_a][jjb + 32], vb + 8);
         __syncthreads();
         if (j0 + BK < je) load_regs(j0 + BK);
 #pragma unroll
         for (int fm = 0; fm < 2; ++fm)
 #pragma unroll
             for (int fp = 0; fp < 2; ++fp)
-                acc[fm][fp] = mfma_tile64<TA, TB>(
+                acc[fm][fp] = mfma_tile2<TA, TB>(
                     &a_lds[wm + fm * 16 + (l & 15)][0],
                     &b_lds[wp + fp * 16 + (l & 15)][0], acc[fm][fp]);
         __syncthreads();
@@ -512,55 +488,46 @@ conv_bwd_data_s2_kernel(const T* __restrict__ dy, const float* __restrict__ w,
     const int je = min(J, js + njc * BK);
     f32x4 acc[2][2] = {};
     const int cc_a = tid >> 2, jjb = (tid & 3) * 8;
-    float va[16], vb[16];
+    float va[8], vb[8];
     auto load_regs = [&](int j0) {
         const int c = c0 + cc_a;
 #pragma unroll
-        for (int h = 0; h < 2; ++h)
-#pragma unroll
-            for (int j8 = 0; j8 < 8; ++j8) {
-                const int j = j0 + jjb + h * 32 + j8;
-                float v = 0.f;
-                if (c < gm.Cin && j < J) {
-                    const int cout = j / tap2, rr = j - cout * tap2;
-                    const int kh = lkh[rr / nkw];
-                    const int kw = lkw[rr - (rr / nkw) * nkw];
-                    v = w[(long)(g * gm.Cout + cout) * K + c * kk2
-                          + kh * gm.khw + kw];
-                }
-                va[h * 8 + j8] = v;
+        for (int j8 = 0; j8 < 8; ++j8) {
+            const int j = j0 + jjb + j8;
+            va[j8] = 0.f;
+            if (c < gm.Cin && j < J) {
+                const int cout = j / tap2, rr = j - cout * tap2;
+                const int kh = lkh[rr / nkw], kw = lkw[rr - (rr / nkw) * nkw];
+                va[j8] = w[(long)(g * gm.Cout + cout) * K + c * kk2
+                           + kh * gm.khw + kw];
             }
+        }
 #pragma unroll
-        for (int h = 0; h < 2; ++h)
-#pragma unroll
-            for (int j8 = 0; j8 < 8; ++j8) {
-                const int j = j0 + jjb + h * 32 + j8;
-                float v = 0.f;
-                if (j < J) {
-                    const int cout = j / tap2, rr = j - cout * tap2;
-                    const int a = rr / nkw, b = rr - a * nkw;
-                    const int oh = t_oh2[cc_a] + skh[a];
-                    const int ow = t_ow2[cc_a] + skw[b];
-                    if (oh >= 0 && oh < gm.OH && ow >= 0 && ow < gm.OW)
-                        v = ld_f32(dy + t_dybase[cc_a] + (long)cout * OHW
-                                   + oh * gm.OW + ow);
-                }
-                vb[h * 8 + j8] = v;
+        for (int j8 = 0; j8 < 8; ++j8) {
+            const int j = j0 + jjb + j8;
+            vb[j8] = 0.f;
+            if (j < J) {
+                const int cout = j / tap2, rr = j - cout * tap2;
+                const int a = rr / nkw, b = rr - a * nkw;
+                const int oh = t_oh2[cc_a] + skh[a];
+                const int ow = t_ow2[cc_a] + skw[b];
+                if (oh >= 0 && oh < gm.OH && ow >= 0 && ow < gm.OW)
+                    vb[j8] = ld_f32(dy + t_dybase[cc_a] + (long)cout * OHW
+                                    + oh * gm.OW + ow);
             }
+        }
     };
     if (js < je) load_regs(js);
     for (int j0 = js; j0 < je; j0 += BK) {
         st8_lds(&a_lds[cc_a][jjb], va);
-        st8_lds(&a_lds[cc_a][jjb + 32], va + 8);
         st8_lds(&b_lds[cc_a][jjb], vb);
-        st8_lds(&b_lds[cc_a][jjb + 32], vb + 8);
         __syncthreads();
         if (j0 + BK < je) load_regs(j0 + BK);
 #pragma unroll
         for (int fm = 0; fm < 2; ++fm)
 #pragma unroll
             for (int fp = 0; fp < 2; ++fp)
-                acc[fm][fp] = mfma_tile64<TA, TB>(
+                acc[fm][fp] = mfma_tile2<TA, TB>(
                     &a_lds[wm + fm * 16 + (l & 15)][0],
                     &b_lds[wp + fp * 16 + (l & 15)][0], acc[fm][fp]);
         __syncthreads();
@@ -653,50 +620,42 @@ conv_bwd_weight_kernel(const T* __restrict__ dy, const T* __restrict__ x,
             }
         }
         __syncthreads();
-        {   // a tile rows: 64 rows x 64 p; thread owns two 8-p slabs
+        {   // a tile rows: 64 rows x 32 p; thread owns 8 contiguous p
             const int mm = (tid >> 2) & 63, ppb = (tid & 3) * 8;
             const int m = m0 + mm;
             float v[8];
 #pragma unroll
-            for (int h = 0; h < 2; ++h) {
-#pragma unroll
-                for (int j = 0; j < 8; ++j) {
-                    const int pp = ppb + h * 32 + j;
-                    v[j] = (m < M && t_dyb[pp] >= 0)
-                               ? ld_f32(dy + t_dyb[pp] + (long)m * OHW)
-                               : 0.f;
-                }
-                st8_lds(&a_lds[mm][ppb + h * 32], v);
+            for (int j = 0; j < 8; ++j) {
+                const int pp = ppb + j;
+                v[j] = (m < M && t_dyb[pp] >= 0)
+                           ? ld_f32(dy + t_dyb[pp] + (long)m * OHW) : 0.f;
             }
+            st8_lds(&a_lds[mm][ppb], v);
         }
         {
             const int kk = (tid >> 2) & 63, ppb = (tid & 3) * 8;
             const int k = k0 + kk;
             float v[8];
 #pragma unroll
-            for (int h = 0; h < 2; ++h) {
-#pragma unroll
-                for (int j = 0; j < 8; ++j) {
-                    const int pp = ppb + h * 32 + j;
-                    v[j] = 0.f;
-                    if (k < K) {
-                        const int ih = t_ihb[pp] + t_kkh[kk];
-                        const int iw = t_iwb[pp] + t_kkw[kk];
-                        if (ih >= 0 && ih < gm.H && iw >= 0 && iw < gm.W)
-                            v[j] = ld_f32(x + t_xb[pp]
-                                          + (long)t_kcin[kk] * HW
-                                          + ih * gm.W + iw);
-                    }
+            for (int j = 0; j < 8; ++j) {
+                const int pp = ppb + j;
+                v[j] = 0.f;
+                if (k < K) {
+                    const int ih = t_ihb[pp] + t_kkh[kk];
+                    const int iw = t_iwb[pp] + t_kkw[kk];
+                    if (ih >= 0 && ih < gm.H && iw >= 0 && iw < gm.W)
+                        v[j] = ld_f32(x + t_xb[pp] + (long)t_kcin[kk] * HW
+                                      + ih * gm.W + iw);
                 }
-                st8_lds(&b_lds[kk][ppb + h * 32], v);
             }
+            st8_lds(&b_lds[kk][ppb], v);
         }
         __syncthreads();
 #pragma unroll
         for (int fm = 0; fm < 2; ++fm)
 #pragma unroll
             for (int fp = 0; fp < 2; ++fp)
-                acc[fm][fp] = mfma_tile64<TA, TB>(
+                acc[fm][fp] = mfma_tile2<TA, TB>(
                     &a_lds[wm + fm * 16 + (l & 15)][0],
                     &b_lds[wp + fp * 16 + (l & 15)][0], acc[fm][fp]);
         __syncthreads();
