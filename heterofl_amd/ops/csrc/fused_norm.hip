@@ -159,13 +159,16 @@ gn_relu_fwd_kernel(const T* __restrict__ x, const float* __restrict__ gamma,
     }
 }
 
+// dgamma/dbeta go to per-(n,g) partials (pgamma/pbeta, shape (N, C)),
+// reduced over n in fixed order by gn_grad_reduce_kernel — deterministic
+// under hipGraph replay (no atomics).
 template <typename T>
 __global__ void __launch_bounds__(256)
 gn_relu_bwd_kernel(const T* __restrict__ dy, const T* __restrict__ x,
                    const float* __restrict__ gamma, const float* __restrict__ beta,
                    const float* __restrict__ mean, const float* __restrict__ invstd,
-                   T* __restrict__ dx, float* __restrict__ dgamma,
-                   float* __restrict__ dbeta, int N, int C, int HW, int G) {
+                   T* __restrict__ dx, float* __restrict__ pgamma,
+                   float* __restrict__ pbeta, int N, int C, int HW, int G) {
     const int ng = blockIdx.x;
     const int n = ng / G, j = ng - n * G;
     const int cpg = C / G;
@@ -173,8 +176,25 @@ gn_relu_bwd_kernel(const T* __restrict__ dy, const T* __restrict__ x,
     const long base = (long)n * C * HW + (long)j * cpg * HW;
     const int M = cpg * HW;
     const float m = mean[ng], is = invstd[ng];
-    // pass 1: s1 = sum(dpre*g), s2 = sum(dpre*g*xhat); also atomically
-    // accumulate per-channel dgamma/dbeta (summed over N by atomics).
+    // per-channel partials: one thread per channel, serial over HW
+    if (pgamma) {
+        for (int cc = threadIdx.x; cc < cpg; cc += blockDim.x) {
+            const int c = j * cpg + cc;
+            const float g = gamma ? gamma[c] : 1.f;
+            const float b = beta ? beta[c] : 0.f;
+            float sg = 0.f, sb = 0.f;
+            for (int hw = 0; hw < HW; ++hw) {
+                const long off = base + (long)cc * HW + hw;
+                const float xh = (ld_f32(x + off) - m) * is;
+                const float pre = xh * g + b;
+                const float d = pre > 0.f ? ld_f32(dy + off) : 0.f;
+                sg += d * xh;
+                sb += d;
+            }
+            pgamma[(long)n * C + c] = sg;
+            pbeta[(long)n * C + c] = sb;
+        }
+    }
     float s1 = 0.f, s2 = 0.f;
     for (int i = threadIdx.x; i < M; i += blockDim.x) {
         const int c = j * cpg + i / HW;
@@ -183,10 +203,6 @@ gn_relu_bwd_kernel(const T* __restrict__ dy, const T* __restrict__ x,
         const float xh = (ld_f32(x + base + i) - m) * is;
         const float pre = xh * g + b;
         const float d = pre > 0.f ? ld_f32(dy + base + i) : 0.f;
-        if (dgamma) {
-            atomicAdd(dgamma + c, d * xh);
-            atomicAdd(dbeta + c, d);
-        }
         s1 += d * g;
         s2 += d * g * xh;
     }
@@ -322,6 +338,22 @@ std::vector<at::Tensor> gn_relu_fwd(at::Tensor x, at::Tensor gamma,
     return {y, mean, invstd};
 }
 
+__global__ void __launch_bounds__(256)
+gn_grad_reduce_kernel(const float* __restrict__ pgamma,
+                      const float* __restrict__ pbeta,
+                      float* __restrict__ dgamma, float* __restrict__ dbeta,
+                      int N, int C) {
+    const int c = blockIdx.x * blockDim.x + threadIdx.x;
+    if (c >= C) return;
+    float sg = 0.f, sb = 0.f;
+    for (int n = 0; n < N; ++n) {
+        sg += pgamma[(long)n * C + c];
+        sb += pbeta[(long)n * C + c];
+    }
+    dgamma[c] = sg;
+    dbeta[c] = sb;
+}
+
 std::vector<at::Tensor> gn_relu_bwd(at::Tensor dy, at::Tensor x,
                                     at::Tensor gamma, at::Tensor beta,
                                     at::Tensor mean, at::Tensor invstd,
@@ -330,8 +362,10 @@ std::vector<at::Tensor> gn_relu_bwd(at::Tensor dy, at::Tensor x,
     const int HW = x.numel() / (N * C);
     auto dx = at::empty_like(x);
     auto opts = x.options().dtype(at::kFloat);
-    auto dgamma = at::zeros({C}, opts);
-    auto dbeta = at::zeros({C}, opts);
+    auto pgamma = at::empty({N, C}, opts);
+    auto pbeta = at::empty({N, C}, opts);
+    auto dgamma = at::empty({C}, opts);
+    auto dbeta = at::empty({C}, opts);
     auto dyc = dy.contiguous();
     auto stream = at::hip::getCurrentHIPStream();
     DISPATCH_FT(x.scalar_type(), {
@@ -342,8 +376,12 @@ std::vector<at::Tensor> gn_relu_bwd(at::Tensor dy, at::Tensor x,
                            gamma.defined() ? gamma.data_ptr<float>() : nullptr,
                            beta.defined() ? beta.data_ptr<float>() : nullptr,
                            mean.data_ptr<float>(), invstd.data_ptr<float>(),
-                           (scalar_t*)dx.data_ptr(), dgamma.data_ptr<float>(),
-                           dbeta.data_ptr<float>(), N, C, HW, (int)G);
+                           (scalar_t*)dx.data_ptr(), pgamma.data_ptr<float>(),
+                           pbeta.data_ptr<float>(), N, C, HW, (int)G);
     });
+    hipLaunchKernelGGL(gn_grad_reduce_kernel, dim3((C + 255) / 256),
+                       dim3(256), 0, stream, pgamma.data_ptr<float>(),
+                       pbeta.data_ptr<float>(), dgamma.data_ptr<float>(),
+                       dbeta.data_ptr<float>(), N, C);
     return {dx, dgamma, dbeta};
 }
