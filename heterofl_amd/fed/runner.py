@@ -155,7 +155,13 @@ class FedRunner:
     def stats(self):
         """sBN statistics pass: rebuild the global model with tracked running
         stats and run the full train set in train mode under no_grad
-        (reference: src/train_classifier_fed.py:127-138)."""
+        (reference: src/train_classifier_fed.py:127-138).
+
+        Multi-rank (C2 of SURVEY §2b): the train set is sharded across ranks
+        and the cumulative running stats are combined by a batch-count
+        weighted all-reduce over RCCL — momentum=None BN keeps the running
+        stats as the mean over batches, so the weighted mean over ranks
+        reproduces the sequential pass (equal-size batches)."""
         cfg = self.cfg
         if self.is_lm:
             return self.global_model  # LM path has no sBN pass (reference:
@@ -170,14 +176,49 @@ class FedRunner:
             # can use large batches on GPU instead of the train batch of 10
             # (5000 launches -> ~100 per round)
             scfg = dict(cfg)
+            scfg['shuffle'] = dict(cfg['shuffle'])
+            scfg['shuffle']['train'] = False  # stats don't need a shuffle;
+            # unshuffled batches make the pass deterministic and identical
+            # across GPU counts
             if torch.cuda.is_available():
                 scfg['batch_size'] = dict(cfg['batch_size'])
                 scfg['batch_size']['train'] = 500
-            loader = make_data_loader({'train': self.dataset['train']}, scfg)['train']
+            train = self.dataset['train']
+            n_batches = 0
+            if self.dist_ctx is not None and self.dist_ctx.world_size > 1:
+                rank, world = self.dist_ctx.rank, self.dist_ctx.world_size
+                bs = scfg['batch_size']['train']
+                # contiguous whole-batch shards so every batch is full-size
+                total_b = len(train) // bs
+                my_b = list(range(rank, total_b, world))
+                idx = [i for b in my_b for i in range(b * bs, (b + 1) * bs)]
+                train = SplitDataset(train, idx)
+            loader = make_data_loader({'train': train}, scfg)['train']
             for input in loader:
                 input = collate(input)
                 input = to_device(input, cfg['device'])
                 test_model(input)
+                n_batches += 1
+            if self.dist_ctx is not None and self.dist_ctx.world_size > 1:
+                import torch.distributed as dist
+                bufs = []
+                for mod in test_model.modules():
+                    if hasattr(mod, 'running_mean') and                             mod.running_mean is not None:
+                        bufs.append(mod.running_mean)
+                        bufs.append(mod.running_var)
+                if bufs:
+                    w = float(n_batches)
+                    flat = torch.cat([b.reshape(-1).float() * w
+                                      for b in bufs])
+                    cnt = torch.tensor([w], device=flat.device)
+                    dist.all_reduce(flat)
+                    dist.all_reduce(cnt)
+                    flat /= cnt
+                    off = 0
+                    for b in bufs:
+                        n = b.numel()
+                        b.copy_(flat[off:off + n].view_as(b))
+                        off += n
         return test_model
 
     # ------------------------------------------------------------------- test

@@ -81,3 +81,62 @@ def _worker(rank, world, port, tmpdir):
 def test_distributed_combine_matches_sequential(tmp_path):
     port = 29541
     mp.spawn(_worker, args=(2, port, str(tmp_path)), nprocs=2, join=True)
+
+
+def _stats_worker(rank, world, port, tmpdir):
+    os.environ['MASTER_ADDR'] = '127.0.0.1'
+    os.environ['MASTER_PORT'] = str(port)
+    torch.distributed.init_process_group('gloo', rank=rank, world_size=world)
+    from heterofl_amd.config import default_config
+    from heterofl_amd.data import fetch_dataset, split_dataset
+    from heterofl_amd.fed import FedRunner
+    from heterofl_amd.models import make_model
+    from heterofl_amd.parallel import init_distributed
+    from heterofl_amd.utils import process_dataset, make_optimizer
+
+    cfg = default_config()
+    cfg['device'] = 'cpu'
+    cfg['engine'] = 'sequential'
+    cfg = make_cfg_local(cfg)
+    torch.manual_seed(0)
+    ds = fetch_dataset('MNIST', synthetic=True, synthetic_size=80)
+    process_dataset(ds, cfg)
+    torch.manual_seed(7)
+    data_split, label_split = split_dataset(ds, 4, 'iid', cfg['classes_size'])
+    torch.manual_seed(1)
+    model = make_model(cfg)
+    ctx = init_distributed(backend='gloo')
+    runner = FedRunner(cfg, ds, data_split, label_split, model,
+                       make_optimizer(model, cfg['lr'], cfg), dist_ctx=ctx)
+    tm = runner.stats()
+    # sequential oracle on one rank
+    runner1 = FedRunner(dict(cfg, world_size=1), ds, data_split, label_split,
+                        make_model(cfg),
+                        make_optimizer(model, cfg['lr'], cfg))
+    runner1.federation.global_parameters = runner.federation.global_parameters
+    runner1.global_model = runner.global_model
+    tm1 = runner1.stats()
+    for (k, v), (k1, v1) in zip(tm.state_dict().items(),
+                                tm1.state_dict().items()):
+        if 'running' in k:
+            diff = (v.float() - v1.float()).abs().max().item()
+            assert diff < 1e-4, (rank, k, diff)
+    torch.distributed.destroy_process_group()
+
+
+def make_cfg_local(cfg):
+    cfg = dict(cfg)
+    cfg['metric_name'] = {'train': {'Local': ['Local-Loss']},
+                          'test': {'Global': ['Global-Loss']}}
+    c = make_cfg(cfg, '1_4_0.5_iid_fix_a1_bn_1_1',
+                 data_name='MNIST', model_name='conv')
+    c['num_epochs'] = {'global': 1, 'local': 1}
+    c['world_size'] = 2
+    # batch 10 divides the 80-sample set -> equal-size batches
+    c['batch_size'] = {'train': 10, 'test': 50}
+    return c
+
+
+def test_distributed_sbn_stats_matches_sequential(tmp_path):
+    mp.spawn(_stats_worker, args=(2, 29551, str(tmp_path)), nprocs=2,
+             join=True)
