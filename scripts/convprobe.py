@@ -2,7 +2,8 @@
 """Run one conv shape in a loop for PMC profiling."""
 import sys
 import torch
-sys.path.insert(0, '.')
+import os
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
 from heterofl_amd.ops import require_native
 
 ext = require_native()
