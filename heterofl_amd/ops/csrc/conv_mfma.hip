@@ -716,7 +716,7 @@ at::Tensor conv_fwd(at::Tensor x, at::Tensor w, at::Tensor bias,
     const int kchunks = (K + BK - 1) / BK;
     const int tiles = ((M + BM - 1) / BM) * ((P + BP - 1) / BP) * gm.G;
     int splitk = 1;
-    while (tiles * splitk < 512 && splitk * 2 <= kchunks / 2) splitk *= 2;
+    while (tiles * splitk < 1024 && splitk * 2 <= kchunks / 2) splitk *= 2;
     dim3 grid((M + BM - 1) / BM, (P + BP - 1) / BP, gm.G * splitk);
     auto stream = at::hip::getCurrentHIPStream();
     at::Tensor partial;
@@ -799,7 +799,7 @@ at::Tensor conv_bwd_data(at::Tensor dy, at::Tensor w, int64_t groups,
     const int nyp = (qtile + BP - 1) / BP;
     const int tiles = ((gm.Cin + BM - 1) / BM) * nyp * (s2 ? 4 : 1) * gm.G;
     int splitk = 1;
-    while (tiles * splitk < 512 && splitk * 2 <= jchunks / 2) splitk *= 2;
+    while (tiles * splitk < 1024 && splitk * 2 <= jchunks / 2) splitk *= 2;
     dim3 grid((gm.Cin + BM - 1) / BM, nyp * (s2 ? 4 : 1), gm.G * splitk);
     auto stream = at::hip::getCurrentHIPStream();
     at::Tensor partial;
@@ -886,7 +886,7 @@ at::Tensor conv_bwd_weight(at::Tensor dy, at::Tensor x, int64_t groups,
     const int mk_tiles = ((gm.Cout + BM - 1) / BM) * ((K + BP - 1) / BP)
                          * gm.G;
     int splitp = 1;
-    while (mk_tiles * splitp < 512 && splitp * BK * 4 < P) splitp *= 2;
+    while (mk_tiles * splitp < 1024 && splitp * BK * 4 < P) splitp *= 2;
     auto dw = at::empty({(long)gm.G * gm.Cout, gm.Cin, gm.khw, gm.khw},
                         x.options().dtype(at::kFloat));
     auto stream = at::hip::getCurrentHIPStream();
