@@ -143,10 +143,12 @@ attn_fwd_kernel(const T* __restrict__ q, const T* __restrict__ k,
         *(bf16x8*)&b_lds[row][vb8] = tv;
     }
     __syncthreads();
-    // out = P V  (M=S rows, N=d cols, K=S): A = P [i][j], B[k=j][col]=V[j][c]
-    {
+    // out = P V  (M=S rows, N=d<=32 cols, K=S): A = P [i][j],
+    // B[k=j][col]=V[j][c].  d <= 32, so only the wp=0 waves have live
+    // output columns (cols 0..31); wp=32 waves idle here (their gathers
+    // would also run past the 40-element V row pitch).
+    if (wp == 0) {
         f32x4 acc[2][2] = {};
-        // K loop over S in two BK=32 slabs; A fragment reads p_lds rows
 #pragma unroll
         for (int kslab = 0; kslab < 2; ++kslab) {
 #pragma unroll
@@ -156,18 +158,15 @@ attn_fwd_kernel(const T* __restrict__ q, const T* __restrict__ k,
                     const int kb = (l >> 4) * 8 + kslab * 32;
                     bf16x8 a = *(const bf16x8*)
                         &p_lds[wm + fm * 16 + (l & 15)][kb];
-                    // B operand: lane needs V[kb + j][col = wp+fp*16+(l&15)]
-                    // -> gather from b_lds (V stored row-major [j][c])
+                    const int col = fp * 16 + (l & 15);  // 0..31 in-bounds
                     bf16x8 b;
 #pragma unroll
                     for (int j = 0; j < 8; ++j)
-                        b[j] = (__bf16)(float)b_lds[kb + j]
-                                               [wp + fp * 16 + (l & 15)];
+                        b[j] = (__bf16)(float)b_lds[kb + j][col];
                     acc[fm][fp] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
                         a, b, acc[fm][fp], 0, 0, 0);
                 }
         }
-        // write out: rows (l>>4)*4+r, col l&15 within quadrants; N=d<=32
 #pragma unroll
         for (int fm = 0; fm < 2; ++fm)
 #pragma unroll
@@ -175,7 +174,7 @@ attn_fwd_kernel(const T* __restrict__ q, const T* __restrict__ k,
 #pragma unroll
                 for (int r = 0; r < 4; ++r) {
                     const int i = wm + fm * 16 + (l >> 4) * 4 + r;
-                    const int c = wp + fp * 16 + (l & 15);
+                    const int c = fp * 16 + (l & 15);
                     if (i < S && c < d)
                         out[base + (long)i * d + c] = (T)acc[fm][fp][r];
                 }
@@ -291,7 +290,7 @@ attn_bwd_kernel(const T* __restrict__ dout, const T* __restrict__ q,
         }
     }
     __syncthreads();
-    {   // dQ = dS K : A = ds[i][j] rows; B[k=j][col=c] = K[j][c] gather
+    if (wp == 0) {  // dQ = dS K (d <= 32: only wp=0 cols live)
         f32x4 acc[2][2] = {};
 #pragma unroll
         for (int kslab = 0; kslab < 2; ++kslab)
@@ -321,7 +320,7 @@ attn_bwd_kernel(const T* __restrict__ dout, const T* __restrict__ q,
                         dq[base + (long)i * d + c] = (T)acc[fm][fp][r];
                 }
     }
-    {   // dK = dS^T Q : A[row=j][k=i] = dS[i][j] gather; B[k=i][col=c]=Q
+    if (wp == 0) {  // dK = dS^T Q (d <= 32)
         f32x4 acc[2][2] = {};
 #pragma unroll
         for (int kslab = 0; kslab < 2; ++kslab)
@@ -351,7 +350,7 @@ attn_bwd_kernel(const T* __restrict__ dout, const T* __restrict__ q,
                         dk[base + (long)j * d + c] = (T)acc[fm][fp][r];
                 }
     }
-    {   // dV = P^T dO : A[row=j][k=i] = P[i][j] gather; B[k=i][col=c]=dO
+    if (wp == 0) {  // dV = P^T dO (d <= 32)
         f32x4 acc[2][2] = {};
 #pragma unroll
         for (int kslab = 0; kslab < 2; ++kslab)
