@@ -687,6 +687,17 @@ splitp_reduce_kernel(const float* __restrict__ partials,
 // ------------------------------------------------------------ host layer
 #include <torch/extension.h>
 #include <ATen/hip/HIPContext.h>
+#include <cstdlib>
+
+// target concurrent-block count for the split-K/split-P decompositions
+// (tunable: HETEROFL_CONV_SPLIT_TARGET, default 1024 = 4 blocks/CU)
+static int split_target() {
+    static int t = [] {
+        const char* e = std::getenv("HETEROFL_CONV_SPLIT_TARGET");
+        return e ? std::atoi(e) : 1024;
+    }();
+    return t;
+}
 
 #define DISPATCH_CONV_FT(t, ...)                                              \
     if ((t) == at::kFloat) { using scalar_t = float; __VA_ARGS__; }           \
@@ -716,7 +727,7 @@ at::Tensor conv_fwd(at::Tensor x, at::Tensor w, at::Tensor bias,
     const int kchunks = (K + BK - 1) / BK;
     const int tiles = ((M + BM - 1) / BM) * ((P + BP - 1) / BP) * gm.G;
     int splitk = 1;
-    while (tiles * splitk < 1024 && splitk * 2 <= kchunks / 2) splitk *= 2;
+    while (tiles * splitk < split_target() && splitk * 2 <= kchunks / 2) splitk *= 2;
     dim3 grid((M + BM - 1) / BM, (P + BP - 1) / BP, gm.G * splitk);
     auto stream = at::hip::getCurrentHIPStream();
     at::Tensor partial;
@@ -799,7 +810,7 @@ at::Tensor conv_bwd_data(at::Tensor dy, at::Tensor w, int64_t groups,
     const int nyp = (qtile + BP - 1) / BP;
     const int tiles = ((gm.Cin + BM - 1) / BM) * nyp * (s2 ? 4 : 1) * gm.G;
     int splitk = 1;
-    while (tiles * splitk < 1024 && splitk * 2 <= jchunks / 2) splitk *= 2;
+    while (tiles * splitk < split_target() && splitk * 2 <= jchunks / 2) splitk *= 2;
     dim3 grid((gm.Cin + BM - 1) / BM, nyp * (s2 ? 4 : 1), gm.G * splitk);
     auto stream = at::hip::getCurrentHIPStream();
     at::Tensor partial;
@@ -886,7 +897,7 @@ at::Tensor conv_bwd_weight(at::Tensor dy, at::Tensor x, int64_t groups,
     const int mk_tiles = ((gm.Cout + BM - 1) / BM) * ((K + BP - 1) / BP)
                          * gm.G;
     int splitp = 1;
-    while (mk_tiles * splitp < 1024 && splitp * BK * 4 < P) splitp *= 2;
+    while (mk_tiles * splitp < split_target() && splitp * BK * 4 < P) splitp *= 2;
     auto dw = at::empty({(long)gm.G * gm.Cout, gm.Cin, gm.khw, gm.khw},
                         x.options().dtype(at::kFloat));
     auto stream = at::hip::getCurrentHIPStream();
