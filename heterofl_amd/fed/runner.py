@@ -77,6 +77,11 @@ class FedRunner:
                                      label_split, cfg)
         self.is_lm = cfg['model_name'] == 'transformer'
         use_batched = cfg.get('engine', 'sequential') == 'batched'
+        # the batched engine covers conv/resnet18/resnet34/transformer; the
+        # Bottleneck resnets train on the sequential engine
+        if use_batched and cfg['model_name'] in ('resnet50', 'resnet101',
+                                                 'resnet152'):
+            use_batched = False
         if use_batched and self.is_lm:
             from .batched_lm_trainer import BatchedLMClientTrainer
             self.trainer = BatchedLMClientTrainer(cfg)

@@ -144,3 +144,15 @@ def test_checkpoint_resume_roundtrip(base_cfg, tmp_path, monkeypatch):
     for key in ('cfg', 'epoch', 'data_split', 'label_split', 'model_dict',
                 'optimizer_dict', 'scheduler_dict', 'logger'):
         assert key in saved, key
+
+
+def test_resnet50_bottleneck_falls_back_sequential(base_cfg):
+    """Bottleneck resnets run on the sequential engine (batched engine
+    covers conv/resnet18/34/transformer)."""
+    cfg = make_cfg(base_cfg, '1_2_1_iid_fix_a1_bn_1_1',
+                   data_name='CIFAR10', model_name='resnet50')
+    cfg['engine'] = 'batched'
+    cfg['num_epochs'] = {'global': 1, 'local': 1}
+    runner = _run(cfg, rounds=1, n_data=20)
+    from heterofl_amd.fed.sequential import SequentialClientTrainer
+    assert isinstance(runner.trainer, SequentialClientTrainer)
