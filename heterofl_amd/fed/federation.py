@@ -159,7 +159,10 @@ class Federation:
             ptype = k.split('.')[-1]
             if 'weight' in ptype:
                 if v.dim() > 1:
-                    if 'conv1' in k or 'conv2' in k:
+                    if 'conv1' in k or 'conv2' in k or 'conv3' in k:
+                        # conv3 extends the reference rule (src/fed.py:82)
+                        # to Bottleneck blocks, which the reference's fed
+                        # path never exercised
                         if idx_i is None:
                             idx_i = Axis.full(v.size(1))
                         inp = idx_i
