@@ -124,7 +124,8 @@ def test_transformer_split_rules(base_cfg):
 
 # ---------------------------------------------------------------- transport
 @pytest.mark.parametrize('model_name,data_name', [
-    ('conv', 'MNIST'), ('resnet18', 'CIFAR10'), ('transformer', 'WikiText2')])
+    ('conv', 'MNIST'), ('resnet18', 'CIFAR10'), ('resnet50', 'CIFAR10'),
+    ('transformer', 'WikiText2')])
 def test_distribute_shapes_match_local_model(base_cfg, model_name, data_name):
     """Each distributed slice must load into a model built at that rate."""
     cfg = make_cfg(base_cfg, '1_4_1_iid_fix_a1-b1-c1-e1_bn_1_1',
