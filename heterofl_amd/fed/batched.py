@@ -608,7 +608,7 @@ class BatchedClientTrainer:
                                               bufs, R, device))
                 self._opt_cache[(rate, R)] = cached
             fopt = cached[1]
-            for b in fopt._refs[2]:
+            for b in fopt.bufs:
                 b.zero_()
             opt = None
         else:
@@ -622,8 +622,6 @@ class BatchedClientTrainer:
                 masks[i, label_split[user_idx[m]]] = 1
         shards = [self._shard(user_idx[m]) for m in slots]
         n = sum(sched)
-        steps = len(sched)
-        local_metrics = []
         for _ in range(cfg['num_epochs']['local']):
             # per-client shuffle + epoch-wide augmented gather
             xs, ys = [], []

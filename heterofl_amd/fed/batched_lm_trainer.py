@@ -76,7 +76,7 @@ class BatchedLMClientTrainer:
                                               bufs, R, device))
                 self._opt_cache[(rate, R)] = cached
             fopt = cached[1]
-            for b in fopt._refs[2]:
+            for b in fopt.bufs:
                 b.zero_()
         else:
             opt = torch.optim.SGD(params, lr=lr, momentum=cfg['momentum'],

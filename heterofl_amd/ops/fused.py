@@ -67,7 +67,6 @@ class _FusedMaskedCE(torch.autograd.Function):
     def forward(ctx, scores, labels, mask, metrics):
         ext = require_native()
         scores = scores.contiguous().float()
-        m = mask if mask is not None else torch.Tensor()
         (losses,) = ext.masked_ce_fwd(
             scores, labels.contiguous(),
             m if mask is not None else torch.Tensor(),
@@ -100,9 +99,11 @@ class FusedClipSGD:
 
     def __init__(self, params, grads, bufs, R, device):
         ext = require_native()
-        self._refs = (list(params), list(grads), list(bufs))
+        self.params = list(params)
+        self.grads = list(grads)
+        self.bufs = list(bufs)
         blob, n, clients = ext.build_chunk_table(
-            self._refs[1], self._refs[0], self._refs[2], R, self.CHUNK)
+            self.grads, self.params, self.bufs, R, self.CHUNK)
         self.table = blob
         self.n_chunks = int(n.item())
         self.chunk_client = clients
