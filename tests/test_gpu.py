@@ -596,3 +596,31 @@ def test_native_stack_converges(base_cfg):
         runner.train_round(ep)
     l1 = global_loss()
     assert l1 < l0, (l0, l1)
+
+
+@needs_gpu
+def test_noniid_dynamic_gn_round_gpu(base_cfg):
+    """BASELINE config-3 shape on one GPU: non-IID-2 dynamic a1-e1 with
+    GroupNorm on the native batched stack."""
+    from heterofl_amd.data import fetch_dataset, split_dataset
+    from heterofl_amd.fed import FedRunner
+    from heterofl_amd.models import make_model
+    from heterofl_amd.utils import process_dataset, make_optimizer
+    cfg = make_cfg(base_cfg, '1_10_0.3_non-iid-2_dynamic_a1-e1_gn_1_1',
+                   data_name='CIFAR10', model_name='resnet18')
+    cfg['device'] = 'cuda:0'
+    cfg['engine'] = 'batched'
+    cfg['compute_dtype'] = 'bfloat16'
+    cfg['num_epochs'] = {'global': 2, 'local': 1}
+    torch.manual_seed(0)
+    ds = fetch_dataset('CIFAR10', synthetic=True, synthetic_size=400)
+    process_dataset(ds, cfg)
+    data_split, label_split = split_dataset(ds, 10, 'non-iid-2', 10)
+    model = make_model(cfg).to('cuda:0')
+    opt = make_optimizer(model, cfg['lr'], cfg)
+    runner = FedRunner(cfg, ds, data_split, label_split, model, opt)
+    for ep in (1, 2):
+        runner.train_round(ep)
+    for v in runner.federation.global_parameters.values():
+        if v.is_floating_point():
+            assert torch.isfinite(v).all()
