@@ -69,7 +69,7 @@ class _FusedMaskedCE(torch.autograd.Function):
         scores = scores.contiguous().float()
         (losses,) = ext.masked_ce_fwd(
             scores, labels.contiguous(),
-            m if mask is not None else torch.Tensor(),
+            mask if mask is not None else torch.Tensor(),
             metrics if metrics is not None else torch.Tensor())
         ctx.save_for_backward(scores, labels,
                               mask if mask is not None else torch.Tensor())
