@@ -54,7 +54,7 @@ def main():
     p = argparse.ArgumentParser()
     p.add_argument('--gpus', type=int, default=1)
     p.add_argument('--steps', type=int, default=5)
-    p.add_argument('--warmup', type=int, default=2)
+    p.add_argument('--warmup', type=int, default=3)
     p.add_argument('--dtype', type=str, default='bfloat16',
                    choices=['float32', 'bfloat16', 'fp8'])
     p.add_argument('--engine', type=str, default='batched',
