@@ -162,8 +162,8 @@ conv_fwd_kernel(const T* __restrict__ x, const float* __restrict__ w,
                 const float* __restrict__ bias, const T* __restrict__ residual,
                 T* __restrict__ y, float* __restrict__ partial, ConvGeom gm,
                 int splitk) {
-    __shared__ TA a_lds[BM][LDK];
-    __shared__ TB b_lds[BP][LDK];
+    __shared__ TA a_lds[2][BM][LDK];
+    __shared__ TB b_lds[2][BP][LDK];
     __shared__ int t_ihb[BP], t_iwb[BP];
     __shared__ long t_xbase[BP], t_ybase[BP];
     const int g = blockIdx.z % gm.G;
@@ -233,20 +233,29 @@ conv_fwd_kernel(const T* __restrict__ x, const float* __restrict__ w,
             }
         }
     };
+    // double-buffered k-loop: ONE barrier per iteration; tile t+1's loads
+    // stay in flight across tile t's whole MFMA phase
     load_regs(ks);
+    st8_lds(&a_lds[0][mm_a][kkb], va);
+    st8_lds(&b_lds[0][pp_b][kkb], vb);
+    if (ks + BK < ke) load_regs(ks + BK);
+    __syncthreads();
+    int cur = 0;
     for (int k0 = ks; k0 < ke; k0 += BK) {
-        st8_lds(&a_lds[mm_a][kkb], va);
-        st8_lds(&b_lds[pp_b][kkb], vb);
-        __syncthreads();
-        if (k0 + BK < ke) load_regs(k0 + BK);
 #pragma unroll
         for (int fm = 0; fm < 2; ++fm)
 #pragma unroll
             for (int fp = 0; fp < 2; ++fp)
                 acc[fm][fp] = mfma_tile2<TA, TB>(
-                    &a_lds[wm + fm * 16 + (l & 15)][0],
-                    &b_lds[wp + fp * 16 + (l & 15)][0], acc[fm][fp]);
+                    &a_lds[cur][wm + fm * 16 + (l & 15)][0],
+                    &b_lds[cur][wp + fp * 16 + (l & 15)][0], acc[fm][fp]);
+        if (k0 + BK < ke) {
+            st8_lds(&a_lds[cur ^ 1][mm_a][kkb], va);
+            st8_lds(&b_lds[cur ^ 1][pp_b][kkb], vb);
+            if (k0 + 2 * BK < ke) load_regs(k0 + 2 * BK);
+        }
         __syncthreads();
+        cur ^= 1;
     }
     const long slab = (long)sp * gm.N * gm.G * gm.Cout * OHW;
 #pragma unroll
