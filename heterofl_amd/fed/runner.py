@@ -229,8 +229,14 @@ class FedRunner:
     # ------------------------------------------------------------------- test
     def test(self, test_model, epoch):
         """Per-client Local metrics + Global metrics
-        (reference: src/train_classifier_fed.py:141-169)."""
+        (reference: src/train_classifier_fed.py:141-169).  On GPU the eval
+        batch is enlarged (metric means are batch-size invariant under the
+        logger's count weighting) so evaluation is not launch-bound."""
         cfg = self.cfg
+        if torch.cuda.is_available() and not self.is_lm:
+            cfg = dict(cfg)
+            cfg['batch_size'] = dict(cfg['batch_size'])
+            cfg['batch_size']['test'] = max(cfg['batch_size']['test'], 500)
         metric = Metric()
         logger = self.logger
         with torch.no_grad():

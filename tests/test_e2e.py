@@ -156,3 +156,16 @@ def test_resnet50_bottleneck_falls_back_sequential(base_cfg):
     runner = _run(cfg, rounds=1, n_data=20)
     from heterofl_amd.fed.sequential import SequentialClientTrainer
     assert isinstance(runner.trainer, SequentialClientTrainer)
+
+
+@pytest.mark.parametrize('norm,scale,mask', [
+    ('none', '1', '1'), ('in', '0', '1'), ('ln', '1', '0'),
+    ('gn', '0', '0')])
+def test_ablation_combos(base_cfg, norm, scale, mask):
+    """Ablation grid (reference: src/make_ablation.py:69-88): every
+    norm/scaler/mask combination trains a round end-to-end."""
+    cfg = make_cfg(base_cfg,
+                   f'1_4_0.5_iid_fix_a1-e1_{norm}_{scale}_{mask}',
+                   data_name='CIFAR10', model_name='resnet18')
+    cfg['num_epochs'] = {'global': 1, 'local': 1}
+    _run(cfg, rounds=1, n_data=40)
