@@ -169,3 +169,36 @@ def test_ablation_combos(base_cfg, norm, scale, mask):
                    data_name='CIFAR10', model_name='resnet18')
     cfg['num_epochs'] = {'global': 1, 'local': 1}
     _run(cfg, rounds=1, n_data=40)
+
+
+def test_centralized_entry(base_cfg, tmp_path, monkeypatch):
+    """Non-fed baseline entry trains, evaluates and checkpoints
+    (reference: src/train_classifier.py)."""
+    import os
+    monkeypatch.chdir(tmp_path)
+    import heterofl_amd.entry as entry
+    from heterofl_amd.config import default_config
+    orig_pc = entry.process_control
+
+    def small_pc(cfg):
+        orig_pc(cfg)
+        cfg['num_epochs'] = 2
+
+    monkeypatch.setattr(entry, 'process_control', small_pc)
+    import heterofl_amd.data as data_mod
+    orig_fetch = data_mod.fetch_dataset
+    monkeypatch.setattr(
+        entry, 'fetch_dataset',
+        lambda name, subset=None, synthetic=False: orig_fetch(
+            name, subset, synthetic=True, synthetic_size=40))
+    cfg = default_config()
+    cfg.update({'data_name': 'MNIST', 'model_name': 'conv', 'device': 'cpu',
+                'num_experiments': 1, 'init_seed': 0, 'resume_mode': 0,
+                'batch_size': {'train': 10, 'test': 10}})
+    cfg['control_name'] = '1_4_0.5_iid_fix_a1_bn_1_1'
+    entry.run_centralized_experiment(
+        cfg, 'Accuracy', +1,
+        {'train': ['Loss', 'Accuracy'], 'test': ['Loss', 'Accuracy']})
+    tag = cfg['model_tag']
+    assert os.path.exists('./output/model/{}_checkpoint.pt'.format(tag))
+    assert os.path.exists('./output/model/{}_best.pt'.format(tag))

@@ -624,3 +624,36 @@ def test_noniid_dynamic_gn_round_gpu(base_cfg):
     for v in runner.federation.global_parameters.values():
         if v.is_floating_point():
             assert torch.isfinite(v).all()
+
+
+@needs_gpu
+def test_stats_and_eval_paths_gpu(base_cfg):
+    """sBN stats pass + Local/Global evaluation run on GPU (large-batch
+    stats/eval paths)."""
+    from heterofl_amd.data import fetch_dataset, split_dataset
+    from heterofl_amd.fed import FedRunner
+    from heterofl_amd.logger import Logger
+    from heterofl_amd.models import make_model
+    from heterofl_amd.utils import process_dataset, make_optimizer
+    import tempfile
+    cfg = make_cfg(base_cfg, '1_4_0.5_iid_fix_a1_bn_1_1',
+                   data_name='CIFAR10', model_name='resnet18')
+    cfg['device'] = 'cuda:0'
+    cfg['engine'] = 'batched'
+    cfg['num_epochs'] = {'global': 1, 'local': 1}
+    torch.manual_seed(0)
+    ds = fetch_dataset('CIFAR10', synthetic=True, synthetic_size=80)
+    process_dataset(ds, cfg)
+    data_split, label_split = split_dataset(ds, 4, 'iid', 10)
+    model = make_model(cfg).to('cuda:0')
+    opt = make_optimizer(model, cfg['lr'], cfg)
+    logger = Logger(tempfile.mkdtemp())
+    runner = FedRunner(cfg, ds, data_split, label_split, model, opt,
+                       logger=logger)
+    logger.safe(True)
+    runner.train_round(1)
+    tm = runner.stats()
+    assert any('running_mean' in k for k in tm.state_dict())
+    runner.test(tm, 1)
+    logger.safe(False)
+    assert 'test/Global-Accuracy' in logger.mean
