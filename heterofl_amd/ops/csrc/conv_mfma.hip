@@ -219,17 +219,26 @@ conv_fwd_kernel(const T* __restrict__ x, const float* __restrict__ w,
             const int k = k0 + kkb + j;
             va[j] = (m < M && k < K) ? wrow[j] : 0.f;
         }
+        // incremental k -> (cin, kh, kw): one divide per 8 contiguous k
+        // (runtime integer divides cost ~60 VALU cycles each)
+        int k = k0 + kkb;
+        int cin = k / kk2, r = k - cin * kk2;
+        int kh = r / gm.khw, kw = r - kh * gm.khw;
 #pragma unroll
         for (int j = 0; j < 8; ++j) {
-            const int k = k0 + kkb + j;
             vb[j] = 0.f;
-            if (k < K) {
-                const int cin = k / kk2, r = k - cin * kk2;
-                const int kh = r / gm.khw, kw = r - kh * gm.khw;
+            if (k + j < K) {
                 const int ih = t_ihb[pp_b] + kh, iw = t_iwb[pp_b] + kw;
                 if (ih >= 0 && ih < gm.H && iw >= 0 && iw < gm.W)
                     vb[j] = ld_f32(x + t_xbase[pp_b] + (long)cin * HW
                                    + ih * gm.W + iw);
+            }
+            if (++kw == gm.khw) {
+                kw = 0;
+                if (++kh == gm.khw) {
+                    kh = 0;
+                    ++cin;
+                }
             }
         }
     };
