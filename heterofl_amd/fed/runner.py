@@ -147,8 +147,13 @@ class FedRunner:
 
         with _phase_timer('3.combine'):
             if self.dist_ctx is None:
-                ordered = [trained[m] for m in range(len(user_idx))]
-                self.federation.combine(ordered, param_idx, user_idx)
+                # combine averages whoever reported — a client that failed
+                # mid-round simply contributes nothing (reference tolerance
+                # semantics, src/fed.py:180-298)
+                tmp_d, cnt_d = self.federation.accumulate(
+                    trained, param_idx, user_idx,
+                    slots=sorted(trained.keys()))
+                self.federation.finalize(tmp_d, cnt_d)
             else:
                 from ..parallel.dist import distributed_combine
                 distributed_combine(self.federation, trained, param_idx,

@@ -123,7 +123,7 @@ bn_relu_bwd_kernel(const T* __restrict__ dy, const T* __restrict__ x,
 // ---------------------------------------------------------------- GroupNorm
 // Stats per (sample n, group j) over the group's channels x HW
 // (covers gn=4 / ln=1 / in=C groups, reference: src/models/resnet.py:19-26).
-template <typename T>
+template <typename T, bool STAGE>
 __global__ void __launch_bounds__(256)
 gn_relu_fwd_kernel(const T* __restrict__ x, const float* __restrict__ gamma,
                    const float* __restrict__ beta, T* __restrict__ y,
@@ -133,11 +133,15 @@ gn_relu_fwd_kernel(const T* __restrict__ x, const float* __restrict__ gamma,
     const int n = ng / G, j = ng - n * G;
     const int cpg = C / G;
     __shared__ float scratch[2 * 256 / WAVE];
+    extern __shared__ __attribute__((aligned(16))) char smem[];
+    T* stage = (T*)smem;
     const long base = (long)n * C * HW + (long)j * cpg * HW;
     const int M = cpg * HW;
     float s1 = 0.f, s2 = 0.f;
     for (int i = threadIdx.x; i < M; i += blockDim.x) {
-        const float v = ld_f32(x + base + i);
+        const T raw = x[base + i];
+        if (STAGE) stage[i] = raw;
+        const float v = (float)raw;
         s1 += v;
         s2 += v * v;
     }
@@ -154,7 +158,7 @@ gn_relu_fwd_kernel(const T* __restrict__ x, const float* __restrict__ gamma,
         const int c = j * cpg + i / HW;
         const float g = gamma ? gamma[c] : 1.f;
         const float b = beta ? beta[c] : 0.f;
-        const float v = ld_f32(x + base + i);
+        const float v = STAGE ? (float)stage[i] : ld_f32(x + base + i);
         st_f32(y + base + i, fmaxf((v - mean) * invstd * g + b, 0.f));
     }
 }
@@ -325,15 +329,30 @@ std::vector<at::Tensor> gn_relu_fwd(at::Tensor x, at::Tensor gamma,
     auto mean = at::empty({N * G}, opts);
     auto invstd = at::empty({N * G}, opts);
     auto stream = at::hip::getCurrentHIPStream();
+    const long gsb = ((long)C / G) * (x.numel() / (N * C)) * x.element_size();
     DISPATCH_FT(x.scalar_type(), {
-        hipLaunchKernelGGL(gn_relu_fwd_kernel<scalar_t>, dim3(N * G), dim3(256),
-                           0, stream,
-                           (const scalar_t*)x.data_ptr(),
-                           gamma.defined() ? gamma.data_ptr<float>() : nullptr,
-                           beta.defined() ? beta.data_ptr<float>() : nullptr,
-                           (scalar_t*)y.data_ptr(), mean.data_ptr<float>(),
-                           invstd.data_ptr<float>(), N, C, HW, (int)G,
-                           (float)eps);
+        if (gsb <= 64 * 1024)
+            hipLaunchKernelGGL((gn_relu_fwd_kernel<scalar_t, true>),
+                               dim3(N * G), dim3(256), (int)gsb, stream,
+                               (const scalar_t*)x.data_ptr(),
+                               gamma.defined() ? gamma.data_ptr<float>()
+                                               : nullptr,
+                               beta.defined() ? beta.data_ptr<float>()
+                                              : nullptr,
+                               (scalar_t*)y.data_ptr(), mean.data_ptr<float>(),
+                               invstd.data_ptr<float>(), N, C, HW, (int)G,
+                               (float)eps);
+        else
+            hipLaunchKernelGGL((gn_relu_fwd_kernel<scalar_t, false>),
+                               dim3(N * G), dim3(256), 0, stream,
+                               (const scalar_t*)x.data_ptr(),
+                               gamma.defined() ? gamma.data_ptr<float>()
+                                               : nullptr,
+                               beta.defined() ? beta.data_ptr<float>()
+                                              : nullptr,
+                               (scalar_t*)y.data_ptr(), mean.data_ptr<float>(),
+                               invstd.data_ptr<float>(), N, C, HW, (int)G,
+                               (float)eps);
     });
     return {y, mean, invstd};
 }

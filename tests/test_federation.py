@@ -211,3 +211,33 @@ def test_dynamic_resampling_with_generator(base_cfg):
     fed.make_model_rate(generator=g2)
     assert r1 == fed.model_rate
     assert set(fed.model_rate) <= {1, 0.0625}
+
+
+def test_combine_tolerates_missing_clients(base_cfg):
+    """A client that never reports contributes nothing; entries only it
+    covered keep their previous global values (reference tolerance
+    semantics, src/fed.py:180-298)."""
+    import torch
+    from heterofl_amd.fed.federation import Federation
+    from heterofl_amd.models import make_model
+    from tests.conftest import make_cfg
+    cfg = make_cfg(base_cfg, '1_2_1_iid_fix_a1-e1_bn_1_1',
+                   data_name='CIFAR10', model_name='resnet18')
+    torch.manual_seed(0)
+    model = make_model(cfg, model_rate=1)
+    fed = Federation(model.state_dict(), cfg['model_rate'],
+                     {0: list(range(10)), 1: list(range(10))}, cfg)
+    user_idx = [0, 1]   # user 0 rate 1, user 1 rate 1/16
+    local, pidx = fed.distribute(user_idx)
+    before = {k: v.clone() for k, v in fed.global_parameters.items()}
+    # only the tiny client (slot 1) reports, with modified weights
+    trained = {1: {k: v + 1.0 if v.is_floating_point() else v
+                   for k, v in local[1].items()}}
+    tmp_d, cnt_d = fed.accumulate(trained, pidx, user_idx,
+                                  slots=sorted(trained.keys()))
+    fed.finalize(tmp_d, cnt_d)
+    w = fed.global_parameters['conv1.weight']
+    n1 = pidx[1]['conv1.weight'].out.n
+    # rows the tiny client covers were updated; the rest untouched
+    assert (w[:n1] - (before['conv1.weight'][:n1] + 1.0)).abs().max() < 1e-6
+    assert (w[n1:] - before['conv1.weight'][n1:]).abs().max() < 1e-6
