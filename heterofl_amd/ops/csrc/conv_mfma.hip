@@ -853,6 +853,154 @@ conv_fwd_staged_kernel(const T* __restrict__ x, const float* __restrict__ w,
         }
 }
 
+
+// ----------------------------------- staged stride-1 bwd-data (LDS window)
+// Same idea as conv_fwd_staged_kernel: a BP-pixel dx tile within one
+// sample spanning whole input rows has a dense dy window; stage it
+// coalesced and build the transposed-conv patch matrix from LDS.
+template <typename T, typename TA, typename TB>
+__global__ void __launch_bounds__(256)
+conv_bwd_data_staged_kernel(const T* __restrict__ dy,
+                            const float* __restrict__ w, T* __restrict__ dx,
+                            float* __restrict__ partial, ConvGeom gm,
+                            int splitk) {
+    __shared__ TA a_lds[BM][LDK];
+    __shared__ TB b_lds[BP][LDK];
+    __shared__ float win[WIN_CH][WIN_ROWS][WIN_W];
+    const int g = blockIdx.z % gm.G;
+    const int sp = blockIdx.z / gm.G;
+    const int c0 = blockIdx.x * BM;
+    const int q0 = blockIdx.y * BP;
+    const int kk2 = gm.khw * gm.khw;
+    const int J = gm.Cout * kk2;
+    const int K = gm.Cin * kk2;
+    const int HW = gm.H * gm.W;
+    const int OHW = gm.OH * gm.OW;
+    const int tid = threadIdx.x;
+    const int l = tid & (WAVE - 1);
+    const int wave = tid / WAVE;
+    const int wm = (wave >> 1) * 32;
+    const int wp = (wave & 1) * 32;
+    const int njc = ((J + BK - 1) / BK + splitk - 1) / splitk;
+    const int js = sp * njc * BK;
+    const int je = min(J, js + njc * BK);
+    // tile geometry (stride 1: OH=H, OW=W)
+    const int n = q0 / HW;
+    const int ih0 = (q0 - n * HW) / gm.W;
+    const int nrow_in = BP / gm.W;
+    const int rows_w = nrow_in + gm.khw - 1;
+    const int ohs_min = ih0 + gm.pad - (gm.khw - 1);
+    const long dyn = ((long)n * gm.G * gm.Cout + (long)g * gm.Cout) * OHW;
+    const int cc_a = tid >> 2, jjb = (tid & 3) * 8;
+    // this thread's pixel for the B build
+    const int iwl = (q0 + cc_a - n * HW) - (ih0 + cc_a / gm.W) * gm.W;
+    const int rowl = (cc_a / gm.W) + (gm.khw - 1);   // ih - ih0 + (khw-1)
+    float va[8];
+
+    f32x4 acc[2][2] = {};
+    for (int j0 = js; j0 < je; j0 += BK) {
+        const int co0 = j0 / kk2;
+        const int co1 = min(gm.Cout - 1, (j0 + BK - 1) / kk2);
+        const int nch = co1 - co0 + 1;
+        const int wtot = nch * rows_w * gm.OW;
+        for (int e = tid; e < wtot; e += 256) {
+            const int ww = e % gm.OW;
+            const int rr = (e / gm.OW) % rows_w;
+            const int cc = e / (gm.OW * rows_w);
+            const int ohs = ohs_min + rr;
+            win[cc][rr][ww + gm.pad] =
+                (ohs >= 0 && ohs < gm.OH)
+                    ? ld_f32(dy + dyn + (long)(co0 + cc) * OHW
+                             + ohs * gm.OW + ww)
+                    : 0.f;
+        }
+        for (int e = tid; e < nch * rows_w * gm.pad * 2; e += 256) {
+            const int side = e & 1;
+            const int pe = e >> 1;
+            const int pcol = pe % gm.pad;
+            const int rr = (pe / gm.pad) % rows_w;
+            const int cc = pe / (gm.pad * rows_w);
+            win[cc][rr][side ? gm.pad + gm.OW + pcol : pcol] = 0.f;
+        }
+        {   // A tile: w[g*Cout + cout][c*kk2 + r] rows=c
+            const int c = c0 + cc_a;
+            int j = j0 + jjb;
+            int cout = j / kk2, r = j - cout * kk2;
+            int kh = r / gm.khw, kw = r - kh * gm.khw;
+#pragma unroll
+            for (int j8 = 0; j8 < 8; ++j8) {
+                va[j8] = (c < gm.Cin && j + j8 < J)
+                             ? w[(long)(g * gm.Cout + cout) * K + c * kk2
+                                 + kh * gm.khw + kw]
+                             : 0.f;
+                if (++kw == gm.khw) {
+                    kw = 0;
+                    if (++kh == gm.khw) {
+                        kh = 0;
+                        ++cout;
+                    }
+                }
+            }
+            st8_lds(&a_lds[cc_a][jjb], va);
+        }
+        __syncthreads();
+        {   // B tile from window: row = rowl - kh, col = iwl + 2*pad? no:
+            // needed dy[ohs = ih + pad - kh][ows = iw + pad - kw];
+            // win row = ohs - ohs_min = rowl - kh; win col = ows + pad
+            //         = iwl + 2*pad - kw... stored col = ows + pad where
+            // ows = iw + pad - kw -> col = iw + 2*pad - kw
+            float vb[8];
+            int j = j0 + jjb;
+            int cout = j / kk2, r = j - cout * kk2;
+            int kh = r / gm.khw, kw = r - kh * gm.khw;
+#pragma unroll
+            for (int j8 = 0; j8 < 8; ++j8) {
+                vb[j8] = (j + j8 < J)
+                             ? win[cout - co0][rowl - kh]
+                                  [iwl + 2 * gm.pad - kw]
+                             : 0.f;
+                if (++kw == gm.khw) {
+                    kw = 0;
+                    if (++kh == gm.khw) {
+                        kh = 0;
+                        ++cout;
+                    }
+                }
+            }
+            st8_lds(&b_lds[cc_a][jjb], vb);
+        }
+        __syncthreads();
+#pragma unroll
+        for (int fm = 0; fm < 2; ++fm)
+#pragma unroll
+            for (int fp = 0; fp < 2; ++fp)
+                acc[fm][fp] = mfma_tile2<TA, TB>(
+                    &a_lds[wm + fm * 16 + (l & 15)][0],
+                    &b_lds[wp + fp * 16 + (l & 15)][0], acc[fm][fp]);
+        __syncthreads();
+    }
+    const long slab = (long)sp * gm.N * gm.G * gm.Cin * HW;
+    const long xb0 = ((long)n * gm.G * gm.Cin + (long)g * gm.Cin) * HW
+                     + (q0 - n * HW);
+#pragma unroll
+    for (int fm = 0; fm < 2; ++fm)
+#pragma unroll
+        for (int fp = 0; fp < 2; ++fp) {
+            const int qq = wp + fp * 16 + (l & 15);
+#pragma unroll
+            for (int r = 0; r < 4; ++r) {
+                const int c = c0 + wm + fm * 16 + (l >> 4) * 4 + r;
+                if (c < gm.Cin) {
+                    const long off = xb0 + qq + (long)c * HW;
+                    if (splitk == 1)
+                        st_f32(dx + off, acc[fm][fp][r]);
+                    else
+                        partial[slab + off] = acc[fm][fp][r];
+                }
+            }
+        }
+}
+
 // ------------------------------------------------------------ host layer
 #include <torch/extension.h>
 #include <ATen/hip/HIPContext.h>
@@ -1007,9 +1155,25 @@ at::Tensor conv_bwd_data(at::Tensor dy, at::Tensor w, int64_t groups,
     if (splitk > 1)
         partial = at::empty({(long)splitk * dx.numel()},
                             dy.options().dtype(at::kFloat));
+    const bool stg = !s2 && gm.stride == 1
+                     && (gm.H * gm.W) % BP == 0 && BP % gm.W == 0
+                     && (gm.OW + 2 * gm.pad) <= WIN_W - 2
+                     && (BP / gm.W + gm.khw - 1) <= WIN_ROWS
+                     && (BK / (gm.khw * gm.khw) + 2) <= WIN_CH
+                     && std::getenv("HETEROFL_CONV_NO_STAGED") == nullptr;
     DISPATCH_CONV_FT(dy.scalar_type(), {
         const bool q = fp8 && dy.scalar_type() == at::kBFloat16;
-        if (s2) {
+        if (stg && !q) {
+            hipLaunchKernelGGL((conv_bwd_data_staged_kernel<scalar_t,
+                                                            scalar_t,
+                                                            scalar_t>),
+                               grid, dim3(256), 0, stream,
+                               (const scalar_t*)dyc.data_ptr(),
+                               w.data_ptr<float>(), (scalar_t*)dx.data_ptr(),
+                               splitk > 1 ? partial.data_ptr<float>()
+                                          : nullptr,
+                               gm, splitk);
+        } else if (s2) {
             if (q)
                 hipLaunchKernelGGL((conv_bwd_data_s2_kernel<scalar_t, e4m3,
                                                             e5m2>),
