@@ -1001,6 +1001,144 @@ conv_bwd_data_staged_kernel(const T* __restrict__ dy,
         }
 }
 
+
+// -------------------------------------- staged bwd-weight (x window in LDS)
+// The k-tile (hence the input-channel span) is FIXED per block, so the x
+// window only advances rows as the P loop walks output rows: stage the
+// dense (channels x rows x width) slab per p-step and build the patch
+// tile from LDS.  Channels per 64-wide k-tile: ceil(64/9)+1 <= 9.
+constexpr int WWIN_CH = 9;
+
+template <typename T, typename TA, typename TB>
+__global__ void __launch_bounds__(256)
+conv_bwd_weight_staged_kernel(const T* __restrict__ dy,
+                              const T* __restrict__ x,
+                              float* __restrict__ out, ConvGeom gm,
+                              int splitp) {
+    __shared__ TA a_lds[BM][LDK];
+    __shared__ TB b_lds[BP][LDK];
+    __shared__ float win[WWIN_CH][WIN_ROWS][WIN_W];
+    __shared__ long t_dyb[BK];
+    const int g = blockIdx.z % gm.G;
+    const int sp = blockIdx.z / gm.G;
+    const int m0 = blockIdx.x * BM;
+    const int k0 = blockIdx.y * BP;
+    const int kk2 = gm.khw * gm.khw;
+    const int K = gm.Cin * kk2;
+    const int M = gm.Cout;
+    const int OHW = gm.OH * gm.OW;
+    const int P = gm.N * OHW;
+    const int HW = gm.H * gm.W;
+    const int tid = threadIdx.x;
+    const int l = tid & (WAVE - 1);
+    const int wave = tid / WAVE;
+    const int wm = (wave >> 1) * 32;
+    const int wp = (wave & 1) * 32;
+    const int pchunk0 = (P + splitp - 1) / splitp;
+    const int pchunk = ((pchunk0 + BK - 1) / BK) * BK;  // BK-aligned chunks
+    const int pstart = sp * pchunk;
+    const int pend = min(P, pstart + pchunk);
+    const int cin0 = k0 / kk2;
+    const int nch = min(gm.Cin - 1, (k0 + BP - 1) / kk2) - cin0 + 1;
+    const int nrow_out = BK / gm.OW;          // 32 pixels per p-step
+    const int rows_in = (nrow_out - 1) * gm.stride + gm.khw;
+    // per-thread B ownership: fixed k, 8 consecutive pixels
+    const int kk_b = (tid >> 2) & 63, ppb = (tid & 3) * 8;
+    const int k_b = k0 + kk_b;
+    int cin_b = 0, kh_b = 0, kw_b = 0;
+    if (k_b < K) {
+        cin_b = k_b / kk2;
+        const int r = k_b - cin_b * kk2;
+        kh_b = r / gm.khw;
+        kw_b = r - kh_b * gm.khw;
+    }
+    const int mm_a = tid >> 2;   // A rows (dy), 8 consecutive p via t_dyb
+
+    f32x4 acc[2][2] = {};
+    for (int p0 = pstart; p0 < pend; p0 += BK) {
+        __syncthreads();
+        const int n = p0 / OHW;
+        const int oh0 = (p0 - n * OHW) / gm.OW;
+        const int ih0 = oh0 * gm.stride - gm.pad;
+        const long xn = ((long)n * gm.G * gm.Cin + (long)g * gm.Cin) * HW;
+        if (tid < BK) {
+            const int p = p0 + tid;
+            t_dyb[tid] = (p < pend && p < P)
+                             ? ((long)n * gm.G * gm.Cout
+                                + (long)g * gm.Cout) * OHW + (p - n * OHW)
+                             : -1;
+        }
+        const int wtot = nch * rows_in * gm.W;
+        for (int e = tid; e < wtot; e += 256) {
+            const int ww = e % gm.W;
+            const int rr = (e / gm.W) % rows_in;
+            const int cc = e / (gm.W * rows_in);
+            const int ih = ih0 + rr;
+            win[cc][rr][ww + gm.pad] =
+                (ih >= 0 && ih < gm.H)
+                    ? ld_f32(x + xn + (long)(cin0 + cc) * HW + ih * gm.W + ww)
+                    : 0.f;
+        }
+        for (int e = tid; e < nch * rows_in * gm.pad * 2; e += 256) {
+            const int side = e & 1;
+            const int pe = e >> 1;
+            const int pcol = pe % gm.pad;
+            const int rr = (pe / gm.pad) % rows_in;
+            const int cc = pe / (gm.pad * rows_in);
+            win[cc][rr][side ? gm.pad + gm.W + pcol : pcol] = 0.f;
+        }
+        __syncthreads();
+        {   // A tile: dy rows (coalesced via t_dyb)
+            const int m = m0 + mm_a;
+            float v[8];
+#pragma unroll
+            for (int j = 0; j < 8; ++j) {
+                const int pp = ppb + j;
+                v[j] = (m < M && t_dyb[pp] >= 0)
+                           ? ld_f32(dy + t_dyb[pp] + (long)m * OHW) : 0.f;
+            }
+            st8_lds(&a_lds[mm_a][ppb], v);
+        }
+        {   // B tile from the window: 8 consecutive pixels of one k
+            float v[8];
+#pragma unroll
+            for (int j = 0; j < 8; ++j) {
+                const int pp = ppb + j;
+                v[j] = 0.f;
+                if (k_b < K && p0 + pp < pend) {
+                    const int ohl = (pp / gm.OW) * gm.stride;
+                    const int owp = pp - (pp / gm.OW) * gm.OW;
+                    v[j] = win[cin_b - cin0][ohl + kh_b]
+                              [owp * gm.stride + kw_b];
+                }
+            }
+            st8_lds(&b_lds[kk_b][ppb], v);
+        }
+        __syncthreads();
+#pragma unroll
+        for (int fm = 0; fm < 2; ++fm)
+#pragma unroll
+            for (int fp = 0; fp < 2; ++fp)
+                acc[fm][fp] = mfma_tile2<TA, TB>(
+                    &a_lds[wm + fm * 16 + (l & 15)][0],
+                    &b_lds[wp + fp * 16 + (l & 15)][0], acc[fm][fp]);
+    }
+    __syncthreads();
+    const long GK = (long)gm.G * gm.Cout * K;
+    float* slab = out + (long)sp * GK;
+#pragma unroll
+    for (int fm = 0; fm < 2; ++fm)
+#pragma unroll
+        for (int fp = 0; fp < 2; ++fp)
+#pragma unroll
+            for (int r = 0; r < 4; ++r) {
+                const int m = m0 + wm + fm * 16 + (l >> 4) * 4 + r;
+                const int k = k0 + wp + fp * 16 + (l & 15);
+                if (m < M && k < K)
+                    slab[(long)(g * gm.Cout + m) * K + k] = acc[fm][fp][r];
+            }
+}
+
 // ------------------------------------------------------------ host layer
 #include <torch/extension.h>
 #include <ATen/hip/HIPContext.h>
@@ -1257,11 +1395,26 @@ at::Tensor conv_bwd_weight(at::Tensor dy, at::Tensor x, int64_t groups,
     auto stream = at::hip::getCurrentHIPStream();
     dim3 grid((gm.Cout + BM - 1) / BM, (K + BP - 1) / BP, gm.G * splitp);
     const bool q = fp8 && x.scalar_type() == at::kBFloat16;
+    // staged path: 3x3 kernels (a 64-wide k-tile then spans <= 9 input
+    // channels = WWIN_CH), p-steps inside one sample spanning whole rows
+    const bool stg = gm.khw == 3 && (gm.OH * gm.OW) % BK == 0
+                     && BK % gm.OW == 0
+                     && (gm.W + 2 * gm.pad) <= WIN_W - 2
+                     && ((BK / gm.OW - 1) * gm.stride + gm.khw) <= WIN_ROWS
+                     && std::getenv("HETEROFL_CONV_NO_STAGED") == nullptr;
     if (splitp == 1) {
         DISPATCH_CONV_FT(x.scalar_type(), {
             if (q)
                 hipLaunchKernelGGL((conv_bwd_weight_kernel<scalar_t, e5m2,
                                                            e4m3>),
+                                   grid, dim3(256), 0, stream,
+                                   (const scalar_t*)dyc.data_ptr(),
+                                   (const scalar_t*)x.data_ptr(),
+                                   dw.data_ptr<float>(), gm, 1);
+            else if (stg)
+                hipLaunchKernelGGL((conv_bwd_weight_staged_kernel<scalar_t,
+                                                                  scalar_t,
+                                                                  scalar_t>),
                                    grid, dim3(256), 0, stream,
                                    (const scalar_t*)dyc.data_ptr(),
                                    (const scalar_t*)x.data_ptr(),
@@ -1280,6 +1433,14 @@ at::Tensor conv_bwd_weight(at::Tensor dy, at::Tensor x, int64_t groups,
     DISPATCH_CONV_FT(x.scalar_type(), {
         if (q)
             hipLaunchKernelGGL((conv_bwd_weight_kernel<scalar_t, e5m2, e4m3>),
+                               grid, dim3(256), 0, stream,
+                               (const scalar_t*)dyc.data_ptr(),
+                               (const scalar_t*)x.data_ptr(),
+                               partials.data_ptr<float>(), gm, splitp);
+        else if (stg)
+            hipLaunchKernelGGL((conv_bwd_weight_staged_kernel<scalar_t,
+                                                              scalar_t,
+                                                              scalar_t>),
                                grid, dim3(256), 0, stream,
                                (const scalar_t*)dyc.data_ptr(),
                                (const scalar_t*)x.data_ptr(),
