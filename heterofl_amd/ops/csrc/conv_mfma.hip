@@ -723,7 +723,7 @@ conv_fwd_staged_kernel(const T* __restrict__ x, const float* __restrict__ w,
                        float* __restrict__ partial, ConvGeom gm, int splitk) {
     __shared__ TA a_lds[BM][LDK];
     __shared__ TB b_lds[BP][LDK];
-    __shared__ float win[2][WIN_CH][WIN_ROWS][WIN_W];
+    __shared__ float win[WIN_CH][WIN_ROWS][WIN_W];
     const int g = blockIdx.z % gm.G;
     const int sp = blockIdx.z / gm.G;
     const int m0 = blockIdx.x * BM;
@@ -757,57 +757,33 @@ conv_fwd_staged_kernel(const T* __restrict__ x, const float* __restrict__ w,
     const int ohl_b = (pp_b / gm.OW) * gm.stride;   // window-row base
     const int iwl_b = ow_b * gm.stride;             // window-col base (pre-pad)
     float va[8];
-    // window staging through registers (<= 6 loads/thread), double-buffered
-    // so tile t+1's loads fly during tile t's B-build and MFMA
-    constexpr int WREG = 6;
-    float wr[WREG];
-    int wcnt = 0, wbase = 0;
-    auto win_load = [&](int k0w) {
-        const int cin0 = k0w / kk2;
-        const int nch = min(gm.Cin - 1, (k0w + BK - 1) / kk2) - cin0 + 1;
+
+    f32x4 acc[2][2] = {};
+    for (int k0 = ks; k0 < ke; k0 += BK) {
+        const int cin0 = k0 / kk2;
+        const int cin1 = min(gm.Cin - 1, (k0 + BK - 1) / kk2);
+        const int nch = cin1 - cin0 + 1;
+        // stage the window slab (coalesced along the input width)
         const int wtot = nch * rows_in * gm.W;
-        wcnt = 0;
-        wbase = cin0;
-        for (int e = tid; e < wtot && wcnt < WREG; e += 256, ++wcnt) {
+        for (int e = tid; e < wtot; e += 256) {
             const int ww = e % gm.W;
             const int rr = (e / gm.W) % rows_in;
             const int cc = e / (gm.W * rows_in);
             const int ih = ih0 + rr;
-            wr[wcnt] = (ih >= 0 && ih < gm.H)
-                           ? ld_f32(x + xn + (long)(cin0 + cc) * HW
-                                    + ih * gm.W + ww)
-                           : 0.f;
+            win[cc][rr][ww + gm.pad] =
+                (ih >= 0 && ih < gm.H)
+                    ? ld_f32(x + xn + (long)(cin0 + cc) * HW + ih * gm.W + ww)
+                    : 0.f;
         }
-    };
-    auto win_store = [&](int buf, int k0w) {
-        const int cin0 = k0w / kk2;
-        const int nch = min(gm.Cin - 1, (k0w + BK - 1) / kk2) - cin0 + 1;
-        const int wtot = nch * rows_in * gm.W;
-        int c = 0;
-        for (int e = tid; e < wtot && c < WREG; e += 256, ++c) {
-            const int ww = e % gm.W;
-            const int rr = (e / gm.W) % rows_in;
-            const int cc = e / (gm.W * rows_in);
-            win[buf][cc][rr][ww + gm.pad] = wr[c];
-        }
+        // zero the horizontal pad columns
         for (int e = tid; e < nch * rows_in * gm.pad * 2; e += 256) {
             const int side = e & 1;
             const int pe = e >> 1;
             const int pcol = pe % gm.pad;
             const int rr = (pe / gm.pad) % rows_in;
             const int cc = pe / (gm.pad * rows_in);
-            win[buf][cc][rr][side ? gm.pad + gm.W + pcol : pcol] = 0.f;
+            win[cc][rr][side ? gm.pad + gm.W + pcol : pcol] = 0.f;
         }
-    };
-    f32x4 acc[2][2] = {};
-    // prologue: window 0 staged synchronously
-    win_load(ks);
-    win_store(0, ks);
-    if (ks + BK < ke) win_load(ks + BK);
-    __syncthreads();
-    int cur = 0;
-    for (int k0 = ks; k0 < ke; k0 += BK) {
-        const int cin0 = k0 / kk2;
         // A tile (weights, coalesced fp32)
         {
             const int m = m0 + mm_a;
@@ -819,6 +795,7 @@ conv_fwd_staged_kernel(const T* __restrict__ x, const float* __restrict__ w,
             }
             st8_lds(&a_lds[mm_a][kkb], va);
         }
+        __syncthreads();
         // B tile built from the LDS window
         {
             float vb[8];
@@ -828,7 +805,7 @@ conv_fwd_staged_kernel(const T* __restrict__ x, const float* __restrict__ w,
 #pragma unroll
             for (int j = 0; j < 8; ++j) {
                 vb[j] = (k + j < K && pp_b < tile_p)
-                            ? win[cur][cin - cin0][ohl_b + kh][iwl_b + kw]
+                            ? win[cin - cin0][ohl_b + kh][iwl_b + kw]
                             : 0.f;
                 if (++kw == gm.khw) {
                     kw = 0;
@@ -840,11 +817,7 @@ conv_fwd_staged_kernel(const T* __restrict__ x, const float* __restrict__ w,
             }
             st8_lds(&b_lds[pp_b][kkb], vb);
         }
-        __syncthreads();   // a/b tiles ready; all reads of win[cur] done
-        if (k0 + BK < ke) {
-            win_store(cur ^ 1, k0 + BK);   // waits its in-flight loads
-            if (k0 + 2 * BK < ke) win_load(k0 + 2 * BK);
-        }
+        __syncthreads();
 #pragma unroll
         for (int fm = 0; fm < 2; ++fm)
 #pragma unroll
@@ -852,8 +825,7 @@ conv_fwd_staged_kernel(const T* __restrict__ x, const float* __restrict__ w,
                 acc[fm][fp] = mfma_tile2<TA, TB>(
                     &a_lds[wm + fm * 16 + (l & 15)][0],
                     &b_lds[wp + fp * 16 + (l & 15)][0], acc[fm][fp]);
-        __syncthreads();   // win[cur^1] written + MFMA done
-        cur ^= 1;
+        __syncthreads();
     }
     const long slab = (long)sp * gm.N * gm.G * gm.Cout * OHW;
     const long yb0 = ((long)n * gm.G * gm.Cout + (long)g * gm.Cout) * OHW
@@ -1217,14 +1189,11 @@ at::Tensor conv_fwd(at::Tensor x, at::Tensor w, at::Tensor bias,
     if (splitk > 1)
         partial = at::empty({(long)splitk * y.numel()},
                             x.options().dtype(at::kFloat));
-    const bool staged = (gm.OH * gm.OW) % BP == 0 && BP % gm.OW == 0
+    const bool staged = (P % BP == 0 || gm.N * gm.OH * gm.OW >= BP)
+                        && (gm.OH * gm.OW) % BP == 0 && BP % gm.OW == 0
                         && (gm.W + 2 * gm.pad) <= WIN_W - 2
                         && ((BP / gm.OW - 1) * gm.stride + gm.khw) <= WIN_ROWS
                         && (BK / (gm.khw * gm.khw) + 2) <= WIN_CH
-                        // window fits the 6-register staging pipeline
-                        && (BK / (gm.khw * gm.khw) + 2)
-                           * ((BP / gm.OW - 1) * gm.stride + gm.khw)
-                           * gm.W <= 6 * 256
                         && std::getenv("HETEROFL_CONV_NO_STAGED") == nullptr;
     DISPATCH_CONV_FT(x.scalar_type(), {
         if (staged && !fp8) {
