@@ -118,3 +118,18 @@ def test_process_make_stats(tmp_path):
                     result_dir=str(tmp_path))
     assert abs(st['Params'] - 51) < 1e-6
     assert abs(st['Ratio'] - 0.51) < 1e-6
+
+
+def test_make_sweep_generator(tmp_path, monkeypatch):
+    """make.py emits a bash sweep (reference: src/make.py:88-101)."""
+    import subprocess, sys, os
+    out = subprocess.run(
+        [sys.executable, os.path.join(os.path.dirname(__file__), '..',
+                                      'make.py'),
+         '--num_gpus', '8', '--num_experiments', '1'],
+        cwd=str(tmp_path), capture_output=True, text=True, timeout=60)
+    assert out.returncode == 0, out.stderr
+    sh = [f for f in os.listdir(tmp_path) if f.endswith('.sh')]
+    assert sh, out.stdout
+    body = open(str(tmp_path / sh[0])).read()
+    assert 'HIP_VISIBLE_DEVICES' in body and 'control_name' in body
