@@ -493,35 +493,29 @@ class BatchedClientTrainer:
         out = []
         # optional: split each rate group into HETEROFL_GROUP_SPLIT
         # subgroups so independent client chains replay on extra concurrent
-        # streams (fills the chip when one group's kernels underfill it)
+        # streams (experimental; subgroups get distinct graph tags so
+        # equal-size subgroups never share a captured graph)
         import os as _os
         gsplit = int(_os.environ.get('HETEROFL_GROUP_SPLIT', '1'))
-        if gsplit > 1 and self.device.type == 'cuda':
-            split_groups = {}
-            for (rate, sched), slots in groups.items():
-                if len(slots) >= 2 * gsplit:
-                    per = (len(slots) + gsplit - 1) // gsplit
-                    for si in range(0, len(slots), per):
-                        split_groups[(rate, sched, si)] = slots[si:si + per]
-                else:
-                    split_groups[(rate, sched, 0)] = slots
-            groups = {(r, s): v for (r, s, _), v in split_groups.items()} \
-                if len(split_groups) == len(groups) else None
-            if groups is None:
-                groups = {}
-                for (rate, sched, si), slots in split_groups.items():
-                    groups[(rate, sched, si)] = slots
+        split_groups = {}
+        for (rate, sched), slots in groups.items():
+            if gsplit > 1 and self.device.type == 'cuda' \
+                    and len(slots) >= 2 * gsplit:
+                per = (len(slots) + gsplit - 1) // gsplit
+                for si in range(0, len(slots), per):
+                    split_groups[(rate, sched, si)] = slots[si:si + per]
+            else:
+                split_groups[(rate, sched, 0)] = slots
         # graph-eligible groups are prepared on the default stream, then
         # their replay trains run CONCURRENTLY on per-group HIP streams (the
         # rate-e group's tiny latency-bound steps hide under rate-a's)
         graphed = []
-        for key, slots in groups.items():
-            rate, sched = key[0], key[1]
+        for (rate, sched, tag), slots in split_groups.items():
             use_graph = (self.device.type == 'cuda'
                          and cfg.get('hip_graphs', True)
                          and len(set(sched)) == 1)
             if use_graph:
-                graphed.append((rate, sched, slots))
+                graphed.append((rate, sched, slots, tag))
             else:
                 out.extend(self._train_group(rate, sched, slots, user_idx,
                                              local_parameters, label_split,
@@ -530,12 +524,13 @@ class BatchedClientTrainer:
             preps = [self._prepare_graphed(rate, sched, slots, user_idx,
                                            [local_parameters[m]
                                             for m in slots],
-                                           label_split, lr)
-                     for rate, sched, slots in graphed]
+                                           label_split, lr, tag)
+                     for rate, sched, slots, tag in graphed]
             from .runner import _phase_timer
             with _phase_timer('2d.replay'):
                 if len(preps) > 1:
-                    if not hasattr(self, '_streams') or                             len(self._streams) < len(preps):
+                    if not hasattr(self, '_streams') or \
+                            len(self._streams) < len(preps):
                         self._streams = [torch.cuda.Stream()
                                          for _ in range(len(preps))]
                     cur = torch.cuda.current_stream()
@@ -553,15 +548,18 @@ class BatchedClientTrainer:
                 else:
                     gs, x_cat, y_cat, nspe = preps[0]
                     gs.run_epochs(x_cat, y_cat, nspe)
-            for (rate, sched, slots), (gs, _, _, _) in zip(graphed, preps):
+            for (rate, sched, slots, tag), (gs, _, _, _) in zip(graphed,
+                                                                preps):
                 out.extend(self._finish_graphed(
                     gs, slots, [local_parameters[m] for m in slots], logger))
         return out
 
-    def _graph_step(self, rate, sched, R, lr):
-        """Get/create the captured training step for this group shape."""
+    def _graph_step(self, rate, sched, R, lr, tag=0):
+        """Get/create the captured training step for this group shape.  The
+        tag keeps concurrently-replayed subgroups (HETEROFL_GROUP_SPLIT) on
+        distinct graphs/buffers."""
         from .graphs import GraphedGroupStep
-        key = (rate, R, tuple(sched), lr)
+        key = (rate, R, tuple(sched), lr, tag)
         if key not in self._graph_cache:
             cfg = self.cfg
             model = self._batched_model(rate, R)
@@ -666,7 +664,7 @@ class BatchedClientTrainer:
         return list(zip(slots, cpu_or_dev))
 
     def _prepare_graphed(self, rate, sched, slots, user_idx, locals_list,
-                         label_split, lr):
+                         label_split, lr, tag=0):
         """Default-stream phase of the hipGraph path: build/capture, pack,
         masks, stage all epochs' augmented data.  Returns (gs, x, y, nspe)."""
         from .runner import _phase_timer
@@ -674,7 +672,7 @@ class BatchedClientTrainer:
         R = len(slots)
         device = self.device
         with _phase_timer('2a.graph_build'):
-            gs = self._graph_step(rate, sched, R, lr)
+            gs = self._graph_step(rate, sched, R, lr, tag)
             if gs.graph is None:
                 gs.ensure_captured()
         with _phase_timer('2b.pack'):
