@@ -132,14 +132,21 @@ def _read_idx(path):
 def _load_mnist_raw(root, split):
     prefix = 'train' if split == 'train' else 't10k'
     raw = os.path.join(root, 'raw')
-    for ext in ('', '.gz'):
-        ip = os.path.join(raw, f'{prefix}-images-idx3-ubyte{ext}')
-        lp = os.path.join(raw, f'{prefix}-labels-idx1-ubyte{ext}')
-        if os.path.exists(ip) and os.path.exists(lp):
-            img = torch.from_numpy(_read_idx(ip).copy())
-            target = _read_idx(lp).astype(np.int64).tolist()
-            return img, target
-    return None
+
+    def find(stem):
+        for ext in ('', '.gz'):
+            p = os.path.join(raw, stem + ext)
+            if os.path.exists(p):
+                return p
+        return None
+
+    ip = find(f'{prefix}-images-idx3-ubyte')
+    lp = find(f'{prefix}-labels-idx1-ubyte')
+    if ip is None or lp is None:
+        return None
+    img = torch.from_numpy(_read_idx(ip).copy())
+    target = _read_idx(lp).astype(np.int64).tolist()
+    return img, target
 
 
 def _load_cifar_raw(root, split, name):
