@@ -23,7 +23,7 @@ cfg['data_name'] = 'CIFAR10'
 cfg['model_name'] = 'resnet18'
 cfg['device'] = 'cuda:0'
 cfg['engine'] = 'batched'
-cfg['compute_dtype'] = 'bfloat16'
+cfg['compute_dtype'] = os.environ.get('HETEROFL_SOAK_DTYPE', 'bfloat16')
 cfg['metric_name'] = {'train': {'Local': ['Local-Loss']},
                       'test': {'Global': ['Global-Loss']}}
 process_control(cfg)
