@@ -81,6 +81,32 @@ def test_hyperparameters_wikitext2():
                                   'dropout': 0.2}
 
 
+def test_hyperparameters_centralized_none():
+    """'none' data_split_mode branches used by the centralized entries
+    (reference: src/utils.py:166-170, 188-192, 207-211)."""
+    cfg = _cfg('0_1_1_none_fix_a1_bn_1_1', 'MNIST')
+    assert cfg['num_epochs'] == 200
+    assert cfg['batch_size'] == {'train': 100, 'test': 500}
+    assert cfg['milestones'] == [100]
+    cfg = _cfg('0_1_1_none_fix_a1_bn_1_1', 'CIFAR10')
+    assert cfg['num_epochs'] == 400
+    assert cfg['batch_size'] == {'train': 100, 'test': 500}
+    assert cfg['milestones'] == [150, 250]
+    cfg = _cfg('0_1_1_none_fix_a1_bn_1_1', 'WikiText2')
+    assert cfg['num_epochs'] == 100
+    assert cfg['batch_size'] == {'train': 100, 'test': 100}
+    assert cfg['milestones'] == [25, 50]
+    with pytest.raises(ValueError):
+        _cfg('0_1_1_bogus_fix_a1_bn_1_1', 'CIFAR10')
+
+
+def test_hyperparameters_mnist_noniid():
+    cfg = _cfg('1_100_0.1_non-iid-2_fix_a1_bn_1_1', 'MNIST')
+    assert cfg['num_epochs'] == {'global': 400, 'local': 5}
+    assert cfg['milestones'] == [200]
+    assert cfg['batch_size'] == {'train': 10, 'test': 50}
+
+
 def test_scale_mask_flags():
     cfg = _cfg('1_10_0.1_iid_fix_a1_bn_0_0')
     assert cfg['scale'] is False and cfg['mask'] is False
