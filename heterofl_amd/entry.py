@@ -238,6 +238,64 @@ def run_centralized_experiment(cfg, pivot_metric, pivot_sign, metric_name):
         logger.safe(False)
 
 
+def run_centralized_eval(cfg, metric_name):
+    """Centralized (non-fed) evaluation entry (reference:
+    src/test_classifier.py:41-91 / src/test_transformer.py): load
+    {tag}_best.pt, re-run the sBN stats pass over the train set (vision
+    only), evaluate the test set, save ./output/result/{tag}.pt."""
+    from .data import make_data_loader, BatchDataset
+    from .metrics import Metric
+    from .utils import collate, to_device
+    cfg['metric_name'] = metric_name
+    process_control(cfg)
+    seeds = list(range(cfg['init_seed'], cfg['init_seed'] + cfg['num_experiments']))
+    for seed in seeds:
+        cfg['model_tag'] = model_tag_of(seed, cfg)
+        print('Eval: {}'.format(cfg['model_tag']))
+        torch.manual_seed(seed)
+        if cfg['device'].startswith('cuda') and not torch.cuda.is_available():
+            cfg['device'] = 'cpu'
+        dataset = fetch_dataset(cfg['data_name'], cfg['subset'],
+                                synthetic=cfg.get('synthetic', False))
+        process_dataset(dataset, cfg)
+        model = make_model(cfg, model_rate=cfg['global_model_rate']).to(cfg['device'])
+        last_epoch, _, _, model, _, _, train_logger = resume(
+            model, cfg['model_tag'], load_tag='best', strict=False)
+        logger = Logger(os.path.join('output', 'runs',
+                                     'test_{}'.format(cfg['model_tag'])))
+        metric = Metric()
+        is_lm = cfg['model_name'] == 'transformer'
+        logger.safe(True)
+        with torch.no_grad():
+            if not is_lm:
+                test_model = make_model(cfg, model_rate=cfg['global_model_rate'],
+                                        track=True).to(cfg['device'])
+                test_model.load_state_dict(model.state_dict(), strict=False)
+                test_model.train(True)
+                loader = make_data_loader({'train': dataset['train']}, cfg)['train']
+                for input in loader:
+                    test_model(to_device(collate(input), cfg['device']))
+            else:
+                test_model = model
+            test_model.train(False)
+            if is_lm:
+                ds = BatchDataset(dataset['test'], cfg['bptt'])
+                test_batches = (ds[i] for i in range(len(ds)))
+            else:
+                loader = make_data_loader({'test': dataset['test']}, cfg)['test']
+                test_batches = (collate(b) for b in loader)
+            for input in test_batches:
+                input = to_device(input, cfg['device'])
+                output = test_model(input)
+                ev = metric.evaluate(cfg['metric_name']['test'], input, output)
+                logger.append(ev, 'test', n=input['label'].size(0))
+        logger.write('test', cfg['metric_name']['test'])
+        logger.safe(False)
+        result = {'cfg': cfg, 'epoch': last_epoch,
+                  'logger': {'train': train_logger, 'test': logger}}
+        save(result, './output/result/{}.pt'.format(cfg['model_tag']))
+
+
 def run_fed_eval(cfg, metric_name, result_key='test'):
     """Evaluation entry (reference: src/test_classifier_fed.py:41-60): load
     {tag}_best.pt, re-run sBN stats (vision), evaluate, save

@@ -202,3 +202,55 @@ def test_centralized_entry(base_cfg, tmp_path, monkeypatch):
     tag = cfg['model_tag']
     assert os.path.exists('./output/model/{}_checkpoint.pt'.format(tag))
     assert os.path.exists('./output/model/{}_best.pt'.format(tag))
+    # centralized eval entry (reference: src/test_classifier.py): loads the
+    # best checkpoint, re-runs sBN stats, saves ./output/result/{tag}.pt
+    entry.run_centralized_eval(
+        cfg, {'train': ['Loss', 'Accuracy'], 'test': ['Loss', 'Accuracy']})
+    assert os.path.exists('./output/result/{}.pt'.format(tag))
+    from heterofl_amd.utils import load
+    result = load('./output/result/{}.pt'.format(tag))
+    assert 'test/Accuracy' in result['logger']['test'].mean
+
+
+def test_centralized_lm_entry(base_cfg, tmp_path, monkeypatch):
+    """Centralized masked-LM train + eval entries
+    (reference: src/train_transformer.py, src/test_transformer.py)."""
+    import os
+    monkeypatch.chdir(tmp_path)
+    import heterofl_amd.entry as entry
+    from heterofl_amd.config import default_config
+    orig_pc = entry.process_control
+
+    def small_pc(cfg):
+        orig_pc(cfg)
+        cfg['num_epochs'] = 1
+        cfg['transformer'] = {'embedding_size': 16, 'num_heads': 2,
+                              'hidden_size': 16, 'num_layers': 1,
+                              'dropout': 0.0}
+        cfg['bptt'] = 8
+
+    monkeypatch.setattr(entry, 'process_control', small_pc)
+    import heterofl_amd.data as data_mod
+    orig_fetch = data_mod.fetch_dataset
+    monkeypatch.setattr(
+        entry, 'fetch_dataset',
+        lambda name, subset=None, synthetic=False: orig_fetch(
+            name, subset, synthetic=True, synthetic_size=400))
+    cfg = default_config()
+    cfg.update({'data_name': 'WikiText2', 'subset': 'label',
+                'model_name': 'transformer', 'device': 'cpu',
+                'num_experiments': 1, 'init_seed': 0, 'resume_mode': 0,
+                'batch_size': {'train': 4, 'test': 4}})
+    cfg['control'] = dict(cfg['control'], data_split_mode='none')
+    cfg['control_name'] = '0_1_1_none_fix_a1_bn_1_1'
+    entry.run_centralized_experiment(
+        cfg, 'Perplexity', -1,
+        {'train': ['Loss', 'Perplexity'], 'test': ['Loss', 'Perplexity']})
+    tag = cfg['model_tag']
+    assert os.path.exists('./output/model/{}_best.pt'.format(tag))
+    entry.run_centralized_eval(
+        cfg, {'train': ['Loss', 'Perplexity'],
+              'test': ['Loss', 'Perplexity']})
+    from heterofl_amd.utils import load
+    result = load('./output/result/{}.pt'.format(tag))
+    assert 'test/Perplexity' in result['logger']['test'].mean
