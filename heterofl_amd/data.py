@@ -174,6 +174,46 @@ def _load_cifar_raw(root, split, name):
     return torch.from_numpy(np.concatenate(imgs).copy()), targets
 
 
+_IMG_EXTENSIONS = ('.png', '.jpg', '.jpeg', '.bmp', '.ppm', '.pgm', '.webp')
+
+
+def _load_folder_raw(root, split, size=None):
+    """Class-per-subdirectory image tree -> (uint8 (N,H,W,C) tensor, targets,
+    classes).  Covers the reference's ImageFolder/ImageNet/Omniglot family
+    (reference: src/datasets/folder.py:9-61, src/datasets/imagenet.py:14-60):
+    labels come from the sorted subdirectory names of ``root/<split>``; images
+    are decoded with PIL and resized to a common size (``size`` or the first
+    image's size)."""
+    base = os.path.join(root, split)
+    if not os.path.isdir(base):
+        return None
+    try:
+        from PIL import Image
+    except ImportError:
+        return None
+    classes = sorted(d for d in os.listdir(base)
+                     if os.path.isdir(os.path.join(base, d)))
+    if not classes:
+        return None
+    imgs, targets = [], []
+    for label, cls in enumerate(classes):
+        cdir = os.path.join(base, cls)
+        for fn in sorted(os.listdir(cdir)):
+            if not fn.lower().endswith(_IMG_EXTENSIONS):
+                continue
+            with Image.open(os.path.join(cdir, fn)) as im:
+                im = im.convert('RGB')
+                if size is None:
+                    size = im.size
+                if im.size != size:
+                    im = im.resize(size)
+                imgs.append(np.asarray(im, dtype=np.uint8))
+            targets.append(label)
+    if not imgs:
+        return None
+    return torch.from_numpy(np.stack(imgs)), targets, classes
+
+
 def _load_wikitext_raw(root, split):
     for fn in (f'wiki.{split}.tokens', f'{split}.txt'):
         p = os.path.join(root, fn)
@@ -228,6 +268,38 @@ def fetch_dataset(data_name, subset='label', synthetic=False,
             img, target = raw
             dataset[split] = VisionDataset(data_name, img, target,
                                            classes_size, split == 'train')
+    elif data_name in ('ImageFolder', 'ImageNet', 'Omniglot'):
+        # folder-tree image datasets (reference: src/datasets/folder.py,
+        # imagenet.py, omniglot.py — defined but unused by the three
+        # benchmark tasks; supported here for parity)
+        train_raw = _load_folder_raw(root, 'train')
+        if train_raw is None:
+            if not synthetic:
+                raise FileNotFoundError(
+                    f'{data_name} folder tree not found under {root}; expects '
+                    f'{root}/train/<class>/*.png (pass synthetic=True '
+                    f'otherwise)')
+            classes_size = 10
+            for split in ('train', 'test'):
+                n = synthetic_size or 1000
+                if split == 'test':
+                    n = max(n // 5, 50)
+                img, target = _synthetic_vision('CIFAR10', n, classes_size,
+                                                seed=0 if split == 'train' else 1)
+                dataset[split] = VisionDataset(data_name, img, target,
+                                               classes_size, split == 'train')
+        else:
+            img, target, classes = train_raw
+            classes_size = len(classes)
+            dataset['train'] = VisionDataset(data_name, img, target,
+                                             classes_size, True)
+            test_raw = _load_folder_raw(root, 'test',
+                                        size=(img.size(2), img.size(1)))
+            if test_raw is None:
+                test_raw = train_raw
+            timg, ttarget = test_raw[0], test_raw[1]
+            dataset['test'] = VisionDataset(data_name, timg, ttarget,
+                                            classes_size, False)
     elif data_name in ('PennTreebank', 'WikiText2', 'WikiText103'):
         vocab = None
         for split in ('train', 'test'):

@@ -207,6 +207,29 @@ def test_cifar_pickle_parser_roundtrip(tmp_path):
     assert ds['train'][0]['img'].shape == (3, 32, 32)
 
 
+def test_imagefolder_parser_roundtrip(tmp_path):
+    """Class-per-subdir image tree (reference: src/datasets/folder.py)."""
+    import numpy as np
+    from PIL import Image
+    from heterofl_amd.data import fetch_dataset
+    rng = np.random.RandomState(0)
+    for split, n in [('train', 3), ('test', 2)]:
+        for cls in ['cat', 'dog']:
+            d = tmp_path / split / cls
+            d.mkdir(parents=True)
+            for i in range(n):
+                arr = rng.randint(0, 256, (16, 16, 3), dtype=np.uint8)
+                Image.fromarray(arr).save(d / f'{i}.png')
+    ds = fetch_dataset('ImageFolder', root=str(tmp_path))
+    assert len(ds['train']) == 6 and len(ds['test']) == 4
+    item = ds['train'][0]
+    assert item['img'].shape == (3, 16, 16)
+    assert ds['train'].classes_size == 2
+    # sorted dir names define labels: cat=0, dog=1
+    assert int(ds['train'][0]['label']) == 0
+    assert int(ds['train'][5]['label']) == 1
+
+
 def test_wikitext_raw_parser(tmp_path):
     from heterofl_amd.data import fetch_dataset
     (tmp_path / 'wiki.train.tokens').write_text('a b c a b c d e')
