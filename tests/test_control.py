@@ -1,5 +1,7 @@
 """control_name grammar tests against hand-computed expansions
 (reference semantics: src/utils.py:113-215)."""
+import os
+
 import numpy as np
 import pytest
 
@@ -144,6 +146,39 @@ def test_process_make_stats(tmp_path):
                     result_dir=str(tmp_path))
     assert abs(st['Params'] - 51) < 1e-6
     assert abs(st['Ratio'] - 0.51) < 1e-6
+
+
+def test_process_plots(tmp_path):
+    """Learning-curve + interpolation plots (reference: src/process.py
+    visualization half)."""
+    pytest.importorskip('matplotlib')
+    from process import (crawl_results, aggregate, make_learning_curves,
+                         make_interpolation_plot)
+    from heterofl_amd.logger import Logger
+    from heterofl_amd.utils import save
+    rdir = tmp_path / 'result'
+    rdir.mkdir()
+    # two interpolation points x2 seeds, with per-round history
+    for mode, acc in [('a3-e7', 60.0), ('a7-e3', 70.0)]:
+        for seed in (0, 1):
+            lg = Logger(str(tmp_path / 'runs' / f'{seed}_{mode}'))
+            for ep in range(3):
+                lg.safe(True)
+                lg.append({'Global-Accuracy': acc + ep + seed}, 'test', n=10)
+                lg.safe(False)
+                if ep < 2:   # keep the final epoch's running mean, as the
+                    lg.reset()   # entries do before pickling the logger
+            tag = f'{seed}_CIFAR10_label_resnet18_1_100_0.1_iid_fix_{mode}_bn_1_1'
+            save({'cfg': {}, 'epoch': 3,
+                  'logger': {'train': None, 'test': lg}},
+                 str(rdir / f'{tag}.pt'))
+    results = crawl_results(str(rdir))
+    assert len(results) == 2 and all(len(v) == 2 for v in results.values())
+    curves = make_learning_curves(results, str(tmp_path / 'vis'))
+    assert len(curves) == 2 and all(os.path.exists(p) for p in curves)
+    table = aggregate(results)
+    interp = make_interpolation_plot(table, str(tmp_path / 'vis'))
+    assert len(interp) == 1 and os.path.exists(interp[0])
 
 
 def test_make_sweep_generator(tmp_path, monkeypatch):
