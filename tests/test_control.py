@@ -196,6 +196,46 @@ def test_make_sweep_generator(tmp_path, monkeypatch):
     assert 'HIP_VISIBLE_DEVICES' in body and 'control_name' in body
 
 
+def test_summary_cli(tmp_path):
+    """summary.py writes one cost file per width level
+    (reference: src/summary.py:44-47)."""
+    import subprocess, sys
+    root = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    out = subprocess.run(
+        [sys.executable, os.path.join(root, 'summary.py'),
+         '--data_name', 'MNIST', '--model_name', 'conv',
+         '--control_name', '1_4_0.5_iid_fix_a1-e1_bn_1_1',
+         '--synthetic', '1'],
+        cwd=str(tmp_path), capture_output=True, text=True, timeout=300,
+        env=dict(os.environ, PYTHONPATH=root))
+    assert out.returncode == 0, out.stderr
+    from heterofl_amd.utils import load
+    for lv in 'abcde':
+        p = tmp_path / 'output' / 'result' / f'MNIST_conv_{lv}.pt'
+        assert p.exists(), (lv, out.stdout)
+    a = load(str(tmp_path / 'output' / 'result' / 'MNIST_conv_a.pt'))
+    e = load(str(tmp_path / 'output' / 'result' / 'MNIST_conv_e.pt'))
+    assert a['num_params'] > 50 * e['num_params']
+
+
+def test_make_ablation_generator(tmp_path):
+    """make_ablation.py emits the norm/scaler/mask ablation sweep
+    (reference: src/make_ablation.py:69-88)."""
+    import subprocess, sys
+    root = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    out = subprocess.run(
+        [sys.executable, os.path.join(root, 'make_ablation.py'),
+         '--num_gpus', '4', '--num_experiments', '1'],
+        cwd=str(tmp_path), capture_output=True, text=True, timeout=60)
+    assert out.returncode == 0, out.stderr
+    sh = [f for f in os.listdir(tmp_path) if f.endswith('.sh')]
+    assert sh, out.stdout
+    body = open(str(tmp_path / sh[0])).read()
+    # ablation axes present: a non-bn norm, scaler off, mask off
+    assert '_gn_' in body or '_in_' in body or '_ln_' in body
+    assert '_0_1' in body or '_1_0' in body or '_0_0' in body
+
+
 def test_mnist_idx_parser_roundtrip(tmp_path):
     """Real-file MNIST idx parser (reference: src/datasets/mnist.py)."""
     import gzip
