@@ -121,20 +121,23 @@ def test_checkpoint_resume_roundtrip(base_cfg, tmp_path, monkeypatch):
                       'mask': '1'}
     cfg['control_name'] = '1_4_0.5_iid_fix_a1_bn_1_1'
     import heterofl_amd.data as data_mod
+    import heterofl_amd.entry as entry_mod
     orig = data_mod.fetch_dataset
     monkeypatch.setattr(
         'heterofl_amd.entry.fetch_dataset',
         lambda name, subset=None, synthetic=False: orig(
             name, subset, synthetic=True, synthetic_size=40))
+    orig_pc = entry_mod.process_control
+
+    def small_pc(c):
+        orig_pc(c)
+        c['num_epochs'] = {'global': 2, 'local': 1}
+
+    monkeypatch.setattr(entry_mod, 'process_control', small_pc)
     metric_name = {'train': {'Local': ['Local-Loss', 'Local-Accuracy']},
                    'test': {'Local': ['Local-Loss', 'Local-Accuracy'],
                             'Global': ['Global-Loss', 'Global-Accuracy']}}
-    cfg2 = dict(cfg)
-    cfg2['num_epochs'] = 2  # process_control overrides to dataset default;
-    run_fed_experiment(dict(cfg, num_epochs=2), 'Global-Accuracy', +1,
-                       metric_name)
-    # process_control sets num_epochs from the dataset table; patch the
-    # saved checkpoint contract instead: files must exist with the tag
+    run_fed_experiment(dict(cfg), 'Global-Accuracy', +1, metric_name)
     tag = '0_MNIST_label_conv_1_4_0.5_iid_fix_a1_bn_1_1'
     ck = './output/model/{}_checkpoint.pt'.format(tag)
     assert os.path.exists(ck), os.listdir('./output/model')
@@ -144,6 +147,20 @@ def test_checkpoint_resume_roundtrip(base_cfg, tmp_path, monkeypatch):
     for key in ('cfg', 'epoch', 'data_split', 'label_split', 'model_dict',
                 'optimizer_dict', 'scheduler_dict', 'logger'):
         assert key in saved, key
+    assert saved['epoch'] == 3  # next epoch to run after 2 rounds
+    # resume_mode=1: restarts from the saved epoch (past the end here, so
+    # the run completes immediately but must load the checkpoint cleanly)
+    run_fed_experiment(dict(cfg, resume_mode=1), 'Global-Accuracy', +1,
+                       metric_name)
+    # resume_mode=2: reload weights + splits only, train from epoch 1 with a
+    # fresh logger (reference: src/train_classifier_fed.py:57-69)
+    run_fed_experiment(dict(cfg, resume_mode=2), 'Global-Accuracy', +1,
+                       metric_name)
+    saved2 = load(ck)
+    assert saved2['epoch'] == 3
+    # the data/label splits must survive both resumes unchanged
+    assert saved2['data_split']['train'].keys() == \
+        saved['data_split']['train'].keys()
 
 
 def test_resnet50_bottleneck_falls_back_sequential(base_cfg):
