@@ -140,3 +140,75 @@ def make_cfg_local(cfg):
 def test_distributed_sbn_stats_matches_sequential(tmp_path):
     mp.spawn(_stats_worker, args=(2, 29551, str(tmp_path)), nprocs=2,
              join=True)
+
+
+def test_round_sampling_rank_invariant():
+    """Every rank draws identical active users and dynamic rates from the
+    round-seeded generator — the no-communication reproducibility contract
+    (fed/runner.py:98-132)."""
+    from heterofl_amd.config import default_config
+    from heterofl_amd.control import process_control
+    from heterofl_amd.fed.runner import sample_active_users
+    from heterofl_amd.fed.federation import Federation
+    from heterofl_amd.models import make_model
+
+    cfg = default_config()
+    cfg['device'] = 'cpu'
+    cfg['data_name'] = 'CIFAR10'
+    cfg['model_name'] = 'resnet18'
+    cfg['control'] = {'fed': '1', 'num_users': '100', 'frac': '0.1',
+                      'data_split_mode': 'iid', 'model_split_mode': 'dynamic',
+                      'model_mode': 'a1-e1', 'norm': 'bn', 'scale': '1',
+                      'mask': '1'}
+    process_control(cfg)
+    cfg['classes_size'] = 10
+
+    def round_gen(epoch, seed=0):
+        g = torch.Generator()
+        g.manual_seed((seed * 1000003 + epoch) % (2 ** 63 - 1))
+        return g
+
+    model = make_model(cfg, model_rate=1.0)
+    label_split = {i: [0, 1] for i in range(100)}
+    seen = []
+    for epoch in (1, 2):
+        per_rank = []
+        for rank in range(2):   # both "ranks" replay the same drawing
+            g = round_gen(epoch)
+            users = sample_active_users(cfg, epoch, generator=g)
+            fed = Federation(model.state_dict(), cfg['model_rate'],
+                             label_split, cfg)
+            fed.make_model_rate(generator=g)
+            per_rank.append((users, list(fed.model_rate)))
+        assert per_rank[0] == per_rank[1], epoch
+        assert len(per_rank[0][0]) == 10          # ceil(0.1 * 100)
+        assert set(per_rank[0][1]) <= {1, 0.0625}  # dynamic a1-e1 levels
+        seen.append(per_rank[0])
+    assert seen[0] != seen[1]   # different rounds draw differently
+
+
+def test_dynamic_rate_distribution():
+    """Dynamic-mode multinomial follows the mode proportions
+    (reference: src/fed.py:15-24 semantics)."""
+    from heterofl_amd.config import default_config
+    from heterofl_amd.control import process_control
+    from heterofl_amd.fed.federation import Federation
+    from heterofl_amd.models import make_model
+
+    cfg = default_config()
+    cfg['device'] = 'cpu'
+    cfg['data_name'] = 'CIFAR10'
+    cfg['model_name'] = 'resnet18'
+    cfg['control'] = {'fed': '1', 'num_users': '1000', 'frac': '0.1',
+                      'data_split_mode': 'iid', 'model_split_mode': 'dynamic',
+                      'model_mode': 'a1-e3', 'norm': 'bn', 'scale': '1',
+                      'mask': '1'}
+    process_control(cfg)
+    cfg['classes_size'] = 10
+    model = make_model(cfg, model_rate=1.0)
+    fed = Federation(model.state_dict(), cfg['model_rate'],
+                     {i: [0] for i in range(1000)}, cfg)
+    g = torch.Generator().manual_seed(7)
+    fed.make_model_rate(generator=g)
+    frac_full = sum(1 for r in fed.model_rate if r == 1) / 1000
+    assert 0.18 < frac_full < 0.32   # expected 0.25
