@@ -212,3 +212,28 @@ def test_dynamic_rate_distribution():
     fed.make_model_rate(generator=g)
     frac_full = sum(1 for r in fed.model_rate if r == 1) / 1000
     assert 0.18 < frac_full < 0.32   # expected 0.25
+
+
+def test_bench_torchrun_contract(tmp_path):
+    """bench.py under torchrun (the driver's exact N>1 launch shape) prints
+    ONE JSON line from rank 0 with the BASELINE metric/config."""
+    import json
+    import subprocess
+    import sys
+    root = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    env = dict(os.environ, HETEROFL_BENCH_SPU='10', OMP_NUM_THREADS='2')
+    out = subprocess.run(
+        [sys.executable, '-m', 'torch.distributed.run', '--nnodes=1',
+         '--nproc-per-node', '2', '--master-addr', '127.0.0.1',
+         '--master-port', '29531', os.path.join(root, 'bench.py'),
+         '--gpus', '2', '--steps', '1', '--warmup', '0'],
+        cwd=root, capture_output=True, text=True, timeout=600, env=env)
+    assert out.returncode == 0, out.stderr[-2000:]
+    lines = [l for l in out.stdout.splitlines() if l.startswith('{')]
+    assert len(lines) == 1, out.stdout
+    rec = json.loads(lines[0])
+    assert rec['metric'] == 'local-train samples/sec/node'
+    assert rec['n_gpus'] == 2 and rec['scaling'] == 'weak'
+    assert rec['config']['num_users'] == 200   # weak scaling: 100 users/GPU
+    assert rec['config']['active_clients'] == 20
+    assert rec['value'] > 0
