@@ -241,3 +241,43 @@ def test_combine_tolerates_missing_clients(base_cfg):
     # rows the tiny client covers were updated; the rest untouched
     assert (w[:n1] - (before['conv1.weight'][:n1] + 1.0)).abs().max() < 1e-6
     assert (w[n1:] - before['conv1.weight'][n1:]).abs().max() < 1e-6
+
+
+def test_combine_label_split_filtering_transformer(base_cfg):
+    """Transformer output layers (embedding vocab rows + decoder.linear2
+    rows incl. bias) aggregate only over clients holding that token
+    (reference: src/fed.py:263-288)."""
+    cfg = make_cfg(base_cfg, '1_2_1_iid_fix_a1-a1_bn_1_1',
+                   data_name='WikiText2', model_name='transformer')
+    cfg['transformer'] = {'embedding_size': 16, 'num_heads': 2,
+                          'hidden_size': 16, 'num_layers': 1, 'dropout': 0.0}
+    cfg['bptt'] = 8
+    cfg['classes_size'] = 12
+    cfg['num_tokens'] = 12
+    model = make_model(cfg, model_rate=1)
+    label_split = {0: [0, 1], 1: [1, 2]}
+    fed = _fed(cfg, model, rates=[1.0, 1.0], label_split=label_split)
+    user_idx = [0, 1]
+    local, pidx = fed.distribute(user_idx)
+    emb_key = next(k for k in fed.global_parameters
+                   if k.split('.')[-2] == 'embedding' and k.endswith('weight'))
+    dec_key = next(k for k in fed.global_parameters
+                   if 'decoder' in k and 'linear2' in k and k.endswith('weight'))
+    dec_bias = dec_key.replace('weight', 'bias')
+    before = {k: fed.global_parameters[k].clone()
+              for k in (emb_key, dec_key, dec_bias)}
+    for m, fill in [(0, 1.0), (1, 3.0)]:
+        for k in local[m]:
+            if local[m][k].is_floating_point():
+                local[m][k] = torch.full_like(local[m][k], fill)
+    fed.combine(local, pidx, user_idx)
+    for k in (emb_key, dec_key, dec_bias):
+        w = fed.global_parameters[k]
+        assert torch.allclose(w[0], torch.full_like(w[0], 1.0)), k
+        assert torch.allclose(w[1], torch.full_like(w[1], 2.0)), k
+        assert torch.allclose(w[2], torch.full_like(w[2], 3.0)), k
+        assert torch.equal(w[3:], before[k][3:]), k
+    # a non-output layer aggregates over BOTH clients everywhere
+    q = next(k for k in fed.global_parameters if 'linear_q.weight' in k)
+    wq = fed.global_parameters[q]
+    assert torch.allclose(wq, torch.full_like(wq, 2.0))
