@@ -17,8 +17,9 @@ def fixed_combinations():
         for combo in itertools.combinations(modes, k):
             dynamic.append('-'.join(combo))
     interp = []
-    for hi, lo in [('a', 'b'), ('a', 'c'), ('a', 'd'), ('a', 'e')]:
-        for n in range(1, 10):
+    # all 10 level pairs x 9 ratios = 90 (reference: src/make.py:62-66)
+    for n in range(1, 10):
+        for hi, lo in itertools.combinations(levels, 2):
             interp.append('{}{}-{}{}'.format(hi, n, lo, 10 - n))
     return modes, dynamic, interp
 

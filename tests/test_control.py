@@ -196,6 +196,19 @@ def test_make_sweep_generator(tmp_path, monkeypatch):
     assert 'HIP_VISIBLE_DEVICES' in body and 'control_name' in body
 
 
+def test_make_sweep_composition():
+    """Sweep sizes pin the reference's enumeration: 5 fixed levels, 26
+    multi-level dynamic combinations, 90 interpolation ratios
+    (reference: src/make.py:55-66)."""
+    from make import fixed_combinations
+    modes, dynamic, interp = fixed_combinations()
+    assert len(modes) == 5
+    assert len(dynamic) == 26      # C(5,2)+C(5,3)+C(5,4)+C(5,5)
+    assert len(interp) == 90       # 10 level pairs x 9 ratios
+    assert 'a1-b1-c1-d1-e1' in dynamic
+    assert 'a5-e5' in interp and 'd9-e1' in interp
+
+
 def test_summary_cli(tmp_path):
     """summary.py writes one cost file per width level
     (reference: src/summary.py:44-47)."""
