@@ -271,3 +271,22 @@ def test_centralized_lm_entry(base_cfg, tmp_path, monkeypatch):
     from heterofl_amd.utils import load
     result = load('./output/result/{}.pt'.format(tag))
     assert 'test/Perplexity' in result['logger']['test'].mean
+
+
+@pytest.mark.parametrize('model_name', ['resnet34', 'resnet101', 'resnet152'])
+def test_deep_resnet_factories(base_cfg, model_name):
+    """Deep resnet factories build and run forward/backward at full and
+    fractional width (reference: src/models/resnet.py:161-208)."""
+    from tests.conftest import make_cfg
+    from heterofl_amd.models import make_model
+    cfg = make_cfg(base_cfg, '1_2_1_iid_fix_a1_bn_1_1',
+                   data_name='CIFAR10', model_name=model_name)
+    for rate in (1.0, 0.25):
+        model = make_model(cfg, model_rate=rate)
+        x = torch.randn(2, 3, 32, 32)
+        out = model({'img': x, 'label': torch.tensor([0, 1]),
+                     'label_split': torch.arange(10)})
+        assert out['score'].shape == (2, 10)
+        out['loss'].backward()
+        assert all(p.grad is not None for p in model.parameters()
+                   if p.requires_grad)
