@@ -290,3 +290,42 @@ def test_deep_resnet_factories(base_cfg, model_name):
         out['loss'].backward()
         assert all(p.grad is not None for p in model.parameters()
                    if p.requires_grad)
+
+
+def test_golden_path_trajectory(base_cfg):
+    """Golden-path regression: the BASELINE config-1 loss
+    trajectory under fixed seeds is pinned so any refactor that changes the
+    engine's numerics is caught immediately.  Values captured on this
+    torch build; tolerance 1e-4 absorbs BLAS nondeterminism, not semantic
+    changes."""
+    cfg = make_cfg(base_cfg, '1_10_0.1_iid_fix_a1_bn_1_1',
+                   data_name='MNIST', model_name='conv')
+    cfg['engine'] = 'sequential'
+    cfg['num_epochs'] = {'global': 3, 'local': 1}
+    torch.manual_seed(0)
+    ds = fetch_dataset('MNIST', synthetic=True, synthetic_size=100)
+    process_dataset(ds, cfg)
+    torch.manual_seed(7)
+    data_split, label_split = split_dataset(ds, 10, 'iid', cfg['classes_size'])
+    torch.manual_seed(1)
+    model = make_model(cfg)
+    opt = make_optimizer(model, cfg['lr'], cfg)
+    runner = FedRunner(cfg, ds, data_split, label_split, model, opt)
+
+    def global_loss():
+        model.load_state_dict(runner.federation.global_parameters)
+        model.train(True)
+        with torch.no_grad():
+            batch = {'img': torch.stack([ds['train'][i]['img']
+                                         for i in range(100)]),
+                     'label': torch.tensor(ds['train'].target)}
+            return model(batch)['loss'].item()
+
+    traj = []
+    for ep in range(1, 4):
+        torch.manual_seed(100 + ep)   # pin per-round batch shuffling
+        runner.train_round(ep)
+        traj.append(global_loss())
+    golden = [2.297021, 2.291516, 2.288667]
+    for got, want in zip(traj, golden):
+        assert abs(got - want) < 1e-4, (traj, golden)
