@@ -101,3 +101,18 @@ def test_recur_and_collate_and_to_device():
     assert c['img'].shape == (2, 3)
     d = to_device({'t': torch.ones(1)}, 'cpu')
     assert d['t'].device.type == 'cpu'
+
+
+def test_no_undefined_names_static():
+    """Static undefined-name sweep over the package (scripts/namecheck.py):
+    GPU-only code paths never execute in CPU CI, so a typo there would
+    otherwise only surface on the GPU box."""
+    import importlib.util
+    import os
+    root = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    spec = importlib.util.spec_from_file_location(
+        'namecheck', os.path.join(root, 'scripts', 'namecheck.py'))
+    mod = importlib.util.module_from_spec(spec)
+    spec.loader.exec_module(mod)
+    problems = mod.check_tree(os.path.join(root, 'heterofl_amd'))
+    assert not problems, problems
