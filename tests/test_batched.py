@@ -217,3 +217,31 @@ def test_batched_lm_matches_sequential(base_cfg):
             a, b = seq_out[m][k].float(), bt_out[m][k].float()
             diff = (a - b).abs().max().item()
             assert diff < 5e-4, (m, k, diff)
+
+
+def test_batched_engine_e2e_conv(base_cfg):
+    """Full federated round with engine='batched' and the 4-block conv model
+    (BatchedConv path, reference model: src/models/conv.py)."""
+    from heterofl_amd.data import fetch_dataset, split_dataset
+    from heterofl_amd.fed import FedRunner
+    from heterofl_amd.utils import process_dataset, make_optimizer
+    cfg = make_cfg(base_cfg, '1_4_0.5_iid_fix_a1-e1_bn_1_1',
+                   data_name='MNIST', model_name='conv')
+    cfg['engine'] = 'batched'
+    cfg['num_epochs'] = {'global': 1, 'local': 1}
+    torch.manual_seed(0)
+    ds = fetch_dataset('MNIST', synthetic=True, synthetic_size=80)
+    process_dataset(ds, cfg)
+    data_split, label_split = split_dataset(ds, 4, 'iid', cfg['classes_size'])
+    model = make_model(cfg)
+    opt = make_optimizer(model, cfg['lr'], cfg)
+    runner = FedRunner(cfg, ds, data_split, label_split, model, opt)
+    before = {k: v.clone() for k, v in
+              runner.federation.global_parameters.items()}
+    runner.train_round(1)
+    # training moved the global parameters
+    moved = any((runner.federation.global_parameters[k] - v).abs().max() > 0
+                for k, v in before.items() if v.is_floating_point())
+    assert moved
+    tm = runner.stats()
+    runner.test(tm, 1)

@@ -281,3 +281,21 @@ def test_combine_label_split_filtering_transformer(base_cfg):
     q = next(k for k in fed.global_parameters if 'linear_q.weight' in k)
     wq = fed.global_parameters[q]
     assert torch.allclose(wq, torch.full_like(wq, 2.0))
+
+
+def test_single_client_partial_rate_combine(base_cfg):
+    """One client at rate<1: combine writes exactly that client's slice and
+    leaves every untouched entry of the global tensors unchanged."""
+    cfg = make_cfg(base_cfg, '1_1_1_iid_fix_a1_bn_1_1',
+                   data_name='CIFAR10', model_name='resnet18')
+    model = make_model(cfg, model_rate=1)
+    fed = _fed(cfg, model, rates=[0.5])
+    before = {k: v.clone() for k, v in fed.global_parameters.items()}
+    local, pidx = fed.distribute([0])
+    filled = {k: torch.full_like(v, 5.0) if v.is_floating_point() else v
+              for k, v in local[0].items()}
+    fed.combine([filled], pidx, [0])
+    w = fed.global_parameters['layer1.0.conv1.weight']  # (64,64,3,3)
+    assert torch.allclose(w[:32, :32], torch.full_like(w[:32, :32], 5.0))
+    assert torch.equal(w[32:], before['layer1.0.conv1.weight'][32:])
+    assert torch.equal(w[:32, 32:], before['layer1.0.conv1.weight'][:32, 32:])
