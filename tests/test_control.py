@@ -327,3 +327,45 @@ def test_wikitext_raw_parser(tmp_path):
     assert ds['train'].token.numel() == 8
     # unseen-at-train token maps to <unk>... 'x' was added during test read
     assert ds['test'].token.numel() == 3
+
+
+def test_noniid_split_properties():
+    """non-iid-2: every user's train indices carry only their 2 assigned
+    classes, and the train indices partition the kept samples
+    (reference: src/data.py:79-110)."""
+    import torch
+    from heterofl_amd.data import fetch_dataset, split_dataset
+    torch.manual_seed(0)
+    ds = fetch_dataset('MNIST', synthetic=True, synthetic_size=400)
+    data_split, label_split = split_dataset(ds, 10, 'non-iid-2',
+                                            classes_size=10)
+    all_idx = []
+    for u in range(10):
+        labels = {int(ds['train'].target[i]) for i in data_split['train'][u]}
+        assert labels <= set(label_split[u]), (u, labels, label_split[u])
+        assert len(label_split[u]) <= 2
+        all_idx.extend(data_split['train'][u])
+    assert len(all_idx) == len(set(all_idx))   # disjoint shards
+    # 2 shards x 10 users / 10 classes = 2 shards per class -> full coverage
+    assert len(all_idx) == len(ds['train'])
+    # test split follows the SAME label assignment
+    for u in range(10):
+        tl = {int(ds['test'].target[i]) for i in data_split['test'][u]}
+        assert tl <= set(label_split[u]), u
+
+
+def test_iid_split_properties():
+    """iid: equal disjoint shards; label_split records observed labels
+    (reference: src/data.py:61-76)."""
+    import torch
+    from heterofl_amd.data import fetch_dataset, split_dataset
+    torch.manual_seed(0)
+    ds = fetch_dataset('MNIST', synthetic=True, synthetic_size=100)
+    data_split, label_split = split_dataset(ds, 4, 'iid', classes_size=10)
+    sizes = [len(data_split['train'][u]) for u in range(4)]
+    assert sizes == [25, 25, 25, 25]
+    seen = sum((data_split['train'][u] for u in range(4)), [])
+    assert len(seen) == len(set(seen))
+    for u in range(4):
+        labels = {int(ds['train'].target[i]) for i in data_split['train'][u]}
+        assert labels == set(label_split[u])
