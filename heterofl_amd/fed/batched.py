@@ -184,31 +184,60 @@ class BBlock(nn.Module):
         return self.conv2(self.n2(out), residual=shortcut)
 
 
+class BBottleneck(nn.Module):
+    """Pre-activation Bottleneck for R grouped clients (local counterpart:
+    models/resnet.py:47-72; reference: src/models/resnet.py:53-101).  The
+    1x1 convs ride the grouped-conv GEMM fast path; the residual is fused
+    into conv3's epilogue on the native path."""
+    expansion = 4
+
+    def __init__(self, R, in_planes, planes, stride, rate, norm, scale):
+        super().__init__()
+        self.n1 = BNormReLU(R, in_planes, norm, rate, scale)
+        self.conv1 = BConv2d(R, in_planes, planes, 1, 1, 0, bias=False)
+        self.n2 = BNormReLU(R, planes, norm, rate, scale)
+        self.conv2 = BConv2d(R, planes, planes, 3, stride, 1, bias=False)
+        self.n3 = BNormReLU(R, planes, norm, rate, scale)
+        self.conv3 = BConv2d(R, planes, self.expansion * planes, 1, 1, 0,
+                             bias=False)
+        if stride != 1 or in_planes != self.expansion * planes:
+            self.shortcut = BConv2d(R, in_planes, self.expansion * planes,
+                                    1, stride, 0, bias=False)
+
+    def forward(self, x):
+        out = self.n1(x)
+        shortcut = self.shortcut(out) if hasattr(self, 'shortcut') else x
+        out = self.conv1(out)
+        out = self.conv2(self.n2(out))
+        return self.conv3(self.n3(out), residual=shortcut)
+
+
 class BatchedResNet(nn.Module):
-    """R same-rate clients' ResNet18/34 in one grouped model.  state_dict
-    keys mirror models.resnet.ResNet with stacked shapes."""
+    """R same-rate clients' ResNet (basic Block or Bottleneck) in one
+    grouped model.  state_dict keys mirror models.resnet.ResNet with
+    stacked shapes."""
 
     def __init__(self, R, data_shape, hidden_size, num_blocks, num_classes,
-                 rate, norm, scale):
+                 rate, norm, scale, block=BBlock):
         super().__init__()
         self.R = R
         self.num_classes = num_classes
         self.data_ch = data_shape[0]
         self.in_planes = hidden_size[0]
         self.conv1 = BConv2d(R, data_shape[0], hidden_size[0], 3, 1, 1, bias=False)
-        self.layer1 = self._make_layer(R, hidden_size[0], num_blocks[0], 1, rate, norm, scale)
-        self.layer2 = self._make_layer(R, hidden_size[1], num_blocks[1], 2, rate, norm, scale)
-        self.layer3 = self._make_layer(R, hidden_size[2], num_blocks[2], 2, rate, norm, scale)
-        self.layer4 = self._make_layer(R, hidden_size[3], num_blocks[3], 2, rate, norm, scale)
-        self.n4 = BNormReLU(R, hidden_size[3], norm, rate, scale)
-        self.linear = BLinear(R, hidden_size[3], num_classes)
+        self.layer1 = self._make_layer(R, block, hidden_size[0], num_blocks[0], 1, rate, norm, scale)
+        self.layer2 = self._make_layer(R, block, hidden_size[1], num_blocks[1], 2, rate, norm, scale)
+        self.layer3 = self._make_layer(R, block, hidden_size[2], num_blocks[2], 2, rate, norm, scale)
+        self.layer4 = self._make_layer(R, block, hidden_size[3], num_blocks[3], 2, rate, norm, scale)
+        self.n4 = BNormReLU(R, hidden_size[3] * block.expansion, norm, rate, scale)
+        self.linear = BLinear(R, hidden_size[3] * block.expansion, num_classes)
 
-    def _make_layer(self, R, planes, num_blocks, stride, rate, norm, scale):
+    def _make_layer(self, R, block, planes, num_blocks, stride, rate, norm, scale):
         strides = [stride] + [1] * (num_blocks - 1)
         layers = []
         for s in strides:
-            layers.append(BBlock(R, self.in_planes, planes, s, rate, norm, scale))
-            self.in_planes = planes
+            layers.append(block(R, self.in_planes, planes, s, rate, norm, scale))
+            self.in_planes = planes * block.expansion
         return nn.Sequential(*layers)
 
     def forward(self, x):
@@ -462,10 +491,14 @@ class BatchedClientTrainer:
             scaler_rate = rate / cfg['global_model_rate']
             if 'resnet' in name:
                 hidden = [int(np.ceil(rate * h)) for h in cfg['resnet']['hidden_size']]
-                nb = {'resnet18': [2, 2, 2, 2], 'resnet34': [3, 4, 6, 3]}[name]
+                nb, blk = {'resnet18': ([2, 2, 2, 2], BBlock),
+                           'resnet34': ([3, 4, 6, 3], BBlock),
+                           'resnet50': ([3, 4, 6, 3], BBottleneck),
+                           'resnet101': ([3, 4, 23, 3], BBottleneck),
+                           'resnet152': ([3, 8, 36, 3], BBottleneck)}[name]
                 model = BatchedResNet(R, cfg['data_shape'], hidden, nb,
                                       cfg['classes_size'], scaler_rate,
-                                      cfg['norm'], cfg['scale'])
+                                      cfg['norm'], cfg['scale'], block=blk)
             elif name == 'conv':
                 hidden = [int(np.ceil(rate * h)) for h in cfg['conv']['hidden_size']]
                 model = BatchedConv(R, cfg['data_shape'], hidden,
