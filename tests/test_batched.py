@@ -5,6 +5,7 @@ The grouped-model trainer must reproduce per-client training exactly
 per-client clip and SGD decompose per client.
 """
 import numpy as np
+import pytest
 import torch
 import torch.nn.functional as F
 
@@ -320,3 +321,45 @@ def test_batched_bottleneck_engine_opt_in(base_cfg, monkeypatch):
     runner = FedRunner(cfg, ds, data_split, label_split, model, opt)
     assert isinstance(runner.trainer, BatchedClientTrainer)
     runner.train_round(1)
+
+
+@pytest.mark.parametrize('norm', ['gn', 'in', 'ln', 'none'])
+def test_batched_step_equivalence_norms(base_cfg, norm):
+    """Batched == sequential for every norm flavor (gn is BASELINE config 3;
+    in/ln/none are the ablation grid, reference: src/make_ablation.py)."""
+    cfg = make_cfg(base_cfg, f'1_3_1_iid_fix_b1_{norm}_1_1',
+                   data_name='CIFAR10', model_name='resnet18')
+    cfg['global_model_rate'] = 1.0
+    rate, lr, steps, n = 0.5, 0.1, 2, 4
+    locals_ = _local_models(cfg, rate)
+    torch.manual_seed(0)
+    data = [torch.randn(n, R, 3, 32, 32) for _ in range(steps)]
+    labels = [torch.randint(0, CLASSES, (n, R)) for _ in range(steps)]
+    masks = torch.ones(R, CLASSES)
+    seq_states = []
+    for r in range(R):
+        d = [data[s][:, r] for s in range(steps)]
+        l = [labels[s][:, r] for s in range(steps)]
+        seq_states.append(_train_sequential(locals_[r], d, l,
+                                            torch.arange(CLASSES), lr, steps))
+    locals2 = _local_models(cfg, rate)
+    bm = _batched_resnet(cfg, rate)
+    pack_states(bm, [m.state_dict() for m in locals2])
+    bm.train(True)
+    params = list(bm.parameters())
+    opt = torch.optim.SGD(params, lr=lr, momentum=0.9, weight_decay=5e-4)
+    for s in range(steps):
+        xb = data[s].reshape(n, R * 3, 32, 32)
+        opt.zero_grad()
+        scores = bm(xb)
+        losses = batched_masked_ce(scores, labels[s], masks)
+        losses.sum().backward()
+        per_client_clip_(params, R, 1.0)
+        opt.step()
+    outs = unpack_states(bm, list(seq_states[0].keys()))
+    for r in range(R):
+        for k in seq_states[r]:
+            a, b = seq_states[r][k], outs[r][k]
+            diff = (a - b).abs().max().item()
+            scale = a.abs().max().item() + 1e-8
+            assert diff / max(scale, 1.0) < 5e-4, (r, k, diff, scale)
