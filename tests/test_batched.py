@@ -323,11 +323,14 @@ def test_batched_bottleneck_engine_opt_in(base_cfg, monkeypatch):
     runner.train_round(1)
 
 
-@pytest.mark.parametrize('norm', ['gn', 'in', 'ln', 'none'])
-def test_batched_step_equivalence_norms(base_cfg, norm):
-    """Batched == sequential for every norm flavor (gn is BASELINE config 3;
-    in/ln/none are the ablation grid, reference: src/make_ablation.py)."""
-    cfg = make_cfg(base_cfg, f'1_3_1_iid_fix_b1_{norm}_1_1',
+@pytest.mark.parametrize('norm,scale', [
+    ('gn', '1'), ('in', '1'), ('ln', '1'), ('none', '1'),
+    ('gn', '0'), ('none', '0')])
+def test_batched_step_equivalence_norms(base_cfg, norm, scale):
+    """Batched == sequential for every norm flavor and scaler setting
+    (gn is BASELINE config 3; in/ln/none and scaler-off are the ablation
+    grid, reference: src/make_ablation.py)."""
+    cfg = make_cfg(base_cfg, f'1_3_1_iid_fix_b1_{norm}_{scale}_1',
                    data_name='CIFAR10', model_name='resnet18')
     cfg['global_model_rate'] = 1.0
     rate, lr, steps, n = 0.5, 0.1, 2, 4
