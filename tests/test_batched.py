@@ -366,3 +366,57 @@ def test_batched_step_equivalence_norms(base_cfg, norm, scale):
             diff = (a - b).abs().max().item()
             scale = a.abs().max().item() + 1e-8
             assert diff / max(scale, 1.0) < 5e-4, (r, k, diff, scale)
+
+
+def test_batched_lm_matches_sequential_halfwidth(base_cfg):
+    """Half-width LM clients (per-head sliced q/k/v, Federation.distribute
+    shapes): BatchedTransformer == sequential per-client training."""
+    from heterofl_amd.data import fetch_dataset, split_dataset
+    from heterofl_amd.fed import Federation
+    from heterofl_amd.fed.batched_lm_trainer import BatchedLMClientTrainer
+    from heterofl_amd.fed.sequential import SequentialClientTrainer
+    from heterofl_amd.data import SplitDataset, BatchDataset
+    from heterofl_amd.utils import process_dataset
+
+    cfg = make_cfg(base_cfg, '1_3_1_iid_fix_a1_bn_1_1',
+                   data_name='WikiText2', model_name='transformer')
+    cfg['num_epochs'] = {'global': 1, 'local': 1}
+    cfg['transformer']['dropout'] = 0.0
+    cfg['mask_rate'] = 0.0
+    cfg['metric_name'] = {'train': {'Local': ['Local-Loss']}}
+    torch.manual_seed(0)
+    ds = fetch_dataset('WikiText2', synthetic=True, synthetic_size=40_000)
+    ds['train'].vocab.itos = ds['train'].vocab.itos[:500]
+    ds['train'].token = ds['train'].token % 500
+    ds['test'].token = ds['test'].token % 500
+    process_dataset(ds, cfg)
+    cfg['num_tokens'] = 500
+    data_split, label_split = split_dataset(ds, 3, 'iid')
+    torch.manual_seed(1)
+    global_model = make_model(cfg)
+    user_idx = [0, 1, 2]
+    rates = {u: 0.5 for u in user_idx}
+    fed = Federation(global_model.state_dict(), [0.5] * 3, label_split, cfg)
+    local_parameters, _ = fed.distribute(user_idx)
+
+    def make_loader(user):
+        return BatchDataset(SplitDataset(ds['train'],
+                                         data_split['train'][user]),
+                            cfg['bptt'])
+
+    seq = SequentialClientTrainer(cfg)
+    torch.manual_seed(5)
+    seq_out = dict(seq.train_clients(
+        [0, 1, 2], user_idx, [dict(l) for l in local_parameters], rates,
+        make_loader, label_split, 0.1))
+    bt = BatchedLMClientTrainer(cfg)
+    bt.set_data(ds, data_split)
+    torch.manual_seed(5)
+    bt_out = dict(bt.train_clients(
+        [0, 1, 2], user_idx, [dict(l) for l in local_parameters], rates,
+        make_loader, label_split, 0.1))
+    for m in range(3):
+        for k in seq_out[m]:
+            a, b = seq_out[m][k].float(), bt_out[m][k].float()
+            diff = (a - b).abs().max().item()
+            assert diff < 5e-4, (m, k, diff)
