@@ -657,3 +657,70 @@ def test_stats_and_eval_paths_gpu(base_cfg):
     runner.test(tm, 1)
     logger.safe(False)
     assert 'test/Global-Accuracy' in logger.mean
+
+
+@needs_gpu
+def test_mfma_conv_bottleneck_shapes():
+    """Bottleneck-specific conv geometries (1x1 s1 with 4x expansion /
+    contraction, wide 1x1 s2 shortcuts) vs F.conv2d — the shapes the
+    grouped BBottleneck engine adds over resnet18/34
+    (fed/batched.py::BBottleneck; ROUND2.md item 6)."""
+    import torch.nn.functional as F
+    from heterofl_amd.ops.fused import grouped_conv
+    torch.manual_seed(0)
+    shapes = [
+        # (G, N, Cin, H, Cout, k, stride, pad)
+        (3, 10, 64, 32, 256, 1, 1, 0),    # conv3 expansion (L1)
+        (3, 10, 256, 32, 64, 1, 1, 0),    # conv1 contraction (L1)
+        (3, 10, 256, 16, 512, 1, 2, 0),   # shortcut s2 (L2)
+        (2, 10, 512, 4, 2048, 1, 1, 0),   # conv3 expansion (L4)
+        (2, 10, 2048, 4, 512, 1, 1, 0),   # conv1 contraction (L4)
+        (3, 10, 4, 32, 16, 1, 1, 0),      # rate-1/16 ragged expansion
+    ]
+    for G, N, Cin, H, Cout, k, s, p in shapes:
+        x = torch.randn(N, G * Cin, H, H, device='cuda', requires_grad=True)
+        w = (torch.randn(G * Cout, Cin, k, k, device='cuda') * 0.1
+             ).requires_grad_()
+        y = grouped_conv(x, w, None, G, s, p)
+        x2 = x.detach().clone().requires_grad_(True)
+        w2 = w.detach().clone().requires_grad_(True)
+        ref = F.conv2d(x2, w2, None, stride=s, padding=p, groups=G)
+        ferr = (y - ref).abs().max().item()
+        assert ferr < 5e-4 * Cin * k, ('fwd', G, Cin, H, Cout, k, s, ferr)
+        g = torch.randn_like(y)
+        y.backward(g)
+        ref.backward(g)
+        derr = (x.grad - x2.grad).abs().max().item()
+        werr = (w.grad - w2.grad).abs().max().item()
+        assert derr < 5e-4 * Cout * k, ('bwd_data', G, Cin, H, Cout, s, derr)
+        assert werr < 5e-3 * N * H, ('bwd_w', G, Cin, H, Cout, s, werr)
+
+
+@needs_gpu
+def test_batched_bottleneck_block_gpu_matches_cpu():
+    """BBottleneck forward/backward on the native GPU path vs the same
+    module's CPU (torch oracle) path."""
+    from heterofl_amd.fed.batched import BBottleneck
+    torch.manual_seed(0)
+    for stride in (1, 2):
+        blk = BBottleneck(3, 32, 16, stride, 1.0, 'bn', True)
+        blk_gpu = BBottleneck(3, 32, 16, stride, 1.0, 'bn', True).cuda()
+        blk_gpu.load_state_dict(blk.state_dict())
+        blk.train(True)
+        blk_gpu.train(True)
+        x = torch.randn(8, 3 * 32, 16, 16, requires_grad=True)
+        xg = x.detach().clone().cuda().requires_grad_(True)
+        y = blk(x)
+        yg = blk_gpu(xg)
+        ferr = (y - yg.cpu()).abs().max().item()
+        assert ferr < 2e-2, (stride, ferr)
+        g = torch.randn_like(y)
+        y.backward(g)
+        yg.backward(g.cuda())
+        derr = (x.grad - xg.grad.cpu()).abs().max().item()
+        assert derr < 2e-2, (stride, derr)
+        for (k, p), (kg, pg) in zip(blk.named_parameters(),
+                                    blk_gpu.named_parameters()):
+            perr = (p.grad - pg.grad.cpu()).abs().max().item()
+            scale = p.grad.abs().max().item() + 1e-6
+            assert perr / max(scale, 1.0) < 5e-2, (stride, k, perr, scale)
