@@ -63,12 +63,35 @@ def run_fed_experiment(cfg, pivot_metric, pivot_sign, metric_name):
     cfg['pivot_metric'] = pivot_metric
     cfg['metric_name'] = metric_name
     process_control(cfg)
+    _apply_round_cap(cfg)
     seeds = list(range(cfg['init_seed'], cfg['init_seed'] + cfg['num_experiments']))
     for seed in seeds:
         cfg['model_tag'] = model_tag_of(seed, cfg)
         cfg['pivot'] = -float('inf') * pivot_sign if pivot_sign > 0 else float('inf')
         print('Experiment: {}'.format(cfg['model_tag']))
         _run_one(cfg, seed, pivot_sign)
+
+
+def _synthetic_size(default=None):
+    """HETEROFL_SYNTHETIC_SIZE shrinks synthetic datasets for smoke/CI runs
+    of the CLI entries (None = the real dataset's size)."""
+    v = os.environ.get('HETEROFL_SYNTHETIC_SIZE')
+    return int(v) if v else default
+
+
+def _apply_round_cap(cfg):
+    """HETEROFL_MAX_ROUNDS caps the global-round count (smoke/CI runs of the
+    real CLI scripts; the reference has no such knob — round counts come
+    from the dataset tables, src/utils.py:150-212)."""
+    cap = os.environ.get('HETEROFL_MAX_ROUNDS')
+    if not cap:
+        return
+    cap = int(cap)
+    if isinstance(cfg.get('num_epochs'), dict):
+        cfg['num_epochs'] = dict(cfg['num_epochs'],
+                                 **{'global': min(cfg['num_epochs']['global'], cap)})
+    else:
+        cfg['num_epochs'] = min(int(cfg['num_epochs']), cap)
 
 
 def _run_one(cfg, seed, pivot_sign):
@@ -79,7 +102,8 @@ def _run_one(cfg, seed, pivot_sign):
         print('[heterofl_amd] no GPU visible; falling back to cpu')
         cfg['device'] = 'cpu'
     dataset = fetch_dataset(cfg['data_name'], cfg['subset'],
-                            synthetic=cfg.get('synthetic', False))
+                            synthetic=cfg.get('synthetic', False),
+                            synthetic_size=_synthetic_size())
     process_dataset(dataset, cfg)
     model = make_model(cfg, model_rate=cfg['global_model_rate']).to(cfg['device'])
     optimizer = make_optimizer(model, cfg['lr'], cfg)
@@ -139,6 +163,7 @@ def run_centralized_experiment(cfg, pivot_metric, pivot_sign, metric_name):
     cfg['pivot_metric'] = pivot_metric
     cfg['metric_name'] = metric_name
     process_control(cfg)
+    _apply_round_cap(cfg)
     seeds = list(range(cfg['init_seed'], cfg['init_seed'] + cfg['num_experiments']))
     for seed in seeds:
         cfg['model_tag'] = model_tag_of(seed, cfg)
@@ -148,7 +173,8 @@ def run_centralized_experiment(cfg, pivot_metric, pivot_sign, metric_name):
         if cfg['device'].startswith('cuda') and not torch.cuda.is_available():
             cfg['device'] = 'cpu'
         dataset = fetch_dataset(cfg['data_name'], cfg['subset'],
-                                synthetic=cfg.get('synthetic', False))
+                                synthetic=cfg.get('synthetic', False),
+                                synthetic_size=_synthetic_size())
         process_dataset(dataset, cfg)
         model = make_model(cfg, model_rate=cfg['global_model_rate']).to(cfg['device'])
         optimizer = make_optimizer(model, cfg['lr'], cfg)
@@ -256,7 +282,8 @@ def run_centralized_eval(cfg, metric_name):
         if cfg['device'].startswith('cuda') and not torch.cuda.is_available():
             cfg['device'] = 'cpu'
         dataset = fetch_dataset(cfg['data_name'], cfg['subset'],
-                                synthetic=cfg.get('synthetic', False))
+                                synthetic=cfg.get('synthetic', False),
+                                synthetic_size=_synthetic_size())
         process_dataset(dataset, cfg)
         model = make_model(cfg, model_rate=cfg['global_model_rate']).to(cfg['device'])
         last_epoch, _, _, model, _, _, train_logger = resume(
@@ -308,7 +335,8 @@ def run_fed_eval(cfg, metric_name, result_key='test'):
         print('Eval: {}'.format(cfg['model_tag']))
         torch.manual_seed(seed)
         dataset = fetch_dataset(cfg['data_name'], cfg['subset'],
-                                synthetic=cfg.get('synthetic', False))
+                                synthetic=cfg.get('synthetic', False),
+                                synthetic_size=_synthetic_size())
         process_dataset(dataset, cfg)
         if cfg['device'].startswith('cuda') and not torch.cuda.is_available():
             cfg['device'] = 'cpu'

@@ -125,7 +125,7 @@ def test_checkpoint_resume_roundtrip(base_cfg, tmp_path, monkeypatch):
     orig = data_mod.fetch_dataset
     monkeypatch.setattr(
         'heterofl_amd.entry.fetch_dataset',
-        lambda name, subset=None, synthetic=False: orig(
+        lambda name, subset=None, synthetic=False, **kw: orig(
             name, subset, synthetic=True, synthetic_size=40))
     orig_pc = entry_mod.process_control
 
@@ -206,7 +206,7 @@ def test_centralized_entry(base_cfg, tmp_path, monkeypatch):
     orig_fetch = data_mod.fetch_dataset
     monkeypatch.setattr(
         entry, 'fetch_dataset',
-        lambda name, subset=None, synthetic=False: orig_fetch(
+        lambda name, subset=None, synthetic=False, **kw: orig_fetch(
             name, subset, synthetic=True, synthetic_size=40))
     cfg = default_config()
     cfg.update({'data_name': 'MNIST', 'model_name': 'conv', 'device': 'cpu',
@@ -251,7 +251,7 @@ def test_centralized_lm_entry(base_cfg, tmp_path, monkeypatch):
     orig_fetch = data_mod.fetch_dataset
     monkeypatch.setattr(
         entry, 'fetch_dataset',
-        lambda name, subset=None, synthetic=False: orig_fetch(
+        lambda name, subset=None, synthetic=False, **kw: orig_fetch(
             name, subset, synthetic=True, synthetic_size=400))
     cfg = default_config()
     cfg.update({'data_name': 'WikiText2', 'subset': 'label',
@@ -329,3 +329,27 @@ def test_golden_path_trajectory(base_cfg):
     golden = [2.297021, 2.291516, 2.288667]
     for got, want in zip(traj, golden):
         assert abs(got - want) < 1e-4, (traj, golden)
+
+
+def test_cli_train_classifier_fed_subprocess(tmp_path):
+    """The real train_classifier_fed.py CLI (argv -> checkpoint) runs
+    end-to-end; HETEROFL_MAX_ROUNDS caps the round count for CI."""
+    import json
+    import os
+    import subprocess
+    import sys
+    root = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    env = dict(os.environ, HETEROFL_MAX_ROUNDS='2', PYTHONPATH=root,
+               HETEROFL_SYNTHETIC_SIZE='40')
+    out = subprocess.run(
+        [sys.executable, os.path.join(root, 'train_classifier_fed.py'),
+         '--data_name', 'MNIST', '--model_name', 'conv',
+         '--control_name', '1_4_0.5_iid_fix_a1_bn_1_1',
+         '--device', 'cpu', '--synthetic', '1'],
+        cwd=str(tmp_path), capture_output=True, text=True, timeout=900,
+        env=env)
+    assert out.returncode == 0, out.stderr[-2000:]
+    tag = '0_MNIST_label_conv_1_4_0.5_iid_fix_a1_bn_1_1'
+    assert (tmp_path / 'output' / 'model' / f'{tag}_checkpoint.pt').exists(), \
+        out.stdout[-2000:]
+    assert (tmp_path / 'output' / 'model' / f'{tag}_best.pt').exists()
