@@ -162,6 +162,21 @@ class FedRunner:
                 distributed_combine(self.federation, trained, param_idx,
                                     user_idx, self.dist_ctx)
         self.global_model.load_state_dict(self.federation.global_parameters)
+        if self.logger is not None:
+            # per-round console line (the engine-level equivalent of the
+            # reference's per-batch Train Epoch/ETA prints,
+            # src/train_classifier_fed.py:108-119 — the graphed inner loop
+            # has no per-batch host involvement to log from)
+            total = cfg['num_epochs']['global'] if \
+                isinstance(cfg['num_epochs'], dict) else cfg['num_epochs']
+            info = {'info': ['Model: {}'.format(cfg.get('model_tag', '')),
+                             'Train Epoch: {}({:.0f}%)'.format(
+                                 epoch, 100.0 * epoch / max(total, 1)),
+                             'Learning rate: {:.4g}'.format(lr)]}
+            self.logger.append(info, 'train', mean=False)
+            flat = [m for group in cfg['metric_name']['train'].values()
+                    for m in group]
+            self.logger.write('train', flat)
         return user_idx
 
     # ------------------------------------------------------------------ stats
