@@ -133,7 +133,16 @@ def _run_one(cfg, seed, pivot_sign):
         if cfg['scheduler_name'] == 'ReduceLROnPlateau':
             scheduler.step(metrics=logger.mean['train/{}'.format(cfg['pivot_metric'])])
         else:
-            scheduler.step()
+            # the server-side optimizer never steps by design (it only
+            # carries the lr the scheduler decays; local optimizers are
+            # fresh per client per round — reference
+            # src/train_classifier_fed.py:79-82,195) so torch's
+            # "step() before lr_scheduler.step()" warning does not apply
+            import warnings
+            with warnings.catch_warnings():
+                warnings.filterwarnings(
+                    'ignore', message='.*lr_scheduler.step.*')
+                scheduler.step()
         logger.safe(False)
         save_result = {
             'cfg': cfg, 'epoch': epoch + 1, 'data_split': data_split,
