@@ -325,8 +325,11 @@ def test_wikitext_raw_parser(tmp_path):
     ds = fetch_dataset('WikiText2', root=str(tmp_path))
     assert len(ds['train'].vocab) >= 6   # <unk> + a b c d e
     assert ds['train'].token.numel() == 8
-    # unseen-at-train token maps to <unk>... 'x' was added during test read
+    # vocab is train-built (reference: src/datasets/lm.py:157-164);
+    # unseen-at-train 'x' maps to <unk>
     assert ds['test'].token.numel() == 3
+    unk = ds['train'].vocab['<unk>']
+    assert int(ds['test'].token[2]) == unk
 
 
 def test_noniid_split_properties():
