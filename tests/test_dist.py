@@ -237,3 +237,34 @@ def test_bench_torchrun_contract(tmp_path):
     assert rec['config']['num_users'] == 200   # weak scaling: 100 users/GPU
     assert rec['config']['active_clients'] == 20
     assert rec['value'] > 0
+
+
+def _collective_worker(rank, world, port, tmpdir):
+    os.environ['MASTER_ADDR'] = '127.0.0.1'
+    os.environ['MASTER_PORT'] = str(port)
+    torch.distributed.init_process_group('gloo', rank=rank, world_size=world)
+    from heterofl_amd.parallel import init_distributed
+    from heterofl_amd.parallel.dist import (broadcast_state_dict,
+                                            allreduce_bn_stats)
+    ctx = init_distributed(backend='gloo')
+    # broadcast: rank 0's values win
+    sd = {'w': torch.full((4, 3), float(rank)),
+          'b': torch.full((5,), float(rank) + 10),
+          'n': torch.tensor(rank, dtype=torch.long)}  # int: untouched
+    broadcast_state_dict(sd, ctx, src=0)
+    assert torch.equal(sd['w'], torch.zeros(4, 3)), rank
+    assert torch.equal(sd['b'], torch.full((5,), 10.0)), rank
+    assert sd['n'].item() == rank   # non-float buffers are not broadcast
+    # allreduce: sums partials across ranks
+    parts = [torch.full((3,), float(rank + 1)), torch.tensor([2.0 * rank])]
+    out = allreduce_bn_stats(parts, ctx)
+    assert torch.equal(out[0], torch.full((3,), 3.0)), rank   # 1 + 2
+    assert torch.equal(out[1], torch.tensor([2.0])), rank     # 0 + 2
+    torch.distributed.destroy_process_group()
+
+
+def test_collective_helpers():
+    """broadcast_state_dict / allreduce_bn_stats utilities (C1/C2 helpers;
+    the round engine itself needs no W_g broadcast — every rank finalizes
+    identical global params from the deterministic padded combine)."""
+    mp.spawn(_collective_worker, args=(2, 29561, ''), nprocs=2, join=True)
