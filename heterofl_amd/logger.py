@@ -1,14 +1,17 @@
-"""Running-mean tracker with per-key counters, epoch history snapshots and
-console line assembly (reference: src/logger.py:8-87).
+"""Metric logger: per-key weighted series with round snapshots and console
+line assembly.
 
-The reference writes TensorBoard summaries; tensorboard is optional here —
-when it is absent the Logger writes a JSONL event stream instead, so the
-object stays picklable into checkpoints exactly like the reference
-(reference: src/train_classifier_fed.py:85-88).
+Plays the role of the reference's Logger (src/logger.py:8-87) and keeps its
+*external* contract — `append/write/safe/reset/flush`, a `mean` mapping, a
+`history` of per-round snapshots, and picklability into checkpoints
+(src/train_classifier_fed.py:85-88) — but the internals are a different
+design: each key accumulates a weighted sum and a total weight, and the mean
+is derived on demand instead of being maintained by an incremental running
+update.  TensorBoard is optional; without it scalars go to a JSONL event
+stream so runs stay inspectable in no-network environments.
 """
 import json
 import os
-from collections import defaultdict
 
 from .utils.core import makedir_exist_ok
 
@@ -20,77 +23,119 @@ except Exception:
     _HAS_TB = False
 
 
+class _MeanView(dict):
+    """Dict of derived means; missing keys read as 0 (callers probe pivot
+    metrics that may not have been logged yet)."""
+
+    def __missing__(self, key):
+        return 0
+
+
+def _is_scalar(x):
+    return isinstance(x, (int, float)) and not isinstance(x, bool)
+
+
 class Logger:
     def __init__(self, log_path):
         self.log_path = log_path
         self.writer = None
-        self.tracker = defaultdict(int)
-        self.counter = defaultdict(int)
-        self.mean = defaultdict(int)
-        self.history = defaultdict(list)
-        self.iterator = defaultdict(int)
+        self._wsum = {}    # key -> sum of weight*value over appends
+        self._wtot = {}    # key -> sum of weights
+        self.tracker = {}  # key -> most recent raw appended value (any type)
+        self.history = {}  # key -> [snapshot mean per round]
+        self._step = {}    # key -> scalar event counter for the writer
 
+    # -------------------------------------------------------------- derived
+    @property
+    def mean(self):
+        """Weighted mean per key, derived from the accumulated series."""
+        view = _MeanView()
+        for k, s in self._wsum.items():
+            w = self._wtot.get(k, 0)
+            if w:
+                view[k] = s / w
+        return view
+
+    # -------------------------------------------------------------- rounds
     def safe(self, write):
-        """Bracket a round: open the writer on True, snapshot means into
-        history and close on False (reference: src/logger.py:18-27)."""
+        """Bracket a round: True opens the event writer; False closes it and
+        snapshots every key's current mean into `history`."""
         if write:
-            self.writer = SummaryWriter(self.log_path) if _HAS_TB else _Jsonl(self.log_path)
-        else:
-            if self.writer is not None:
-                self.writer.close()
-                self.writer = None
-            for name in self.mean:
-                self.history[name].append(self.mean[name])
-        return
+            if _HAS_TB:
+                self.writer = SummaryWriter(self.log_path)
+            else:
+                self.writer = _Jsonl(self.log_path)
+            return
+        if self.writer is not None:
+            self.writer.close()
+            self.writer = None
+        for k, m in self.mean.items():
+            self.history.setdefault(k, []).append(m)
 
     def reset(self):
-        self.tracker = defaultdict(int)
-        self.counter = defaultdict(int)
-        self.mean = defaultdict(int)
-        return
+        self._wsum = {}
+        self._wtot = {}
+        self.tracker = {}
 
+    # -------------------------------------------------------------- logging
     def append(self, result, tag, n=1, mean=True):
-        for k in result:
-            name = '{}/{}'.format(tag, k)
-            self.tracker[name] = result[k]
-            if mean and isinstance(result[k], (int, float)):
-                self.counter[name] += n
-                self.mean[name] = ((self.counter[name] - n) * self.mean[name]
-                                   + n * result[k]) / self.counter[name]
-        return
+        """Record a batch of named values under `tag/`; scalars with
+        mean=True join the weighted series with weight n (= batch size)."""
+        for name, value in result.items():
+            key = '{}/{}'.format(tag, name)
+            self.tracker[key] = value
+            if mean and _is_scalar(value):
+                self._wsum[key] = self._wsum.get(key, 0.0) + n * value
+                self._wtot[key] = self._wtot.get(key, 0) + n
 
     def write(self, tag, metric_names):
-        names = ['{}/{}'.format(tag, k) for k in metric_names]
-        evaluation_info = []
-        for name in names:
-            tag_, k = name.split('/')
-            if isinstance(self.tracker[name], (int, float)):
-                s = self.mean[name]
-                evaluation_info.append('{}: {:.4f}'.format(k, s))
-                if self.writer is not None:
-                    self.iterator[name] += 1
-                    self.writer.add_scalar(name, s, self.iterator[name])
-            elif isinstance(self.tracker[name], list) and self.tracker[name]:
-                evaluation_info.append('{}: {}'.format(k, self.tracker[name][0]))
-        info_name = '{}/info'.format(tag)
-        info = self.tracker[info_name]
-        if isinstance(info, dict) and 'info' in info:
-            info = info['info']
+        """Emit one console line (info prefix + current means) and push each
+        numeric mean to the event writer."""
+        parts = []
+        info = self.tracker.get('{}/info'.format(tag))
+        if isinstance(info, dict):
+            info = info.get('info')
         if isinstance(info, list):
-            print(' | '.join(info[:3] + evaluation_info))
-        else:
-            print(' | '.join(evaluation_info))
-        return
+            parts.extend(str(x) for x in info[:3])
+        means = self.mean
+        for name in metric_names:
+            key = '{}/{}'.format(tag, name)
+            raw = self.tracker.get(key)
+            if _is_scalar(raw):
+                m = means[key]
+                parts.append('{}: {:.4f}'.format(name, m))
+                if self.writer is not None:
+                    step = self._step.get(key, 0) + 1
+                    self._step[key] = step
+                    self.writer.add_scalar(key, m, step)
+            elif isinstance(raw, list) and raw:
+                parts.append('{}: {}'.format(name, raw[0]))
+        print(' | '.join(parts))
 
     def flush(self):
         if self.writer is not None:
             self.writer.flush()
 
-    # keep the pickled checkpoint logger usable (writer handles are dropped)
+    # ------------------------------------------------------------- pickling
     def __getstate__(self):
-        state = self.__dict__.copy()
+        # writer handles never survive the checkpoint
+        state = dict(self.__dict__)
         state['writer'] = None
         return state
+
+    def __setstate__(self, state):
+        # migrate checkpoints pickled by the earlier running-mean layout
+        # (mean/counter/iterator attributes) into the series form
+        if '_wsum' not in state:
+            counter = state.pop('counter', {}) or {}
+            old_mean = state.pop('mean', {}) or {}
+            state['_wsum'] = {k: old_mean.get(k, 0) * w
+                              for k, w in counter.items()}
+            state['_wtot'] = dict(counter)
+            state['_step'] = dict(state.pop('iterator', {}) or {})
+            state.setdefault('tracker', {})
+            state['history'] = dict(state.get('history', {}) or {})
+        self.__dict__.update(state)
 
 
 class _Jsonl:
@@ -101,7 +146,8 @@ class _Jsonl:
         self.f = open(os.path.join(path, 'events.jsonl'), 'a')
 
     def add_scalar(self, name, value, step):
-        self.f.write(json.dumps({'name': name, 'value': value, 'step': step}) + '\n')
+        self.f.write(json.dumps({'name': name, 'value': value,
+                                 'step': step}) + '\n')
 
     def flush(self):
         self.f.flush()
