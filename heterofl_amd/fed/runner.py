@@ -77,13 +77,12 @@ class FedRunner:
                                      label_split, cfg)
         self.is_lm = cfg['model_name'] == 'transformer'
         use_batched = cfg.get('engine', 'sequential') == 'batched'
-        # Bottleneck resnets default to the sequential engine; the grouped
-        # BBottleneck engine (oracle-equivalent on CPU, tests/test_batched.py)
-        # is opt-in until its GPU shapes get a hardware numerics pass
-        # (ROUND2.md item 6)
+        # The grouped BBottleneck engine passed its GPU numerics suite
+        # (GPUTEST_r01) so Bottleneck resnets now use the batched engine by
+        # default; HETEROFL_BATCHED_BOTTLENECK=0 forces the sequential oracle
         if use_batched and cfg['model_name'] in ('resnet50', 'resnet101',
                                                  'resnet152') and \
-                os.environ.get('HETEROFL_BATCHED_BOTTLENECK', '0') != '1':
+                os.environ.get('HETEROFL_BATCHED_BOTTLENECK', '1') == '0':
             use_batched = False
         if use_batched and self.is_lm:
             from .batched_lm_trainer import BatchedLMClientTrainer
