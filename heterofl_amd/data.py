@@ -129,19 +129,30 @@ def _read_idx(path):
     return data.reshape(shape)
 
 
-def _load_mnist_raw(root, split):
+def _load_mnist_raw(root, split, data_name='MNIST'):
     prefix = 'train' if split == 'train' else 't10k'
     raw = os.path.join(root, 'raw')
 
-    def find(stem):
-        for ext in ('', '.gz'):
-            p = os.path.join(raw, stem + ext)
-            if os.path.exists(p):
-                return p
+    def find(stems):
+        for stem in stems:
+            for ext in ('', '.gz'):
+                p = os.path.join(raw, stem + ext)
+                if os.path.exists(p):
+                    return p
         return None
 
-    ip = find(f'{prefix}-images-idx3-ubyte')
-    lp = find(f'{prefix}-labels-idx1-ubyte')
+    img_stems = [f'{prefix}-images-idx3-ubyte']
+    lbl_stems = [f'{prefix}-labels-idx1-ubyte']
+    if data_name == 'EMNIST':
+        # EMNIST raw files are named emnist-<subset>-{train,test}-*
+        # (the reference uses the balanced split, src/datasets/mnist.py)
+        word = 'train' if split == 'train' else 'test'
+        for subset in ('balanced', 'byclass', 'bymerge', 'digits',
+                       'letters', 'mnist'):
+            img_stems.append(f'emnist-{subset}-{word}-images-idx3-ubyte')
+            lbl_stems.append(f'emnist-{subset}-{word}-labels-idx1-ubyte')
+    ip = find(img_stems)
+    lp = find(lbl_stems)
     if ip is None or lp is None:
         return None
     img = torch.from_numpy(_read_idx(ip).copy())
@@ -250,7 +261,7 @@ def fetch_dataset(data_name, subset='label', synthetic=False,
             raw = None
             if not synthetic:
                 if data_name in ('MNIST', 'FashionMNIST', 'EMNIST'):
-                    raw = _load_mnist_raw(root, split)
+                    raw = _load_mnist_raw(root, split, data_name)
                 else:
                     raw = _load_cifar_raw(root, split, data_name)
                 if raw is None:

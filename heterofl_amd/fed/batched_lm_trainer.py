@@ -110,12 +110,15 @@ class BatchedLMClientTrainer:
                     opt.step()
                 if logger is not None:
                     with torch.no_grad():
-                        n = tokens.size(1) * tokens.size(2)
+                        # weight = window row count, the reference's n
+                        # (src/train_transformer_fed.py:168-171)
+                        n = tokens.size(1)
                         for i in range(R):
                             l = losses[i].item()
                             logger.append({'Local-Loss': l,
                                            'Local-Perplexity':
-                                               float(torch.exp(torch.tensor(l)))},
+                                               float(torch.exp(torch.tensor(
+                                                   min(l, 30.0))))},
                                           'train', n=n)
         template_keys = list(locals_list[0].keys())
         states = unpack_states(model, template_keys)
@@ -161,10 +164,11 @@ class BatchedLMClientTrainer:
             m = gs.metrics.detach().cpu()
             for i in range(R):
                 cnt = max(int(m[i, 2].item()), 1)
-                l = m[i, 0].item() / cnt
-                import math as _math
-                logger.append({'Local-Loss': l,
-                               'Local-Perplexity': float(_math.exp(min(l, 30)))},
+                # row-weighted means of per-window loss and exp(loss) — the
+                # reference's aggregation (mean-of-exp, weighted by
+                # input['label'].size(0); src/train_transformer_fed.py:168-171)
+                logger.append({'Local-Loss': m[i, 0].item() / cnt,
+                               'Local-Perplexity': m[i, 1].item() / cnt},
                               'train', n=cnt)
         template_keys = list(locals_list[0].keys())
         states = unpack_states(gs.model, template_keys)

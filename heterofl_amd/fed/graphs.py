@@ -238,8 +238,15 @@ class LMGraphedStep:
             logits = self.model(tokens)
         losses = lm_masked_ce(logits, tokens, self.masks)
         with torch.no_grad():
-            n = tokens.size(1) * tokens.size(2)
-            self.metrics[:, 0] += losses.detach() * n
+            # reference metric aggregation weights each bptt window by its
+            # row count and logs per-window exp(loss)
+            # (src/train_transformer_fed.py:168-171 with logger.append
+            # n=input['label'].size(0)) — accumulate loss*rows, exp(loss)*rows
+            # and rows device-side so the graphed round reproduces it
+            n = tokens.size(1)
+            ld = losses.detach()
+            self.metrics[:, 0] += ld * n
+            self.metrics[:, 1] += torch.exp(ld.clamp(max=30.0)) * n
             self.metrics[:, 2] += n
         if self._native:
             raw = torch.autograd.grad(losses.sum(), self.params)
