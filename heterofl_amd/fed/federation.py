@@ -374,6 +374,70 @@ class Federation:
             cnt_d[k] = cnt
         return tmp_d, cnt_d
 
+    @staticmethod
+    def _count_only(cnt, ps):
+        """cnt[slice] += 1 for one client's parameter (no values)."""
+        if ps is None or ps.out is None:
+            cnt += 1
+            return
+        out, inp = ps.out, ps.inp
+        if inp is not None and cnt.dim() > 1:
+            if out.is_dense_prefix() and inp.is_dense_prefix():
+                cnt[:out.n, :inp.n] += 1
+            elif inp.is_dense_prefix():
+                oidx = out.indices(cnt.device)
+                cnt.narrow(1, 0, inp.n).index_add_(
+                    0, oidx, torch.ones(oidx.numel(), inp.n, dtype=cnt.dtype,
+                                        device=cnt.device))
+            else:
+                oidx = out.indices(cnt.device).unsqueeze(1)
+                iidx = inp.indices(cnt.device).unsqueeze(0)
+                ones = torch.ones(oidx.numel(), iidx.numel(), dtype=cnt.dtype,
+                                  device=cnt.device)
+                cnt.index_put_((oidx, iidx), ones, accumulate=True)
+        else:
+            if out.is_dense_prefix():
+                cnt[:out.n] += 1
+            else:
+                oidx = out.indices(cnt.device)
+                cnt.index_add_(0, oidx, torch.ones(oidx.numel(),
+                                                   dtype=cnt.dtype,
+                                                   device=cnt.device))
+
+    def count_map(self, param_idx, user_idx, slots=None):
+        """The count half of the combine, computed from index maps alone.
+
+        Counts depend only on (param_idx, label_split, user_idx) — never on
+        client parameter VALUES — and every rank builds param_idx for ALL of
+        the round's clients (distribute slices tensors per-slot but index
+        maps are global).  So in the multi-rank combine each rank can derive
+        the full count tensor locally and the collective only has to move
+        the value accumulators: half the payload of reducing counts too,
+        bit-exactly (parallel/dist.py:distributed_combine).
+        """
+        gp = self.global_parameters
+        weight_keys = [k for k in gp if 'weight' in k]
+        bias_keys = [k for k in gp if 'bias' in k]
+        if slots is None:
+            slots = range(len(param_idx))
+        cnt_d = OrderedDict()
+        for k, v in gp.items():
+            ptype = k.split('.')[-1]
+            cnt = torch.zeros(v.size(), dtype=torch.float32, device=v.device)
+            is_output = self._output_layer_kind(k, weight_keys, bias_keys)
+            for m in slots:
+                if 'weight' in ptype or 'bias' in ptype:
+                    ps = param_idx[m][k]
+                    if is_output and ps is not None and ps.out is not None:
+                        label_split = torch.tensor(self.label_split[user_idx[m]])
+                        out_idx = ps.out.indices()[label_split]
+                        ps = ParamSlice(Axis.gather(out_idx), ps.inp)
+                    self._count_only(cnt, ps)
+                else:
+                    cnt += 1
+            cnt_d[k] = cnt
+        return cnt_d
+
     def finalize(self, tmp_d, cnt_d):
         """Masked count-weighted average written in place into the global
         tensors (reference: src/fed.py:217-218)."""

@@ -166,6 +166,8 @@ class FedRunner:
             # reference's per-batch Train Epoch/ETA prints,
             # src/train_classifier_fed.py:108-119 — the graphed inner loop
             # has no per-batch host involvement to log from)
+            if self.dist_ctx is not None:
+                self.logger.sync()  # merge each rank's clients' train metrics
             total = cfg['num_epochs']['global'] if \
                 isinstance(cfg['num_epochs'], dict) else cfg['num_epochs']
             info = {'info': ['Model: {}'.format(cfg.get('model_tag', '')),
@@ -173,9 +175,10 @@ class FedRunner:
                                  epoch, 100.0 * epoch / max(total, 1)),
                              'Learning rate: {:.4g}'.format(lr)]}
             self.logger.append(info, 'train', mean=False)
-            flat = [m for group in cfg['metric_name']['train'].values()
-                    for m in group]
-            self.logger.write('train', flat)
+            if self.dist_ctx is None or self.dist_ctx.is_main:
+                flat = [m for group in cfg['metric_name']['train'].values()
+                        for m in group]
+                self.logger.write('train', flat)
         return user_idx
 
     # ------------------------------------------------------------------ stats
@@ -215,10 +218,15 @@ class FedRunner:
             if self.dist_ctx is not None and self.dist_ctx.world_size > 1:
                 rank, world = self.dist_ctx.rank, self.dist_ctx.world_size
                 bs = scfg['batch_size']['train']
-                # contiguous whole-batch shards so every batch is full-size
-                total_b = len(train) // bs
+                # shard whole batches (incl. the partial tail batch) across
+                # ranks: cumulative-BN (momentum=None) weighs every batch
+                # equally, so giving the tail to exactly one rank and
+                # weighting ranks by batch count reproduces the single-rank
+                # pass exactly — tail included
+                total_b = (len(train) + bs - 1) // bs
                 my_b = list(range(rank, total_b, world))
-                idx = [i for b in my_b for i in range(b * bs, (b + 1) * bs)]
+                idx = [i for b in my_b
+                       for i in range(b * bs, min((b + 1) * bs, len(train)))]
                 train = SplitDataset(train, idx)
             loader = make_data_loader({'train': train}, scfg)['train']
             for input in loader:
@@ -253,18 +261,26 @@ class FedRunner:
         """Per-client Local metrics + Global metrics
         (reference: src/train_classifier_fed.py:141-169).  On GPU the eval
         batch is enlarged (metric means are batch-size invariant under the
-        logger's count weighting) so evaluation is not launch-bound."""
+        logger's count weighting) so evaluation is not launch-bound.
+
+        Multi-rank (C3 of SURVEY §2b): the per-user Local loop is sharded by
+        user and the Global set by sample across ranks; logger.sync() then
+        merges the shards' weighted sums so every rank sees the sequential
+        run's exact means (weighted means are additive over disjoint
+        shards)."""
         cfg = self.cfg
         if torch.cuda.is_available() and not self.is_lm:
             cfg = dict(cfg)
             cfg['batch_size'] = dict(cfg['batch_size'])
             cfg['batch_size']['test'] = max(cfg['batch_size']['test'], 500)
+        rank, world = (0, 1) if self.dist_ctx is None else \
+            (self.dist_ctx.rank, self.dist_ctx.world_size)
         metric = Metric()
         logger = self.logger
         with torch.no_grad():
             test_model.train(False)
             if not self.is_lm:
-                for m in range(cfg['num_users']):
+                for m in range(rank, cfg['num_users'], world):
                     loader = make_data_loader(
                         {'test': SplitDataset(self.dataset['test'],
                                               self.data_split['test'][m])}, cfg)['test']
@@ -278,7 +294,11 @@ class FedRunner:
                                              input, output)
                         if logger:
                             logger.append(ev, 'test', input_size)
-                loader = make_data_loader({'test': self.dataset['test']}, cfg)['test']
+                global_test = self.dataset['test']
+                if world > 1:
+                    global_test = SplitDataset(
+                        global_test, list(range(rank, len(global_test), world)))
+                loader = make_data_loader({'test': global_test}, cfg)['test']
                 for input in loader:
                     input = collate(input)
                     input_size = input['label'].size(0)
@@ -290,7 +310,7 @@ class FedRunner:
                         logger.append(ev, 'test', input_size)
             else:
                 ds = BatchDataset(self.dataset['test'], cfg['bptt'])
-                for i in range(len(ds)):
+                for i in range(rank, len(ds), world):
                     input = ds[i]
                     input_size = input['label'].size(0)
                     input = to_device(input, cfg['device'])
@@ -300,8 +320,12 @@ class FedRunner:
                     if logger:
                         logger.append(ev, 'test', input_size)
             if logger:
+                if self.dist_ctx is not None:
+                    logger.sync()
                 info = {'info': ['Test Epoch: {}({:.0f}%)'.format(epoch, 100.)]}
                 logger.append(info, 'test', mean=False)
-                names = cfg['metric_name']['test']
-                flat = sum(names.values(), []) if isinstance(names, dict) else names
-                logger.write('test', flat)
+                if self.dist_ctx is None or self.dist_ctx.is_main:
+                    names = cfg['metric_name']['test']
+                    flat = sum(names.values(), []) if isinstance(names, dict) \
+                        else names
+                    logger.write('test', flat)

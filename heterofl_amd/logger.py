@@ -37,10 +37,14 @@ def _is_scalar(x):
 
 class Logger:
     def __init__(self, log_path):
+        # log_path=None makes a writer-less logger (non-main ranks in
+        # multi-GPU runs accumulate series for sync() but never write files)
         self.log_path = log_path
         self.writer = None
-        self._wsum = {}    # key -> sum of weight*value over appends
-        self._wtot = {}    # key -> sum of weights
+        self._wsum = {}    # key -> sum of weight*value, this rank, unsynced
+        self._wtot = {}    # key -> sum of weights, this rank, unsynced
+        self._ssum = {}    # key -> already rank-merged sum (see sync())
+        self._stot = {}    # key -> already rank-merged weight
         self.tracker = {}  # key -> most recent raw appended value (any type)
         self.history = {}  # key -> [snapshot mean per round]
         self._step = {}    # key -> scalar event counter for the writer
@@ -48,23 +52,45 @@ class Logger:
     # -------------------------------------------------------------- derived
     @property
     def mean(self):
-        """Weighted mean per key, derived from the accumulated series."""
+        """Weighted mean per key, derived from the accumulated series
+        (local pending contributions plus any rank-merged ones)."""
         view = _MeanView()
-        for k, s in self._wsum.items():
-            w = self._wtot.get(k, 0)
+        for k in set(self._wsum) | set(self._ssum):
+            s = self._wsum.get(k, 0.0) + self._ssum.get(k, 0.0)
+            w = self._wtot.get(k, 0) + self._stot.get(k, 0)
             if w:
                 view[k] = s / w
         return view
+
+    def sync(self):
+        """Merge this logger's pending series across ranks (C3 metric
+        reduction — the reference accumulates metrics in one process,
+        src/logger.py:35-55; sharded evaluation needs the shards' weighted
+        sums summed so every rank sees identical means).  Collective: all
+        ranks must call it at the same point.  No-op when not distributed."""
+        import torch.distributed as dist
+        if not dist.is_initialized() or dist.get_world_size() <= 1:
+            return
+        world = dist.get_world_size()
+        gathered = [None] * world
+        dist.all_gather_object(gathered, (self._wsum, self._wtot))
+        for wsum, wtot in gathered:
+            for k, s in wsum.items():
+                self._ssum[k] = self._ssum.get(k, 0.0) + s
+                self._stot[k] = self._stot.get(k, 0) + wtot[k]
+        self._wsum = {}
+        self._wtot = {}
 
     # -------------------------------------------------------------- rounds
     def safe(self, write):
         """Bracket a round: True opens the event writer; False closes it and
         snapshots every key's current mean into `history`."""
         if write:
-            if _HAS_TB:
-                self.writer = SummaryWriter(self.log_path)
-            else:
-                self.writer = _Jsonl(self.log_path)
+            if self.log_path is not None:
+                if _HAS_TB:
+                    self.writer = SummaryWriter(self.log_path)
+                else:
+                    self.writer = _Jsonl(self.log_path)
             return
         if self.writer is not None:
             self.writer.close()
@@ -75,6 +101,8 @@ class Logger:
     def reset(self):
         self._wsum = {}
         self._wtot = {}
+        self._ssum = {}
+        self._stot = {}
         self.tracker = {}
 
     # -------------------------------------------------------------- logging
@@ -126,6 +154,8 @@ class Logger:
     def __setstate__(self, state):
         # migrate checkpoints pickled by the earlier running-mean layout
         # (mean/counter/iterator attributes) into the series form
+        state.setdefault('_ssum', {})
+        state.setdefault('_stot', {})
         if '_wsum' not in state:
             counter = state.pop('counter', {}) or {}
             old_mean = state.pop('mean', {}) or {}
@@ -133,6 +163,8 @@ class Logger:
                               for k, w in counter.items()}
             state['_wtot'] = dict(counter)
             state['_step'] = dict(state.pop('iterator', {}) or {})
+            state['_ssum'] = {}
+            state['_stot'] = {}
             state.setdefault('tracker', {})
             state['history'] = dict(state.get('history', {}) or {})
         self.__dict__.update(state)

@@ -70,15 +70,36 @@ def _flat_allreduce(tensors, ctx):
 def distributed_combine(federation, trained, param_idx, user_idx, ctx):
     """Padded-reduce combine: exact equivalent of the sequential combine on
     the union of all ranks' clients (tested by construction: accumulators are
-    linear in clients, so SUM over ranks == sequential accumulate)."""
+    linear in clients, so SUM over ranks == sequential accumulate).
+
+    Counts are derived locally from the index maps (Federation.count_map):
+    they depend only on (param_idx, label_split), which every rank holds for
+    ALL clients, so only the value accumulators ride the xGMI links — half
+    the payload, bit-identical.  A one-scalar all-reduce first verifies that
+    every assigned client actually reported; if any went missing (failure
+    tolerance, reference src/fed.py:180-298 semantics) the combine falls
+    back to reducing counts too.  HETEROFL_COMBINE_COUNTS=reduce forces the
+    fallback."""
     slots = sorted(trained.keys())
-    tmp_d, cnt_d = federation.accumulate(trained, param_idx, user_idx, slots=slots)
+    tmp_d, cnt_local = federation.accumulate(trained, param_idx, user_idx,
+                                             slots=slots)
     keys = list(tmp_d.keys())
-    buffers = [tmp_d[k] for k in keys] + [cnt_d[k] for k in keys]
-    reduced = _flat_allreduce(buffers, ctx)
-    n = len(keys)
-    tmp_d = {k: reduced[i] for i, k in enumerate(keys)}
-    cnt_d = {k: reduced[n + i] for i, k in enumerate(keys)}
+    dev = tmp_d[keys[0]].device
+    local_counts = os.environ.get('HETEROFL_COMBINE_COUNTS', 'local') == 'local'
+    if local_counts:
+        n_trained = torch.tensor([len(slots)], dtype=torch.float32, device=dev)
+        dist.all_reduce(n_trained, op=dist.ReduceOp.SUM)
+        local_counts = int(n_trained.item()) == len(user_idx)
+    if local_counts:
+        reduced = _flat_allreduce([tmp_d[k] for k in keys], ctx)
+        tmp_d = {k: reduced[i] for i, k in enumerate(keys)}
+        cnt_d = federation.count_map(param_idx, user_idx)
+    else:
+        buffers = [tmp_d[k] for k in keys] + [cnt_local[k] for k in keys]
+        reduced = _flat_allreduce(buffers, ctx)
+        n = len(keys)
+        tmp_d = {k: reduced[i] for i, k in enumerate(keys)}
+        cnt_d = {k: reduced[n + i] for i, k in enumerate(keys)}
     federation.finalize(tmp_d, cnt_d)
 
 

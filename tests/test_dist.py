@@ -99,7 +99,9 @@ def _stats_worker(rank, world, port, tmpdir):
     cfg['engine'] = 'sequential'
     cfg = make_cfg_local(cfg)
     torch.manual_seed(0)
-    ds = fetch_dataset('MNIST', synthetic=True, synthetic_size=80)
+    # 85 samples with batch 10 -> a partial tail batch, exercising the
+    # whole-batch sharding that hands the tail to exactly one rank
+    ds = fetch_dataset('MNIST', synthetic=True, synthetic_size=85)
     process_dataset(ds, cfg)
     torch.manual_seed(7)
     data_split, label_split = split_dataset(ds, 4, 'iid', cfg['classes_size'])
@@ -132,7 +134,6 @@ def make_cfg_local(cfg):
                  data_name='MNIST', model_name='conv')
     c['num_epochs'] = {'global': 1, 'local': 1}
     c['world_size'] = 2
-    # batch 10 divides the 80-sample set -> equal-size batches
     c['batch_size'] = {'train': 10, 'test': 50}
     return c
 
@@ -268,3 +269,166 @@ def test_collective_helpers():
     the round engine itself needs no W_g broadcast — every rank finalizes
     identical global params from the deterministic padded combine)."""
     mp.spawn(_collective_worker, args=(2, 29561, ''), nprocs=2, join=True)
+
+
+def test_fed_cli_torchrun_ws2(tmp_path):
+    """The REAL fed CLI entry (train_classifier_fed.py) under torchrun
+    world_size=2 on gloo: clients shard across ranks, evaluation is sharded
+    + merged, and only rank 0 writes the checkpoint."""
+    import subprocess
+    import sys
+    root = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    env = dict(os.environ, HETEROFL_MAX_ROUNDS='2',
+               HETEROFL_SYNTHETIC_SIZE='40', OMP_NUM_THREADS='2')
+    out = subprocess.run(
+        [sys.executable, '-m', 'torch.distributed.run', '--nnodes=1',
+         '--nproc-per-node', '2', '--master-addr', '127.0.0.1',
+         '--master-port', '29571',
+         os.path.join(root, 'train_classifier_fed.py'),
+         '--data_name', 'MNIST', '--model_name', 'conv',
+         '--device', 'cpu', '--synthetic', '1', '--num_experiments', '1',
+         '--control_name', '1_4_0.5_iid_fix_a1-e1_bn_1_1'],
+        cwd=str(tmp_path), capture_output=True, text=True, timeout=600,
+        env=env)
+    assert out.returncode == 0, out.stderr[-3000:]
+    tag = '0_MNIST_label_conv_1_4_0.5_iid_fix_a1-e1_bn_1_1'
+    ck = tmp_path / 'output' / 'model' / f'{tag}_checkpoint.pt'
+    assert ck.exists(), out.stdout[-2000:]
+    # exactly one rank printed the per-round console lines
+    assert out.stdout.count('Experiment: ') == 1, out.stdout[-2000:]
+
+
+def _missing_client_worker(rank, world, port, tmpdir):
+    """One rank's client never reports: the scalar completeness check must
+    route the combine onto the count-reducing fallback and still match the
+    sequential combine over the REPORTING clients."""
+    os.environ['MASTER_ADDR'] = '127.0.0.1'
+    os.environ['MASTER_PORT'] = str(port)
+    torch.distributed.init_process_group('gloo', rank=rank, world_size=world)
+    from heterofl_amd.config import default_config
+    from heterofl_amd.data import fetch_dataset, split_dataset
+    from heterofl_amd.fed import FedRunner
+    from heterofl_amd.models import make_model
+    from heterofl_amd.parallel import init_distributed
+    from heterofl_amd.parallel.dist import distributed_combine
+    from heterofl_amd.utils import process_dataset, make_optimizer
+    from heterofl_amd.fed.runner import sample_active_users
+
+    cfg = default_config()
+    cfg['device'] = 'cpu'
+    cfg['engine'] = 'sequential'
+    cfg['metric_name'] = {'train': {'Local': ['Local-Loss']},
+                          'test': {'Global': ['Global-Loss']}}
+    cfg = make_cfg(cfg, '1_6_1_iid_fix_a1-e1_bn_1_1',
+                   data_name='MNIST', model_name='conv')
+    cfg['num_epochs'] = {'global': 1, 'local': 1}
+    cfg['world_size'] = world
+    torch.manual_seed(0)
+    ds = fetch_dataset('MNIST', synthetic=True, synthetic_size=60)
+    process_dataset(ds, cfg)
+    torch.manual_seed(7)
+    data_split, label_split = split_dataset(ds, 6, 'iid', cfg['classes_size'])
+    torch.manual_seed(1)
+    model = make_model(cfg)
+    ctx = init_distributed(backend='gloo')
+    runner = FedRunner(cfg, ds, data_split, label_split, model,
+                       make_optimizer(model, cfg['lr'], cfg), dist_ctx=ctx)
+    g = runner._round_generator(1)
+    user_idx = sample_active_users(cfg, 1, generator=g)
+    runner.federation.make_model_rate(generator=g)
+    my_clients = list(range(rank, len(user_idx), world))
+    local_parameters, param_idx = runner.federation.distribute(
+        user_idx, resample=False, slots=my_clients)
+    trained = dict(runner.trainer.train_clients(
+        my_clients, user_idx, local_parameters,
+        runner.federation.model_rate, runner._make_loader,
+        label_split, cfg['lr']))
+    # drop slot 3 (it lives on rank 3 % world)
+    trained.pop(3, None)
+    distributed_combine(runner.federation, trained, param_idx, user_idx, ctx)
+    gp = {k: v.clone() for k, v in runner.federation.global_parameters.items()}
+
+    # sequential oracle over the reporting clients only
+    torch.manual_seed(1)
+    model2 = make_model(dict(cfg, world_size=1))
+    runner2 = FedRunner(dict(cfg, world_size=1), ds, data_split, label_split,
+                        model2, make_optimizer(model2, cfg['lr'], cfg))
+    local2, pidx2 = runner2.federation.distribute(user_idx, resample=False)
+    runner2.federation.model_rate = list(runner.federation.model_rate)
+    local2, pidx2 = runner2.federation.distribute(user_idx, resample=False)
+    trained2 = dict(runner2.trainer.train_clients(
+        list(range(len(user_idx))), user_idx, local2,
+        runner2.federation.model_rate, runner2._make_loader,
+        label_split, cfg['lr']))
+    trained2.pop(3, None)
+    slots2 = sorted(trained2.keys())
+    tmp_d, cnt_d = runner2.federation.accumulate(trained2, pidx2, user_idx,
+                                                 slots=slots2)
+    runner2.federation.finalize(tmp_d, cnt_d)
+    for k, v in runner2.federation.global_parameters.items():
+        if v.is_floating_point():
+            diff = (gp[k] - v).abs().max().item()
+            assert diff < 1e-5, (rank, k, diff)
+    torch.distributed.destroy_process_group()
+
+
+def test_distributed_combine_missing_client_fallback():
+    mp.spawn(_missing_client_worker, args=(2, 29581, ''), nprocs=2, join=True)
+
+
+def _eval_sync_worker(rank, world, port, tmpdir):
+    """Sharded evaluation (C3): per-user Local loop and Global set sharded
+    across ranks; after logger.sync() every rank's means equal the
+    single-rank oracle's."""
+    os.environ['MASTER_ADDR'] = '127.0.0.1'
+    os.environ['MASTER_PORT'] = str(port)
+    torch.distributed.init_process_group('gloo', rank=rank, world_size=world)
+    from heterofl_amd.config import default_config
+    from heterofl_amd.data import fetch_dataset, split_dataset
+    from heterofl_amd.fed import FedRunner
+    from heterofl_amd.logger import Logger
+    from heterofl_amd.models import make_model
+    from heterofl_amd.parallel import init_distributed
+    from heterofl_amd.utils import process_dataset, make_optimizer
+
+    cfg = default_config()
+    cfg['device'] = 'cpu'
+    cfg['engine'] = 'sequential'
+    cfg['metric_name'] = {'train': {'Local': ['Local-Loss']},
+                          'test': {'Local': ['Local-Loss', 'Local-Accuracy'],
+                                   'Global': ['Global-Loss',
+                                              'Global-Accuracy']}}
+    cfg = make_cfg(cfg, '1_5_0.4_iid_fix_a1_bn_1_1',
+                   data_name='MNIST', model_name='conv')
+    cfg['num_epochs'] = {'global': 1, 'local': 1}
+    cfg['world_size'] = world
+    torch.manual_seed(0)
+    ds = fetch_dataset('MNIST', synthetic=True, synthetic_size=60)
+    process_dataset(ds, cfg)
+    torch.manual_seed(7)
+    data_split, label_split = split_dataset(ds, 5, 'iid', cfg['classes_size'])
+    torch.manual_seed(1)
+    model = make_model(cfg)
+    ctx = init_distributed(backend='gloo')
+    logger = Logger(None)
+    runner = FedRunner(cfg, ds, data_split, label_split, model,
+                       make_optimizer(model, cfg['lr'], cfg),
+                       logger=logger, dist_ctx=ctx)
+    tm = runner.stats()
+    runner.test(tm, 1)
+    means = dict(logger.mean)
+
+    # single-rank oracle with the same test model
+    logger1 = Logger(None)
+    runner1 = FedRunner(dict(cfg, world_size=1), ds, data_split, label_split,
+                        make_model(cfg),
+                        make_optimizer(model, cfg['lr'], cfg), logger=logger1)
+    runner1.test(tm, 1)
+    for k, v in logger1.mean.items():
+        assert k in means, (rank, k)
+        assert abs(means[k] - v) < 1e-6, (rank, k, means[k], v)
+    torch.distributed.destroy_process_group()
+
+
+def test_sharded_eval_sync_matches_sequential():
+    mp.spawn(_eval_sync_worker, args=(2, 29591, ''), nprocs=2, join=True)

@@ -163,15 +163,19 @@ def test_checkpoint_resume_roundtrip(base_cfg, tmp_path, monkeypatch):
         saved['data_split']['train'].keys()
 
 
-def test_resnet50_bottleneck_falls_back_sequential(base_cfg):
-    """Bottleneck resnets run on the sequential engine (batched engine
-    covers conv/resnet18/34/transformer)."""
+def test_resnet50_bottleneck_engines(base_cfg, monkeypatch):
+    """Bottleneck resnets default to the batched BBottleneck engine;
+    HETEROFL_BATCHED_BOTTLENECK=0 forces the sequential oracle."""
+    from heterofl_amd.fed.batched import BatchedClientTrainer
+    from heterofl_amd.fed.sequential import SequentialClientTrainer
     cfg = make_cfg(base_cfg, '1_2_1_iid_fix_a1_bn_1_1',
                    data_name='CIFAR10', model_name='resnet50')
     cfg['engine'] = 'batched'
     cfg['num_epochs'] = {'global': 1, 'local': 1}
     runner = _run(cfg, rounds=1, n_data=20)
-    from heterofl_amd.fed.sequential import SequentialClientTrainer
+    assert isinstance(runner.trainer, BatchedClientTrainer)
+    monkeypatch.setenv('HETEROFL_BATCHED_BOTTLENECK', '0')
+    runner = _run(cfg, rounds=1, n_data=20)
     assert isinstance(runner.trainer, SequentialClientTrainer)
 
 
