@@ -124,6 +124,10 @@ class BNormReLU(nn.Module):
         super().__init__()
         self.norm = norm
         self.rate = rate if scale else 1.0
+        # set by the native sBN statistics pass (runner.stats): a callable
+        # sink(module, batch_mean, batch_invstd, x) harvesting per-batch BN
+        # stats from the fused kernel's own reduction
+        self.stats_sink = None
         if norm == 'in':
             self.groups = R * ch
         elif norm == 'ln':
@@ -141,13 +145,22 @@ class BNormReLU(nn.Module):
             if self.training and self.rate != 1.0:
                 x = x / self.rate
             return F.relu(x)
+        kind = 'bn' if self.norm == 'bn' else 'gn'
         if native_ops.use_native(x):
+            if self.stats_sink is not None and kind == 'bn':
+                ext = native_ops.require_native()
+                y, mean, invstd = ext.bn_relu_fwd(x.contiguous(), self.weight,
+                                                  self.bias, 1e-5)
+                self.stats_sink(self, mean, invstd, x)
+                return y
             from ..ops.fused import fused_norm_relu
-            kind = 'bn' if self.norm == 'bn' else 'gn'
             return fused_norm_relu(x, self.weight, self.bias, kind,
                                    self.groups)
+        if self.stats_sink is not None and kind == 'bn':
+            mean = x.mean(dim=(0, 2, 3))
+            var = x.var(dim=(0, 2, 3), unbiased=False)
+            self.stats_sink(self, mean, (var + 1e-5).rsqrt(), x)
         from ..ops.fused import eager_scaler_norm_relu
-        kind = 'bn' if self.norm == 'bn' else 'gn'
         r = self.rate if self.training else 1.0
         return eager_scaler_norm_relu(x, self.weight, self.bias, kind,
                                       self.groups, r)

@@ -200,6 +200,9 @@ class FedRunner:
             test_model = make_model(cfg, model_rate=cfg['global_model_rate'],
                                     track=True).to(cfg['device'])
             test_model.load_state_dict(self.global_model.state_dict(), strict=False)
+            if self._native_stats_ok():
+                self._native_stats(test_model)
+                return test_model
             test_model.train(True)
             # BN momentum=None is a cumulative average over batches, which is
             # batch-size invariant for equal-size batches — so the stats pass
@@ -255,6 +258,111 @@ class FedRunner:
                         b.copy_(flat[off:off + n].view_as(b))
                         off += n
         return test_model
+
+    # ------------------------------------------------------ native sBN pass
+    def _native_stats_ok(self):
+        """Route the sBN statistics pass through the batched native engine
+        (MFMA convs + fused sBN kernels) instead of the eager per-client
+        model.  bn only (other norms track no running stats) and vision only;
+        HETEROFL_NATIVE_STATS=0 forces the eager pass."""
+        from .batched import BatchedClientTrainer
+        return (self.cfg['norm'] == 'bn' and not self.is_lm
+                and isinstance(self.trainer, BatchedClientTrainer)
+                and os.environ.get('HETEROFL_NATIVE_STATS', '1') == '1')
+
+    def _native_stats(self, test_model):
+        """Run the full train set (normalized, augmented, unshuffled,
+        batch 500 on GPU) through a 1-group batched model at the global rate
+        and harvest each fused BN kernel's per-batch mean/var into the
+        cumulative (momentum=None) running stats of test_model
+        (reference semantics: src/train_classifier_fed.py:127-138 with
+        BatchNorm2d(momentum=None), src/models/resnet.py:16-17).
+
+        Multi-rank: whole batches (incl. the partial tail) shard across
+        ranks; per-module stats merge with a batch-count-weighted
+        all-reduce (C2)."""
+        from .batched import pack_states, BNormReLU
+        cfg = self.cfg
+        device = torch.device(cfg['device'])
+        trainer = self.trainer
+        bmodel = trainer._batched_model(cfg['global_model_rate'], 1)
+        pack_states(bmodel, [dict(self.federation.global_parameters)])
+        bmodel.train(True)
+
+        state = {}  # module -> [mean, unbiased_var, n_batches]
+
+        def sink(mod, mean, invstd, x):
+            n, _, h, w = x.shape
+            m = n * h * w
+            mu = mean.float()
+            biased = invstd.float().pow(-2) - 1e-5
+            unb = biased * (m / max(m - 1, 1))
+            s = state.get(mod)
+            if s is None:
+                state[mod] = [mu.clone(), unb.clone(), 1]
+            else:
+                s[2] += 1
+                k = s[2]
+                s[0] += (mu - s[0]) / k
+                s[1] += (unb - s[1]) / k
+
+        bn_mods = [(name, m) for name, m in bmodel.named_modules()
+                   if isinstance(m, BNormReLU) and m.norm == 'bn']
+        for _, m in bn_mods:
+            m.stats_sink = sink
+        try:
+            train = self.dataset['train']
+            img = getattr(self, '_stats_img', None)
+            if img is None:
+                img = train.img
+                if img.dim() == 3:
+                    img = img.unsqueeze(-1)
+                img = img.to(device)
+                self._stats_img = img
+            n_total = img.size(0)
+            bs = 500 if device.type == 'cuda' else cfg['batch_size']['train']
+            total_b = (n_total + bs - 1) // bs
+            rank, world = (0, 1) if self.dist_ctx is None else \
+                (self.dist_ctx.rank, self.dist_ctx.world_size)
+            # the reference's stats pass iterates the TRAIN loader, i.e.
+            # with train-time augmentation active (train_classifier_fed.py:
+            # 127-138 over the train dataset's transforms)
+            for b in range(rank, total_b, world):
+                batch = img[b * bs:min((b + 1) * bs, n_total)]
+                x = trainer.augment(batch, train=True)
+                bmodel(x)
+        finally:
+            for _, m in bn_mods:
+                m.stats_sink = None
+        if self.dist_ctx is not None and self.dist_ctx.world_size > 1:
+            import torch.distributed as dist
+            parts, weights = [], []
+            for _, m in bn_mods:
+                mu, var, k = state[m]
+                parts.append(mu * k)
+                parts.append(var * k)
+                weights.append(float(k))
+            flat = torch.cat([p.reshape(-1) for p in parts])
+            wt = torch.tensor(weights, dtype=torch.float32, device=flat.device)
+            dist.all_reduce(flat)
+            dist.all_reduce(wt)
+            off = 0
+            for i, (_, m) in enumerate(bn_mods):
+                mu, var, k = state[m]
+                n = mu.numel()
+                state[m] = [flat[off:off + n] / wt[i].clamp(min=1),
+                            flat[off + n:off + 2 * n] / wt[i].clamp(min=1),
+                            int(wt[i].item())]
+                off += 2 * n
+        named = dict(test_model.named_modules())
+        for name, m in bn_mods:
+            mu, var, k = state[m]
+            tgt = named[name]
+            tgt.running_mean.copy_(mu.to(tgt.running_mean.dtype))
+            tgt.running_var.copy_(var.to(tgt.running_var.dtype))
+            if tgt.num_batches_tracked is not None:
+                tgt.num_batches_tracked.fill_(k)
+        test_model.train(False)
 
     # ------------------------------------------------------------------- test
     def test(self, test_model, epoch):

@@ -387,3 +387,32 @@ def test_run_to_run_reproducibility(base_cfg):
     a, b = run_once(), run_once()
     for k in a:
         assert torch.equal(a[k], b[k]), k
+
+
+def test_native_stats_pass_matches_eager(base_cfg, monkeypatch):
+    """runner.stats() routed through the batched engine's stats sink
+    (fused-BN mean/var harvest) reproduces the eager per-batch cumulative-BN
+    pass exactly (MNIST: no train-time augmentation, so both passes are
+    deterministic)."""
+    cfg = make_cfg(base_cfg, '1_4_0.5_iid_fix_a1_bn_1_1',
+                   data_name='MNIST', model_name='conv')
+    cfg['engine'] = 'batched'
+    cfg['num_epochs'] = {'global': 1, 'local': 1}
+    torch.manual_seed(0)
+    ds = fetch_dataset('MNIST', synthetic=True, synthetic_size=45)
+    process_dataset(ds, cfg)
+    data_split, label_split = split_dataset(ds, 4, 'iid', cfg['classes_size'])
+    model = make_model(cfg)
+    opt = make_optimizer(model, cfg['lr'], cfg)
+    runner = FedRunner(cfg, ds, data_split, label_split, model, opt)
+    runner.train_round(1)
+    assert runner._native_stats_ok()
+    tm_native = runner.stats()
+    monkeypatch.setenv('HETEROFL_NATIVE_STATS', '0')
+    assert not runner._native_stats_ok()
+    tm_eager = runner.stats()
+    sd_n, sd_e = tm_native.state_dict(), tm_eager.state_dict()
+    for k in sd_e:
+        if 'running' in k or 'num_batches' in k:
+            diff = (sd_n[k].float() - sd_e[k].float()).abs().max().item()
+            assert diff < 1e-4, (k, diff)
