@@ -236,6 +236,8 @@ def _load_wikitext_raw(root, split):
 
 # --------------------------------------------------------------- synthetic
 def _synthetic_vision(data_name, n, classes_size, seed):
+    if os.environ.get('HETEROFL_SYNTHETIC_MODE') == 'learnable':
+        return _learnable_vision(data_name, n, classes_size, seed)
     g = torch.Generator().manual_seed(seed)
     if data_name in ('MNIST', 'FashionMNIST', 'EMNIST'):
         img = torch.randint(0, 256, (n, 28, 28), dtype=torch.uint8, generator=g)
@@ -243,6 +245,58 @@ def _synthetic_vision(data_name, n, classes_size, seed):
         img = torch.randint(0, 256, (n, 32, 32, 3), dtype=torch.uint8, generator=g)
     target = torch.randint(0, classes_size, (n,), generator=g).tolist()
     return img, target
+
+
+def _learnable_vision(data_name, n, classes_size, seed):
+    """Learnable class-conditional synthetic images (the no-network proxy
+    for the BASELINE accuracy study — there is no way to fetch real CIFAR
+    raw files in this environment).
+
+    Each class owns a bank of 64 smooth random templates (low-pass noise
+    upsampled to full resolution, shared between train/test); a sample is a
+    random class template with amplitude jitter, a random circular shift and
+    dense pixel noise.  A CNN must learn the 640 template shapes under
+    noise/shift/crop — pixel means alone do not separate classes — so the
+    federated training dynamics (sBN, lr schedule, heterogeneous widths,
+    masked CE, augmentation) all matter, while the problem stays solvable to
+    high accuracy when training is healthy."""
+    import torch.nn.functional as F
+    if data_name in ('MNIST', 'FashionMNIST', 'EMNIST'):
+        C, H, W = 1, 28, 28
+    else:
+        C, H, W = 3, 32, 32
+    T = 64
+    gb = torch.Generator().manual_seed(777)   # class bank: split-invariant
+    bank = torch.randn(classes_size * T, C, 8, 8, generator=gb)
+    bank = F.interpolate(bank, size=(H, W), mode='bilinear',
+                         align_corners=False)
+    bank = (bank - bank.mean(dim=(1, 2, 3), keepdim=True)) \
+        / bank.std(dim=(1, 2, 3), keepdim=True).clamp(min=1e-6)
+    bank = bank.view(classes_size, T, C, H, W)
+    gs = torch.Generator().manual_seed(10_000 + seed)  # draws: per split
+    y = torch.randint(0, classes_size, (n,), generator=gs)
+    t = torch.randint(0, T, (n,), generator=gs)
+    amp = 0.7 + 0.6 * torch.rand(n, generator=gs)
+    dy = torch.randint(-4, 5, (n,), generator=gs)
+    dx = torch.randint(-4, 5, (n,), generator=gs)
+    out = torch.empty(n, C, H, W, dtype=torch.uint8)
+    chunk = 4096
+    for s in range(0, n, chunk):
+        e = min(s + chunk, n)
+        x = bank[y[s:e], t[s:e]] * amp[s:e].view(-1, 1, 1, 1)
+        x = x + 0.75 * torch.randn(x.shape, generator=gs)
+        # group samples by shift so each group rolls at once
+        key = (dy[s:e] + 4) * 9 + (dx[s:e] + 4)
+        for kv in key.unique().tolist():
+            sel = (key == kv).nonzero(as_tuple=True)[0]
+            sdy, sdx = kv // 9 - 4, kv % 9 - 4
+            if sdy or sdx:
+                x[sel] = torch.roll(x[sel], shifts=(sdy, sdx), dims=(2, 3))
+        out[s:e] = (128 + 48 * x).clamp(0, 255).to(torch.uint8)
+    img = out.permute(0, 2, 3, 1)
+    if C == 1:
+        img = img.squeeze(-1)
+    return img.contiguous(), y.tolist()
 
 
 def fetch_dataset(data_name, subset='label', synthetic=False,

@@ -144,11 +144,18 @@ def _run_one(cfg, seed, pivot_sign):
             classes_size=cfg.get('classes_size'))
     runner = FedRunner(cfg, dataset, data_split, label_split, model, optimizer,
                        logger=logger, dist_ctx=ctx)
-    for epoch in range(last_epoch, cfg['num_epochs']['global'] + 1):
+    # HETEROFL_EVAL_EVERY=k evaluates every k-th round (+ the final round)
+    # instead of the reference's every-round evaluation — a cost knob for
+    # long runs; k=1 (default) is reference behavior
+    eval_every = int(os.environ.get('HETEROFL_EVAL_EVERY', '1'))
+    total_epochs = cfg['num_epochs']['global']
+    for epoch in range(last_epoch, total_epochs + 1):
         logger.safe(True)
         runner.train_round(epoch)
-        test_model = runner.stats()
-        runner.test(test_model, epoch)
+        do_eval = (epoch % eval_every == 0) or epoch == total_epochs
+        if do_eval:
+            test_model = runner.stats()
+            runner.test(test_model, epoch)
         if cfg['scheduler_name'] == 'ReduceLROnPlateau':
             # post-sync means are identical on every rank
             scheduler.step(metrics=logger.mean['train/{}'.format(cfg['pivot_metric'])])
@@ -165,7 +172,8 @@ def _run_one(cfg, seed, pivot_sign):
                 scheduler.step()
         logger.safe(False)
         cur = logger.mean['test/{}'.format(cfg['pivot_metric'])]
-        better = cur > cfg['pivot'] if pivot_sign > 0 else cur < cfg['pivot']
+        better = do_eval and (cur > cfg['pivot'] if pivot_sign > 0
+                              else cur < cfg['pivot'])
         if better:
             cfg['pivot'] = cur
         if is_main:
