@@ -8,18 +8,23 @@
 
 // feat (N, R*C, H, W) -> pooled (N, R, C) [saved for bwd], scores (N, R, J)
 // scores = pooled @ W[r]^T + b[r];  W (R, J, C) fp32, b (R, J) fp32.
+// N is tiled over blockIdx.y (TN rows per block) so the (TN, C) LDS pool
+// slab stays within limits at any batch size — the sBN statistics / eval
+// paths run this head at N=500, not just the train batch.
 template <typename T>
 __global__ void __launch_bounds__(256)
 head_fwd_kernel(const T* __restrict__ feat, const float* __restrict__ w,
                 const float* __restrict__ b, float* __restrict__ pooled,
                 float* __restrict__ scores, int N, int R, int C, int HW,
-                int J) {
+                int J, int TN) {
     const int r = blockIdx.x;
-    extern __shared__ float pool_lds[];  // (N, C)
+    const int n0 = blockIdx.y * TN;
+    const int nt = min(TN, N - n0);
+    extern __shared__ float pool_lds[];  // (TN, C)
     const float inv_hw = 1.f / HW;
-    // pool: each thread owns (n, c) pairs
-    for (int e = threadIdx.x; e < N * C; e += blockDim.x) {
-        const int n = e / C, c = e - n * C;
+    // pool: each thread owns (n, c) pairs of this n-tile
+    for (int e = threadIdx.x; e < nt * C; e += blockDim.x) {
+        const int n = n0 + e / C, c = e - (e / C) * C;
         const T* src = feat + ((long)n * R * C + (long)r * C + c) * HW;
         float s = 0.f;
         for (int i = 0; i < HW; ++i) s += ld_f32(src + i);
@@ -29,10 +34,10 @@ head_fwd_kernel(const T* __restrict__ feat, const float* __restrict__ w,
     }
     __syncthreads();
     // scores: thread owns (n, j)
-    for (int e = threadIdx.x; e < N * J; e += blockDim.x) {
-        const int n = e / J, j = e - n * J;
+    for (int e = threadIdx.x; e < nt * J; e += blockDim.x) {
+        const int n = n0 + e / J, j = e - (e / J) * J;
         const float* wr = w + ((long)r * J + j) * C;
-        const float* pr = pool_lds + n * C;
+        const float* pr = pool_lds + (n - n0) * C;
         float s = b ? b[r * J + j] : 0.f;
         for (int c = 0; c < C; ++c) s += pr[c] * wr[c];
         scores[((long)n * R + r) * J + j] = s;
@@ -99,14 +104,18 @@ std::vector<at::Tensor> head_fwd(at::Tensor feat, at::Tensor w, at::Tensor b,
     auto pooled = at::empty({N, (long)R, C}, opts);
     auto scores = at::empty({N, (long)R, J}, opts);
     auto stream = at::hip::getCurrentHIPStream();
-    const int lds = N * C * sizeof(float);
+    // n-tile size: (TN, C) fp32 slab capped at 32 KB LDS
+    const int TN = std::max(1, std::min(N, 8192 / C));
+    const int lds = TN * C * sizeof(float);
+    const int ny = (N + TN - 1) / TN;
     DISPATCH_HT(feat.scalar_type(), {
-        hipLaunchKernelGGL(head_fwd_kernel<scalar_t>, dim3((int)R), dim3(256),
-                           lds, stream, (const scalar_t*)feat.data_ptr(),
+        hipLaunchKernelGGL(head_fwd_kernel<scalar_t>, dim3((int)R, ny),
+                           dim3(256), lds, stream,
+                           (const scalar_t*)feat.data_ptr(),
                            w.data_ptr<float>(),
                            b.defined() ? b.data_ptr<float>() : nullptr,
                            pooled.data_ptr<float>(), scores.data_ptr<float>(),
-                           N, (int)R, C, HW, J);
+                           N, (int)R, C, HW, J, TN);
     });
     return {scores, pooled};
 }
@@ -127,6 +136,11 @@ std::vector<at::Tensor> head_bwd(at::Tensor dscores, at::Tensor pooled,
                            pooled.options().dtype(feat_dtype));
     auto stream = at::hip::getCurrentHIPStream();
     const int lds = N * J * sizeof(float);
+    // fail loudly instead of poisoning the queue with an over-LDS launch
+    // (the training path keeps N small; fwd tiles N, bwd does not)
+    TORCH_CHECK(lds <= 64 * 1024,
+                "head_bwd: N*J too large for the LDS stage (N=", N,
+                ", J=", J, ")");
     DISPATCH_HT(feat_dtype, {
         hipLaunchKernelGGL(head_bwd_kernel<scalar_t>, dim3((int)R), dim3(256),
                            lds, stream, dsc.data_ptr<float>(),

@@ -770,3 +770,21 @@ def test_resnet50_fed_round_batched_default(base_cfg):
         runner.train_round(ep)
     l1 = global_loss()
     assert l1 < l0, (l0, l1)
+
+
+@needs_gpu
+def test_fused_head_large_batch():
+    """The n-tiled head forward handles eval/stats batch sizes (N=500) that
+    overflowed the old (N, C) LDS slab and poisoned the queue (r2 fix)."""
+    from heterofl_amd.ops.fused import fused_head
+    for C, N in [(512, 500), (64, 500), (512, 10), (512, 37)]:
+        x = torch.randn(N, C, 4, 4, device='cuda:0')
+        w = torch.randn(1, 10, C, device='cuda:0')
+        b = torch.randn(1, 10, device='cuda:0')
+        s = fused_head(x, w, b, 1)
+        pooled = torch.nn.functional.adaptive_avg_pool2d(x, 1).view(N, 1, C)
+        ref = torch.einsum('nrc,rjc->nrj', pooled, w) + b
+        assert (s.float() - ref).abs().max().item() < 1e-2, (C, N)
+        # canary: the queue must still accept launches after the head
+        torch.zeros(8, device='cuda:0').add_(1.0)
+        torch.cuda.synchronize()
