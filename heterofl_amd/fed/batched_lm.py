@@ -19,6 +19,23 @@ import torch.nn.functional as F
 from .. import ops as native_ops
 
 
+class _ShadowParam(torch.autograd.Function):
+    """Route a persistent bf16 shadow of an fp32 master parameter through
+    autograd: forward returns the shadow with NO cast kernel (the fused
+    clip+SGD kernel mirrors every update into it, ops/csrc/clip_sgd.hip),
+    backward upcasts the bf16 grad for the fp32 master.  This removes the
+    ~80 per-step autocast weight-cast launches the round-1 LM profile was
+    dominated by (profiles/kstats_r01_lm.txt)."""
+
+    @staticmethod
+    def forward(ctx, param, shadow):
+        return shadow
+
+    @staticmethod
+    def backward(ctx, g):
+        return g.float(), None
+
+
 class SLinear(nn.Module):
     """R stacked Linears: weight (R, out, in), x (R, B, in) -> (R, B, out)."""
 
@@ -26,14 +43,22 @@ class SLinear(nn.Module):
         super().__init__()
         self.weight = nn.Parameter(torch.empty(R, out_f, in_f).normal_(0, 0.02))
         self.bias = nn.Parameter(torch.zeros(R, out_f))
+        self.register_buffer('w16', None, persistent=False)
+        self.register_buffer('b16', None, persistent=False)
 
     def forward(self, x):
         lead = x.shape[:-1]
         R = self.weight.size(0)
+        if self.w16 is not None and x.is_cuda:
+            w = _ShadowParam.apply(self.weight, self.w16)
+            b = _ShadowParam.apply(self.bias, self.b16)
+            if x.dtype != torch.bfloat16:
+                x = x.to(torch.bfloat16)
+        else:
+            w, b = self.weight, self.bias
         flat = x.reshape(R, -1, x.size(-1))
-        out = torch.baddbmm(self.bias.unsqueeze(1), flat,
-                            self.weight.transpose(1, 2))
-        return out.reshape(*lead, self.weight.size(1))
+        out = torch.baddbmm(b.unsqueeze(1), flat, w.transpose(1, 2))
+        return out.reshape(*lead, w.size(1))
 
 
 class SLayerNorm(nn.Module):
@@ -74,12 +99,54 @@ class SEmbedding(nn.Module):
     def __init__(self, R, V, E):
         super().__init__()
         self.weight = nn.Parameter(torch.empty(R, V, E).normal_(0, 0.02))
+        self.register_buffer('w16', None, persistent=False)
 
     def forward(self, ids):
         R, V, E = self.weight.shape
-        flat = self.weight.reshape(R * V, E)
+        w = self.weight
+        if self.w16 is not None and ids.is_cuda:
+            w = _ShadowParam.apply(self.weight, self.w16)
+        flat = w.reshape(R * V, E)
         off = (torch.arange(R, device=ids.device) * V).view(R, *([1] * (ids.dim() - 1)))
         return F.embedding(ids + off, flat)
+
+
+def enable_bf16_shadows(model):
+    """Create bf16 shadow buffers for every SLinear / SEmbedding master
+    parameter.  Returns {id(param): shadow}; pass the aligned shadow list to
+    FusedClipSGD/GraphClipSGD so the optimizer kernel keeps the shadows
+    current in-graph (no per-step cast launches)."""
+    shadow_map = getattr(model, '_shadow_map', None)
+    if shadow_map is not None:
+        return shadow_map
+    shadow_map = {}
+    for mod in model.modules():
+        if isinstance(mod, SLinear):
+            names = ('weight', 'bias')
+        elif isinstance(mod, SEmbedding):
+            names = ('weight',)
+        else:
+            continue
+        for pname in names:
+            p = getattr(mod, pname)
+            sh = p.detach().to(torch.bfloat16).contiguous()
+            setattr(mod, 'w16' if pname == 'weight' else 'b16', sh)
+            shadow_map[id(p)] = sh
+    model._shadow_map = shadow_map
+    return shadow_map
+
+
+def refresh_shadows(model):
+    """Re-cast every shadow from its fp32 master (after pack_states wrote
+    new round parameters, or after graph-capture state restore)."""
+    shadow_map = getattr(model, '_shadow_map', None)
+    if not shadow_map:
+        return
+    with torch.no_grad():
+        for p in model.parameters():
+            sh = shadow_map.get(id(p))
+            if sh is not None:
+                sh.copy_(p)
 
 
 class STransformerEmbedding(nn.Module):

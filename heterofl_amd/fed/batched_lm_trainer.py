@@ -9,7 +9,8 @@ import torch
 
 from .. import ops as native_ops
 from .batched import pack_states, unpack_states, per_client_clip_
-from .batched_lm import make_batched_transformer, lm_masked_ce
+from .batched_lm import (make_batched_transformer, lm_masked_ce,
+                         enable_bf16_shadows, refresh_shadows)
 
 
 class BatchedLMClientTrainer:
@@ -60,10 +61,15 @@ class BatchedLMClientTrainer:
                                              locals_list, label_split, lr,
                                              logger)
         model = self._model(rate, R)
+        native = native_ops.use_native(device)
+        use_shadows = native and self._amp
+        if use_shadows:
+            smap = enable_bf16_shadows(model)
         pack_states(model, locals_list)
+        if use_shadows:
+            refresh_shadows(model)
         model.train(True)
         params = [p for p in model.parameters() if p.requires_grad]
-        native = native_ops.use_native(device)
         if native:
             from ..ops.fused import FusedClipSGD
             cached = self._opt_cache.get((rate, R))
@@ -71,9 +77,12 @@ class BatchedLMClientTrainer:
                 for p in params:
                     p.grad = torch.zeros_like(p)
                 bufs = [torch.zeros_like(p) for p in params]
+                shadows = [smap.get(id(p)) for p in params] \
+                    if use_shadows else None
                 cached = (model, FusedClipSGD(params,
                                               [p.grad for p in params],
-                                              bufs, R, device))
+                                              bufs, R, device,
+                                              shadows=shadows))
                 self._opt_cache[(rate, R)] = cached
             fopt = cached[1]
             for b in fopt.bufs:
@@ -99,7 +108,7 @@ class BatchedLMClientTrainer:
                 else:
                     opt.zero_grad(set_to_none=True)
                 with torch.autocast('cuda', torch.bfloat16,
-                                    enabled=self._amp):
+                                    enabled=self._amp and not use_shadows):
                     logits = model(tokens)
                 losses = lm_masked_ce(logits, tokens, masks)
                 losses.sum().backward()
@@ -144,11 +153,14 @@ class BatchedLMClientTrainer:
         if key not in self._graph_cache:
             model = make_batched_transformer(cfg, rate, R).to(device)
             model.train(True)
+            if self._amp and native_ops.use_native(device):
+                enable_bf16_shadows(model)
             self._graph_cache[key] = LMGraphedStep(
                 model, R, rows.size(1), bptt, n_full, lr, cfg['momentum'],
                 cfg['weight_decay'], cfg['num_tokens'], self._amp, device)
         gs = self._graph_cache[key]
         pack_states(gs.model, locals_list)
+        refresh_shadows(gs.model)
         masks = None
         if cfg['mask']:
             masks = torch.zeros(R, cfg['num_tokens'], device=device)

@@ -57,6 +57,7 @@ class GraphedGroupStep:
         self._native = (device.type == 'cuda' and native_ops.use_native(device))
         self._capturing = False
         self._captured_grads = None
+        self._shadows = False  # bf16 shadow weights are an LM-path feature
         if self._native:
             native_ops.require_native()
             from ..ops.fused import GraphClipSGD
@@ -146,6 +147,9 @@ class GraphedGroupStep:
                 b.copy_(sb)
             self.metrics.copy_(saved_m)
             self.counter.zero_()
+        if self._shadows:
+            from .batched_lm import refresh_shadows
+            refresh_shadows(self.model)
         torch.cuda.synchronize()
 
     # ------------------------------------------------------------------ API
@@ -212,9 +216,14 @@ class LMGraphedStep:
         self._capturing = False
         self._captured_grads = None
         self._native = (device.type == 'cuda' and native_ops.use_native(device))
+        smap = getattr(model, '_shadow_map', None) or {}
+        self._shadows = bool(smap)
         if self._native:
             from ..ops.fused import GraphClipSGD
-            self.opt = GraphClipSGD(self.params, self.bufs, R, device)
+            self.opt = GraphClipSGD(self.params, self.bufs, R, device,
+                                    shadows=[smap.get(id(p))
+                                             for p in self.params]
+                                    if smap else None)
 
     def _one_step(self):
         w = self.counter - (self.counter // self.n_win) * self.n_win
@@ -234,7 +243,10 @@ class LMGraphedStep:
 
     def _step_on(self, tokens):
         from .batched_lm import lm_masked_ce
-        with torch.autocast('cuda', torch.bfloat16, enabled=self.amp):
+        # with bf16 shadow weights the model computes in bf16 natively and
+        # autocast (with its per-use weight casts) is unnecessary
+        with torch.autocast('cuda', torch.bfloat16,
+                            enabled=self.amp and not self._shadows):
             logits = self.model(tokens)
         losses = lm_masked_ce(logits, tokens, self.masks)
         with torch.no_grad():
@@ -301,6 +313,9 @@ class LMGraphedStep:
                 b.copy_(sb)
             self.metrics.copy_(saved_m)
             self.counter.zero_()
+        if self._shadows:
+            from .batched_lm import refresh_shadows
+            refresh_shadows(self.model)
         torch.cuda.synchronize()
 
     def begin_round(self, rows, masks):

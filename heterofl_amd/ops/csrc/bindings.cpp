@@ -19,11 +19,12 @@ at::Tensor masked_ce_bwd(at::Tensor scores, at::Tensor labels, at::Tensor mask,
 std::vector<at::Tensor> build_chunk_table(std::vector<at::Tensor> grads,
                                           std::vector<at::Tensor> params,
                                           std::vector<at::Tensor> bufs,
-                                          int64_t R, int64_t chunk_elems);
+                                          int64_t R, int64_t chunk_elems,
+                                          std::vector<at::Tensor> shadows);
 void fill_chunk_table(at::Tensor blob, std::vector<at::Tensor> grads,
                       std::vector<at::Tensor> params,
                       std::vector<at::Tensor> bufs, int64_t R,
-                      int64_t chunk_elems);
+                      int64_t chunk_elems, std::vector<at::Tensor> shadows);
 void clip_sgd_step(at::Tensor table_blob, int64_t n_chunks,
                    at::Tensor chunk_client, at::Tensor partials,
                    at::Tensor normsq, double max_norm, double lr,

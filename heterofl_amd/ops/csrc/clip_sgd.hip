@@ -17,6 +17,9 @@ struct Chunk {             // one contiguous piece of one client's slice
     const float* grad;
     float* param;
     float* buf;
+    __hip_bfloat16* shadow;  // optional bf16 mirror of param (LM forward
+                             // reads it; updating it here removes every
+                             // per-step weight-cast launch from the path)
     int len;
     int client;
 };
@@ -70,7 +73,9 @@ clip_sgd_step_kernel(const Chunk* __restrict__ chunks,
         float g = ck.grad[i] * scale + weight_decay * p;
         const float b = momentum * ck.buf[i] + g;
         ck.buf[i] = b;
-        ck.param[i] = p - lr * b;
+        const float pn = p - lr * b;
+        ck.param[i] = pn;
+        if (ck.shadow) ck.shadow[i] = __float2bfloat16(pn);
     }
 }
 
@@ -83,7 +88,8 @@ clip_sgd_step_kernel(const Chunk* __restrict__ chunks,
 std::vector<at::Tensor> build_chunk_table(std::vector<at::Tensor> grads,
                                           std::vector<at::Tensor> params,
                                           std::vector<at::Tensor> bufs,
-                                          int64_t R, int64_t chunk_elems) {
+                                          int64_t R, int64_t chunk_elems,
+                                          std::vector<at::Tensor> shadows) {
     std::vector<Chunk> host;
     for (size_t t = 0; t < grads.size(); ++t) {
         TORCH_CHECK(grads[t].is_contiguous() && params[t].is_contiguous() &&
@@ -96,12 +102,22 @@ std::vector<at::Tensor> build_chunk_table(std::vector<at::Tensor> grads,
         const float* g = grads[t].data_ptr<float>();
         float* p = params[t].data_ptr<float>();
         float* b = bufs[t].data_ptr<float>();
+        __hip_bfloat16* sh = nullptr;
+        if (t < shadows.size() && shadows[t].defined() &&
+            shadows[t].numel() > 0) {
+            TORCH_CHECK(shadows[t].is_contiguous() &&
+                        shadows[t].scalar_type() == at::kBFloat16 &&
+                        shadows[t].numel() == numel,
+                        "shadow must be a contiguous bf16 mirror");
+            sh = (__hip_bfloat16*)shadows[t].data_ptr();
+        }
         for (long r = 0; r < R; ++r) {
             for (long off = 0; off < per; off += chunk_elems) {
                 Chunk ck;
                 ck.grad = g + r * per + off;
                 ck.param = p + r * per + off;
                 ck.buf = b + r * per + off;
+                ck.shadow = sh ? sh + r * per + off : nullptr;
                 ck.len = (int)std::min((long)chunk_elems, per - off);
                 ck.client = (int)r;
                 host.push_back(ck);
@@ -130,7 +146,7 @@ std::vector<at::Tensor> build_chunk_table(std::vector<at::Tensor> grads,
 void fill_chunk_table(at::Tensor blob, std::vector<at::Tensor> grads,
                       std::vector<at::Tensor> params,
                       std::vector<at::Tensor> bufs, int64_t R,
-                      int64_t chunk_elems) {
+                      int64_t chunk_elems, std::vector<at::Tensor> shadows) {
     std::vector<Chunk> host;
     for (size_t t = 0; t < grads.size(); ++t) {
         const long numel = grads[t].numel();
@@ -138,12 +154,17 @@ void fill_chunk_table(at::Tensor blob, std::vector<at::Tensor> grads,
         const float* g = grads[t].data_ptr<float>();
         float* p = params[t].data_ptr<float>();
         float* b = bufs[t].data_ptr<float>();
+        __hip_bfloat16* sh = nullptr;
+        if (t < shadows.size() && shadows[t].defined() &&
+            shadows[t].numel() > 0)
+            sh = (__hip_bfloat16*)shadows[t].data_ptr();
         for (long r = 0; r < R; ++r)
             for (long off = 0; off < per; off += chunk_elems) {
                 Chunk ck;
                 ck.grad = g + r * per + off;
                 ck.param = p + r * per + off;
                 ck.buf = b + r * per + off;
+                ck.shadow = sh ? sh + r * per + off : nullptr;
                 ck.len = (int)std::min((long)chunk_elems, per - off);
                 ck.client = (int)r;
                 host.push_back(ck);

@@ -97,13 +97,15 @@ class FusedClipSGD:
 
     CHUNK = 65536
 
-    def __init__(self, params, grads, bufs, R, device):
+    def __init__(self, params, grads, bufs, R, device, shadows=None):
         ext = require_native()
         self.params = list(params)
         self.grads = list(grads)
         self.bufs = list(bufs)
+        self.shadows = [s if s is not None else torch.Tensor()
+                        for s in (shadows or [])]
         blob, n, clients = ext.build_chunk_table(
-            self.grads, self.params, self.bufs, R, self.CHUNK)
+            self.grads, self.params, self.bufs, R, self.CHUNK, self.shadows)
         self.table = blob
         self.n_chunks = int(n.item())
         self.chunk_client = clients
@@ -198,12 +200,14 @@ class GraphClipSGD:
     autograd.grad) are filled in afterwards with fill_chunk_table."""
 
     CHUNK = 65536
-    CHUNK_BYTES = 32  # sizeof(Chunk): 3 pointers + 2 ints
+    CHUNK_BYTES = 40  # sizeof(Chunk): 4 pointers + 2 ints
 
-    def __init__(self, params, bufs, R, device):
+    def __init__(self, params, bufs, R, device, shadows=None):
         require_native()
         self.params = list(params)
         self.bufs = list(bufs)
+        self.shadows = [s if s is not None else torch.Tensor()
+                        for s in (shadows or [])]
         self.R = R
         clients = []
         for p in self.params:
@@ -220,10 +224,10 @@ class GraphClipSGD:
         self.normsq = torch.zeros(R, dtype=torch.float32, device=device)
 
     def bind(self, grads):
-        """Write (grad, param, buf) pointers into the device table."""
+        """Write (grad, param, buf, shadow) pointers into the device table."""
         ext = require_native()
         ext.fill_chunk_table(self.table, list(grads), self.params, self.bufs,
-                             self.R, self.CHUNK)
+                             self.R, self.CHUNK, self.shadows)
 
     def launch(self, max_norm, lr, momentum, weight_decay):
         ext = require_native()

@@ -788,3 +788,51 @@ def test_fused_head_large_batch():
         # canary: the queue must still accept launches after the head
         torch.zeros(8, device='cuda:0').add_(1.0)
         torch.cuda.synchronize()
+
+
+@needs_gpu
+def test_lm_bf16_shadow_weights():
+    """bf16 shadow weights: the fused optimizer keeps every shadow equal to
+    bf16(master) across steps, and the shadow forward matches the autocast
+    forward at init."""
+    from heterofl_amd.config import default_config
+    from heterofl_amd.control import process_control, CONTROL_FIELDS
+    from heterofl_amd.fed.batched_lm import (make_batched_transformer,
+                                             enable_bf16_shadows,
+                                             refresh_shadows, lm_masked_ce)
+    from heterofl_amd.ops.fused import FusedClipSGD
+    cfg = default_config()
+    control = '1_4_1_iid_fix_a1_bn_1_1'
+    cfg['control'] = dict(zip(CONTROL_FIELDS, control.split('_')))
+    cfg['control_name'] = control
+    cfg['data_name'] = 'WikiText2'
+    cfg['model_name'] = 'transformer'
+    cfg['device'] = 'cuda:0'
+    process_control(cfg)
+    cfg['num_tokens'] = 300
+    torch.manual_seed(0)
+    model = make_batched_transformer(cfg, 1.0, 2).to('cuda:0')
+    model.train(True)
+    smap = enable_bf16_shadows(model)
+    refresh_shadows(model)
+    tokens = torch.randint(0, 300, (2, 3, cfg['bptt']), device='cuda:0')
+    # shadow forward ~ autocast forward (same bf16 weight values)
+    torch.manual_seed(1)
+    logits_sh = model(tokens)
+    assert logits_sh.dtype == torch.bfloat16
+    params = [p for p in model.parameters() if p.requires_grad]
+    for p in params:
+        p.grad = torch.zeros_like(p)
+    bufs = [torch.zeros_like(p) for p in params]
+    fopt = FusedClipSGD(params, [p.grad for p in params], bufs, 2, 'cuda:0',
+                        shadows=[smap.get(id(p)) for p in params])
+    losses = lm_masked_ce(logits_sh, tokens, None)
+    losses.sum().backward()
+    fopt.step(1.0, 0.1, 0.9, 5e-4)
+    torch.cuda.synchronize()
+    # every shadow must equal bf16(master) after the fused step
+    for p in params:
+        sh = smap.get(id(p))
+        if sh is not None:
+            assert torch.equal(sh, p.detach().to(torch.bfloat16)), 'stale shadow'
+    assert torch.isfinite(losses).all()
