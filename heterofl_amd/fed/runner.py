@@ -326,10 +326,16 @@ class FedRunner:
                 (self.dist_ctx.rank, self.dist_ctx.world_size)
             # the reference's stats pass iterates the TRAIN loader, i.e.
             # with train-time augmentation active (train_classifier_fed.py:
-            # 127-138 over the train dataset's transforms)
+            # 127-138 over the train dataset's transforms).  Activations run
+            # in the engine's compute dtype: fp32 MFMA is 1/16 the bf16 rate,
+            # and the model trains in bf16 anyway (BN statistics accumulate
+            # in fp32 inside the kernel either way).
+            amp = getattr(trainer, '_amp', False)
             for b in range(rank, total_b, world):
                 batch = img[b * bs:min((b + 1) * bs, n_total)]
                 x = trainer.augment(batch, train=True)
+                if amp:
+                    x = x.to(torch.bfloat16)
                 bmodel(x)
         finally:
             for _, m in bn_mods:

@@ -44,18 +44,20 @@ def timeit(fn, iters=20):
     return s.elapsed_time(e) / iters * 1000  # us
 
 
-print(f'{"shape":6} {"native_us":>10} {"miopen_us":>10} {"ratio":>6}')
-tot_n = tot_m = 0.0
-for name, G, N, Cin, H, Cout, k, s, p in SHAPES:
-    x = torch.randn(N, G * Cin, H, H, device=dev, dtype=dt)
-    w = torch.randn(G * Cout, Cin, k, k, device=dev) * 0.1
-    tn = timeit(lambda: ext.conv_fwd(x, w, torch.Tensor(), torch.Tensor(),
-                                     G, s, p, 0))
-    tm = timeit(lambda: F.conv2d(x, w.to(dt), None, s, p, 1, G))
-    tot_n += tn
-    tot_m += tm
-    print(f'{name:6} {tn:10.1f} {tm:10.1f} {tn / tm:6.2f}')
-print(f'TOTAL  {tot_n:10.1f} {tot_m:10.1f} {tot_n / tot_m:6.2f}')
+for dt in (torch.float32, torch.bfloat16):
+    print(f'-- dtype {dt} --')
+    print(f'{"shape":6} {"native_us":>10} {"miopen_us":>10} {"ratio":>6}')
+    tot_n = tot_m = 0.0
+    for name, G, N, Cin, H, Cout, k, s, p in SHAPES:
+        x = torch.randn(N, G * Cin, H, H, device=dev, dtype=dt)
+        w = torch.randn(G * Cout, Cin, k, k, device=dev) * 0.1
+        tn = timeit(lambda: ext.conv_fwd(x, w, torch.Tensor(), torch.Tensor(),
+                                         G, s, p, 0))
+        tm = timeit(lambda: F.conv2d(x, w.to(dt), None, s, p, 1, G))
+        tot_n += tn
+        tot_m += tm
+        print(f'{name:6} {tn:10.1f} {tm:10.1f} {tn / tm:6.2f}')
+    print(f'TOTAL  {tot_n:10.1f} {tot_m:10.1f} {tot_n / tot_m:6.2f}')
 
 # whole stats pass A/B
 from heterofl_amd.config import default_config
