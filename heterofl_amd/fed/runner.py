@@ -320,7 +320,11 @@ class FedRunner:
                 img = img.to(device)
                 self._stats_img = img
             n_total = img.size(0)
-            bs = 500 if device.type == 'cuda' else cfg['batch_size']['train']
+            # cumulative BN is batch-size invariant for equal-size batches,
+            # so the GPU pass uses large batches (fewer launches, fuller
+            # kernels); HETEROFL_STATS_BS tunes it
+            bs = int(os.environ.get('HETEROFL_STATS_BS', '2500')) \
+                if device.type == 'cuda' else cfg['batch_size']['train']
             total_b = (n_total + bs - 1) // bs
             rank, world = (0, 1) if self.dist_ctx is None else \
                 (self.dist_ctx.rank, self.dist_ctx.world_size)

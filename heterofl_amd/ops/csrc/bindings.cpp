@@ -57,6 +57,15 @@ at::Tensor lm_ce_bwd(at::Tensor logits, at::Tensor labels, at::Tensor mask,
 std::vector<at::Tensor> head_bwd(at::Tensor dscores, at::Tensor pooled,
                                  at::Tensor w, int64_t R, int64_t H,
                                  int64_t W, bool bf16_feat, bool want_db);
+void rng_bump(at::Tensor seed);
+std::vector<at::Tensor> gelu_drop_fwd(at::Tensor x, at::Tensor seed,
+                                      double rate, double p);
+at::Tensor gelu_drop_bwd(at::Tensor dy, at::Tensor x, at::Tensor mask,
+                         double rate, double p);
+std::vector<at::Tensor> res_drop_fwd(at::Tensor src, at::Tensor h,
+                                     at::Tensor seed, double rate, double p);
+at::Tensor drop_scale_bwd(at::Tensor dt, at::Tensor mask, double rate,
+                          double p);
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("bn_relu_fwd", &bn_relu_fwd, "fused sBN+ReLU forward");
@@ -83,4 +92,11 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("ln_bwd", &ln_bwd, "fused per-client LayerNorm backward");
     m.def("lm_ce_fwd", &lm_ce_fwd, "fused vocab-masked LM CE forward");
     m.def("lm_ce_bwd", &lm_ce_bwd, "fused vocab-masked LM CE backward");
+    m.def("rng_bump", &rng_bump, "advance the device RNG seed cell");
+    m.def("gelu_drop_fwd", &gelu_drop_fwd, "fused scaler+GELU+dropout fwd");
+    m.def("gelu_drop_bwd", &gelu_drop_bwd, "fused scaler+GELU+dropout bwd");
+    m.def("res_drop_fwd", &res_drop_fwd,
+          "fused scaler+dropout+residual-add fwd");
+    m.def("drop_scale_bwd", &drop_scale_bwd,
+          "dropout-mask * 1/rate backward");
 }

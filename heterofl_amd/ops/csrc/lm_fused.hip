@@ -1,0 +1,209 @@
+// Fused LM glue kernels (the LM step carried ~300 small
+// torch elementwise launches — scaler divisions, GELU, dropout, residual
+// adds).  Two elementwise fusions cover the encoder layer's non-GEMM work
+// (reference semantics: src/models/transformer.py:112-119):
+//   gelu_drop:  y = dropout(gelu(x / rate), p)          [Scaler+GELU+drop]
+//   res_drop:   t = src + dropout(h / rate, p)          [Scaler+drop+residual]
+// The dropout mask is drawn by a counter-hash RNG keyed on a device u64
+// seed cell (bumped by its own tiny kernel so hipGraph replays draw fresh
+// masks) and SAVED as uint8 for the exact one-kernel backward.
+#include "common.h"
+
+__device__ __forceinline__ uint32_t mix32(uint32_t h) {
+    h ^= h >> 16;
+    h *= 0x85EBCA6Bu;
+    h ^= h >> 13;
+    h *= 0xC2B2AE35u;
+    h ^= h >> 16;
+    return h;
+}
+
+__device__ __forceinline__ uint32_t rng_at(unsigned long long seed, long i) {
+    uint32_t h = (uint32_t)(seed & 0xFFFFFFFFull) * 0x9E3779B1u
+                 + (uint32_t)(seed >> 32) * 0x85EBCA77u
+                 + (uint32_t)i * 0xC2B2AE3Du + (uint32_t)(i >> 32);
+    return mix32(h);
+}
+
+__global__ void rng_bump_kernel(unsigned long long* seed) {
+    *seed = *seed * 6364136223846793005ull + 1442695040888963407ull;
+}
+
+// GELU (erf form, torch default) and its derivative
+__device__ __forceinline__ float gelu_f(float v) {
+    return 0.5f * v * (1.f + erff(v * 0.70710678f));
+}
+__device__ __forceinline__ float dgelu_f(float v) {
+    return 0.5f * (1.f + erff(v * 0.70710678f))
+           + v * 0.39894228f * __expf(-0.5f * v * v);
+}
+
+template <typename T>
+__global__ void __launch_bounds__(256)
+gelu_drop_fwd_kernel(const T* __restrict__ x, T* __restrict__ y,
+                     unsigned char* __restrict__ mask,
+                     const unsigned long long* __restrict__ seed, long n,
+                     float inv_rate, float p, float inv_keep) {
+    const unsigned long long sd = *seed;
+    const uint32_t thresh = (uint32_t)(p * 4294967296.0);
+    for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+         i += (long)gridDim.x * blockDim.x) {
+        const float g = gelu_f(ld_f32(x + i) * inv_rate);
+        unsigned char keep = 1;
+        if (thresh) keep = rng_at(sd, i) >= thresh;
+        if (mask) mask[i] = keep;
+        st_f32(y + i, keep ? g * inv_keep : 0.f);
+    }
+}
+
+template <typename T>
+__global__ void __launch_bounds__(256)
+gelu_drop_bwd_kernel(const T* __restrict__ dy, const T* __restrict__ x,
+                     const unsigned char* __restrict__ mask,
+                     T* __restrict__ dx, long n, float inv_rate,
+                     float inv_keep) {
+    for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+         i += (long)gridDim.x * blockDim.x) {
+        const float keep = mask ? (float)mask[i] : 1.f;
+        const float v = ld_f32(x + i) * inv_rate;
+        st_f32(dx + i,
+               ld_f32(dy + i) * keep * inv_keep * dgelu_f(v) * inv_rate);
+    }
+}
+
+template <typename T>
+__global__ void __launch_bounds__(256)
+res_drop_fwd_kernel(const T* __restrict__ src, const T* __restrict__ h,
+                    T* __restrict__ t, unsigned char* __restrict__ mask,
+                    const unsigned long long* __restrict__ seed, long n,
+                    float inv_rate, float p, float inv_keep) {
+    const unsigned long long sd = *seed;
+    const uint32_t thresh = (uint32_t)(p * 4294967296.0);
+    for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+         i += (long)gridDim.x * blockDim.x) {
+        unsigned char keep = 1;
+        if (thresh) keep = rng_at(sd, i) >= thresh;
+        if (mask) mask[i] = keep;
+        const float hv = keep ? ld_f32(h + i) * inv_rate * inv_keep : 0.f;
+        st_f32(t + i, ld_f32(src + i) + hv);
+    }
+}
+
+template <typename T>
+__global__ void __launch_bounds__(256)
+drop_scale_bwd_kernel(const T* __restrict__ dt,
+                      const unsigned char* __restrict__ mask,
+                      T* __restrict__ dh, long n, float inv_rate,
+                      float inv_keep) {
+    for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+         i += (long)gridDim.x * blockDim.x) {
+        const float keep = mask ? (float)mask[i] : 1.f;
+        st_f32(dh + i, ld_f32(dt + i) * keep * inv_keep * inv_rate);
+    }
+}
+
+// ------------------------------------------------------------ host layer
+#include <torch/extension.h>
+#include <ATen/hip/HIPContext.h>
+
+#define DISPATCH_LMF(t, ...)                                                  \
+    if ((t) == at::kFloat) { using scalar_t = float; __VA_ARGS__; }           \
+    else if ((t) == at::kBFloat16) { using scalar_t = __hip_bfloat16; __VA_ARGS__; } \
+    else { TORCH_CHECK(false, "unsupported dtype"); }
+
+static dim3 lmf_grid(long n) {
+    long blocks = (n + 256 * 8 - 1) / (256 * 8);
+    if (blocks < 1) blocks = 1;
+    if (blocks > 4096) blocks = 4096;
+    return dim3((unsigned)blocks);
+}
+
+void rng_bump(at::Tensor seed) {
+    auto stream = at::hip::getCurrentHIPStream();
+    hipLaunchKernelGGL(rng_bump_kernel, dim3(1), dim3(1), 0, stream,
+                       (unsigned long long*)seed.data_ptr());
+}
+
+std::vector<at::Tensor> gelu_drop_fwd(at::Tensor x, at::Tensor seed,
+                                      double rate, double p) {
+    TORCH_CHECK(x.is_cuda() && x.is_contiguous());
+    const long n = x.numel();
+    auto y = at::empty_like(x);
+    const bool dropping = p > 0.0;
+    auto mask = at::empty({dropping ? n : 0}, x.options().dtype(at::kByte));
+    auto stream = at::hip::getCurrentHIPStream();
+    DISPATCH_LMF(x.scalar_type(), {
+        hipLaunchKernelGGL(gelu_drop_fwd_kernel<scalar_t>, lmf_grid(n),
+                           dim3(256), 0, stream,
+                           (const scalar_t*)x.data_ptr(),
+                           (scalar_t*)y.data_ptr(),
+                           dropping ? mask.data_ptr<unsigned char>() : nullptr,
+                           (const unsigned long long*)seed.data_ptr(), n,
+                           (float)(1.0 / rate), (float)p,
+                           (float)(1.0 / (1.0 - p)));
+    });
+    if (dropping) rng_bump(seed);
+    return {y, mask};
+}
+
+at::Tensor gelu_drop_bwd(at::Tensor dy, at::Tensor x, at::Tensor mask,
+                         double rate, double p) {
+    const long n = x.numel();
+    auto dyc = dy.contiguous();
+    auto dx = at::empty_like(x);
+    auto stream = at::hip::getCurrentHIPStream();
+    DISPATCH_LMF(x.scalar_type(), {
+        hipLaunchKernelGGL(gelu_drop_bwd_kernel<scalar_t>, lmf_grid(n),
+                           dim3(256), 0, stream,
+                           (const scalar_t*)dyc.data_ptr(),
+                           (const scalar_t*)x.data_ptr(),
+                           mask.numel() ? mask.data_ptr<unsigned char>()
+                                        : nullptr,
+                           (scalar_t*)dx.data_ptr(), n, (float)(1.0 / rate),
+                           (float)(1.0 / (1.0 - p)));
+    });
+    return dx;
+}
+
+std::vector<at::Tensor> res_drop_fwd(at::Tensor src, at::Tensor h,
+                                     at::Tensor seed, double rate, double p) {
+    TORCH_CHECK(src.is_cuda() && src.is_contiguous() && h.is_contiguous());
+    TORCH_CHECK(src.numel() == h.numel());
+    const long n = src.numel();
+    auto t = at::empty_like(src);
+    const bool dropping = p > 0.0;
+    auto mask = at::empty({dropping ? n : 0},
+                          src.options().dtype(at::kByte));
+    auto stream = at::hip::getCurrentHIPStream();
+    DISPATCH_LMF(src.scalar_type(), {
+        hipLaunchKernelGGL(res_drop_fwd_kernel<scalar_t>, lmf_grid(n),
+                           dim3(256), 0, stream,
+                           (const scalar_t*)src.data_ptr(),
+                           (const scalar_t*)h.data_ptr(),
+                           (scalar_t*)t.data_ptr(),
+                           dropping ? mask.data_ptr<unsigned char>() : nullptr,
+                           (const unsigned long long*)seed.data_ptr(), n,
+                           (float)(1.0 / rate), (float)p,
+                           (float)(1.0 / (1.0 - p)));
+    });
+    if (dropping) rng_bump(seed);
+    return {t, mask};
+}
+
+at::Tensor drop_scale_bwd(at::Tensor dt, at::Tensor mask, double rate,
+                          double p) {
+    const long n = dt.numel();
+    auto dtc = dt.contiguous();
+    auto dh = at::empty_like(dtc);
+    auto stream = at::hip::getCurrentHIPStream();
+    DISPATCH_LMF(dt.scalar_type(), {
+        hipLaunchKernelGGL(drop_scale_bwd_kernel<scalar_t>, lmf_grid(n),
+                           dim3(256), 0, stream,
+                           (const scalar_t*)dtc.data_ptr(),
+                           mask.numel() ? mask.data_ptr<unsigned char>()
+                                        : nullptr,
+                           (scalar_t*)dh.data_ptr(), n, (float)(1.0 / rate),
+                           (float)(1.0 / (1.0 - p)));
+    });
+    return dh;
+}

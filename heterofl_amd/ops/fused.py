@@ -315,3 +315,71 @@ class _FusedLMCE(torch.autograd.Function):
 
 def fused_lm_ce(logits, tokens, mask, R):
     return _FusedLMCE.apply(logits, tokens, mask, R)
+
+
+# ------------------------------------------------------- LM glue fusions
+_rng_cells = {}
+
+
+def _rng_cell(device):
+    """Per-device u64 seed cell read by the fused dropout kernels and bumped
+    device-side after each draw — a stable pointer, so hipGraph replays draw
+    fresh masks (torch's philox handles its own ops; these are ours)."""
+    key = str(device)
+    if key not in _rng_cells:
+        _rng_cells[key] = torch.tensor(
+            [torch.initial_seed() & 0x7FFFFFFFFFFFFFFF],
+            dtype=torch.int64, device=device)
+    return _rng_cells[key]
+
+
+class _FusedGeluDropout(torch.autograd.Function):
+    """dropout(gelu(x / rate), p) in one kernel each way, mask saved
+    (reference chain: src/models/transformer.py:116 —
+    dropout1(GELU(scaler(linear1(src)))))."""
+
+    @staticmethod
+    def forward(ctx, x, rate, p):
+        ext = require_native()
+        x = x.contiguous()
+        y, mask = ext.gelu_drop_fwd(x, _rng_cell(x.device), rate, p)
+        ctx.save_for_backward(x, mask)
+        ctx.meta = (rate, p)
+        return y
+
+    @staticmethod
+    def backward(ctx, dy):
+        ext = require_native()
+        x, mask = ctx.saved_tensors
+        rate, p = ctx.meta
+        return ext.gelu_drop_bwd(dy, x, mask, rate, p), None, None
+
+
+def fused_gelu_dropout(x, rate, p):
+    return _FusedGeluDropout.apply(x, rate, p)
+
+
+class _FusedResDropout(torch.autograd.Function):
+    """src + dropout(h / rate, p) in one kernel; backward is dt for src and
+    one mask-scale kernel for h (the LN that follows is already fused)."""
+
+    @staticmethod
+    def forward(ctx, src, h, rate, p):
+        ext = require_native()
+        t, mask = ext.res_drop_fwd(src.contiguous(), h.contiguous(),
+                                   _rng_cell(src.device), rate, p)
+        ctx.save_for_backward(mask)
+        ctx.meta = (rate, p)
+        return t
+
+    @staticmethod
+    def backward(ctx, dt):
+        ext = require_native()
+        (mask,) = ctx.saved_tensors
+        rate, p = ctx.meta
+        dh = ext.drop_scale_bwd(dt, mask, rate, p)
+        return dt, dh, None, None
+
+
+def fused_res_dropout(src, h, rate, p):
+    return _FusedResDropout.apply(src, h, rate, p)
