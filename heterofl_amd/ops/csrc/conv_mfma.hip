@@ -1390,6 +1390,18 @@ at::Tensor conv_bwd_weight(at::Tensor dy, at::Tensor x, int64_t groups,
                          * gm.G;
     int splitp = 1;
     while (mk_tiles * splitp < split_target() && splitp * BK * 4 < P) splitp *= 2;
+    // small-P shapes (P = N*OH*OW few hundred) leave each block only a
+    // handful of BK-steps and the per-step prologue dominates; push the
+    // split further so the chip trades a larger partial reduce for more
+    // concurrent blocks (HETEROFL_BWDW_SMALLP=0 disables)
+    static const bool smallp_boost = [] {
+        const char* e = std::getenv("HETEROFL_BWDW_SMALLP");
+        return !(e && e[0] == '0');
+    }();
+    if (smallp_boost)
+        while (P <= 2048 && mk_tiles * splitp < 4 * split_target()
+               && splitp * BK < P)
+            splitp *= 2;
     auto dw = at::empty({(long)gm.G * gm.Cout, gm.Cin, gm.khw, gm.khw},
                         x.options().dtype(at::kFloat));
     auto stream = at::hip::getCurrentHIPStream();
