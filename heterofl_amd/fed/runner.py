@@ -374,6 +374,68 @@ class FedRunner:
                 tgt.num_batches_tracked.fill_(k)
         test_model.train(False)
 
+    # -------------------------------------------------------- fast GPU eval
+    def _fast_vision_eval(self, test_model, metric, logger, cfg, rank, world):
+        """Loader-free evaluation: stage the raw uint8 test set on device
+        once, run ONE pass of large-batch forwards for the UNMASKED logits,
+        then derive every user's Local (label-split-masked, reference:
+        src/models/resnet.py:152-157) and the Global metrics from those
+        logits.  Same math as the per-user loader loop (the mask only
+        rewrites logits), ~100x fewer host round-trips.  Users and samples
+        shard across ranks exactly like the loader path."""
+        from ..models.functional import masked_cross_entropy
+        from .batched import DeviceAugment
+        import torch.nn.functional as F
+        ds = self.dataset['test']
+        device = torch.device(cfg['device'])
+        staged = getattr(self, '_eval_staged', None)
+        if staged is None:
+            img = ds.img
+            if img.dim() == 3:
+                img = img.unsqueeze(-1)
+            aug = DeviceAugment(cfg['data_name'], device)
+            labels = torch.tensor(ds.target, device=device)
+            staged = (img.to(device), labels, aug)
+            self._eval_staged = staged
+        img, labels, aug = staged
+        n = img.size(0)
+        bs = 2048
+        scores = []
+        for b0 in range(0, n, bs):
+            x = aug(img[b0:b0 + bs], train=False)
+            out = test_model({'img': x, 'label': labels[b0:b0 + bs]})
+            scores.append(out['score'].float())
+        scores = torch.cat(scores)
+        names = cfg['metric_name']['test']
+        # Local metrics: this rank's share of users
+        for m in range(rank, cfg['num_users'], world):
+            idx = torch.as_tensor(self.data_split['test'][m],
+                                  dtype=torch.long, device=device)
+            if idx.numel() == 0:
+                continue
+            y = labels[idx]
+            ls = torch.tensor(self.label_split[m], device=device)
+            sc = scores[idx]
+            if cfg['mask']:
+                sc, loss = masked_cross_entropy(sc, y, ls,
+                                                cfg['classes_size'])
+            else:
+                loss = F.cross_entropy(sc, y)
+            ev = metric.evaluate(names['Local'],
+                                 {'label': y, 'label_split': ls},
+                                 {'score': sc, 'loss': loss})
+            if logger:
+                logger.append(ev, 'test', idx.numel())
+        # Global metrics: this rank's share of samples, unmasked logits
+        gidx = torch.arange(rank, n, world, device=device)
+        gy = labels[gidx]
+        gs = scores[gidx]
+        ev = metric.evaluate(names['Global'], {'label': gy},
+                             {'score': gs,
+                              'loss': F.cross_entropy(gs, gy)})
+        if logger:
+            logger.append(ev, 'test', gidx.numel())
+
     # ------------------------------------------------------------------- test
     def test(self, test_model, epoch):
         """Per-client Local metrics + Global metrics
@@ -397,7 +459,12 @@ class FedRunner:
         logger = self.logger
         with torch.no_grad():
             test_model.train(False)
-            if not self.is_lm:
+            if not self.is_lm and torch.cuda.is_available() and \
+                    os.environ.get('HETEROFL_FAST_EVAL', '1') == '1' and \
+                    hasattr(self.dataset['test'], 'img'):
+                self._fast_vision_eval(test_model, metric, logger, cfg,
+                                       rank, world)
+            elif not self.is_lm:
                 for m in range(rank, cfg['num_users'], world):
                     loader = make_data_loader(
                         {'test': SplitDataset(self.dataset['test'],

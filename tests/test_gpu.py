@@ -883,3 +883,36 @@ def test_fused_gelu_res_dropout():
     # two calls draw different masks (device seed cell bumps)
     y2 = fused_gelu_dropout(x.detach(), 1.0, p)
     assert not torch.equal((y != 0), (y2 != 0))
+
+
+@needs_gpu
+def test_fast_eval_matches_loader_eval(base_cfg, monkeypatch):
+    """The loader-free staged evaluation reproduces the per-user loader
+    loop's logged means exactly (same logits math, different batching)."""
+    from heterofl_amd.data import fetch_dataset, split_dataset
+    from heterofl_amd.fed import FedRunner
+    from heterofl_amd.logger import Logger
+    from heterofl_amd.models import make_model
+    from heterofl_amd.utils import process_dataset, make_optimizer
+    cfg = make_cfg(base_cfg, '1_6_0.5_non-iid-2_fix_a1_bn_1_1',
+                   data_name='CIFAR10', model_name='resnet18')
+    cfg['device'] = 'cuda:0'
+    cfg['engine'] = 'batched'
+    torch.manual_seed(0)
+    ds = fetch_dataset('CIFAR10', synthetic=True, synthetic_size=240)
+    process_dataset(ds, cfg)
+    data_split, label_split = split_dataset(ds, 6, 'non-iid-2', 10)
+    model = make_model(cfg).to('cuda:0')
+    runner = FedRunner(cfg, ds, data_split, label_split, model,
+                       make_optimizer(model, cfg['lr'], cfg))
+    tm = runner.stats()
+    means = {}
+    for fast in ('1', '0'):
+        monkeypatch.setenv('HETEROFL_FAST_EVAL', fast)
+        logger = Logger(None)
+        runner.logger = logger
+        runner.test(tm, 1)
+        means[fast] = dict(logger.mean)
+    for k, v in means['0'].items():
+        assert k in means['1'], k
+        assert abs(means['1'][k] - v) < 5e-3, (k, means['1'][k], v)
