@@ -1390,13 +1390,13 @@ at::Tensor conv_bwd_weight(at::Tensor dy, at::Tensor x, int64_t groups,
                          * gm.G;
     int splitp = 1;
     while (mk_tiles * splitp < split_target() && splitp * BK * 4 < P) splitp *= 2;
-    // small-P shapes (P = N*OH*OW few hundred) leave each block only a
-    // handful of BK-steps and the per-step prologue dominates; push the
-    // split further so the chip trades a larger partial reduce for more
-    // concurrent blocks (HETEROFL_BWDW_SMALLP=0 disables)
+    // measured NEGATIVE (bwdw_sweep r2): pushing split-P further on small-P
+    // shapes (L3: splitp 8 -> 95.6us vs 89.4us at splitp 2) trades too much
+    // partial-reduce traffic for the extra blocks; default off, kept gated
+    // for future study (HETEROFL_BWDW_SMALLP=1)
     static const bool smallp_boost = [] {
         const char* e = std::getenv("HETEROFL_BWDW_SMALLP");
-        return !(e && e[0] == '0');
+        return e && e[0] == '1';
     }();
     if (smallp_boost)
         while (P <= 2048 && mk_tiles * splitp < 4 * split_target()
