@@ -1389,7 +1389,13 @@ at::Tensor conv_bwd_weight(at::Tensor dy, at::Tensor x, int64_t groups,
     const int mk_tiles = ((gm.Cout + BM - 1) / BM) * ((K + BP - 1) / BP)
                          * gm.G;
     int splitp = 1;
-    while (mk_tiles * splitp < split_target() && splitp * BK * 4 < P) splitp *= 2;
+    // measured (bwdw2 r2): small-P shapes that already have >=512 (m,k)
+    // tiles need no split at all — L3 bwdW 89.4us (splitp 2) -> 74.5us
+    // (splitp 1) = 1.00x MIOpen; splitting only pays when tiles are scarce
+    // or P is long enough to amortize the partial reduce
+    if (P > 1024 || mk_tiles < 512)
+        while (mk_tiles * splitp < split_target() && splitp * BK * 4 < P)
+            splitp *= 2;
     // measured NEGATIVE (bwdw_sweep r2): pushing split-P further on small-P
     // shapes (L3: splitp 8 -> 95.6us vs 89.4us at splitp 2) trades too much
     // partial-reduce traffic for the extra blocks; default off, kept gated

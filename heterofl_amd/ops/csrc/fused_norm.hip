@@ -60,6 +60,87 @@ bn_relu_fwd_kernel(const T* __restrict__ x, const float* __restrict__ gamma,
     }
 }
 
+// --------------------------------------------- forward, large-batch (3-kernel)
+// The one-workgroup-per-channel kernel serializes when N*HW is large (the
+// sBN statistics / eval pass runs N=2500): with C blocks only, 3.3 ms per
+// call was the stats pass's real bound (kstats_r02).  Large batches use a
+// grid-parallel 3-kernel path instead: per-(channel, sample-chunk) partial
+// sums -> per-channel finalize -> grid-parallel normalize+ReLU.
+template <typename T>
+__global__ void __launch_bounds__(256)
+bn_sums_kernel(const T* __restrict__ x, float* __restrict__ partials, int N,
+               int C, int HW, int S) {
+    const int c = blockIdx.x;
+    const int s = blockIdx.y;
+    const int chunk = (N + S - 1) / S;
+    const int n1 = min(N, (s + 1) * chunk);
+    const long chan_off = (long)c * HW;
+    const long samp_stride = (long)C * HW;
+    float s1 = 0.f, s2 = 0.f;
+    for (int n = s * chunk; n < n1; ++n) {
+        const T* base = x + n * samp_stride + chan_off;
+        for (int hw = threadIdx.x; hw < HW; hw += blockDim.x) {
+            const float v = ld_f32(base + hw);
+            s1 += v;
+            s2 += v * v;
+        }
+    }
+    __shared__ float scratch[2 * 256 / WAVE];
+    block_reduce2(s1, s2, scratch);
+    if (threadIdx.x == 0) {
+        partials[((long)c * S + s) * 2] = s1;
+        partials[((long)c * S + s) * 2 + 1] = s2;
+    }
+}
+
+__global__ void __launch_bounds__(256)
+bn_finalize_kernel(const float* __restrict__ partials,
+                   const float* __restrict__ gamma,
+                   const float* __restrict__ beta,
+                   float* __restrict__ mean_out,
+                   float* __restrict__ invstd_out,
+                   float* __restrict__ scale_shift, int C, int S,
+                   float inv_m, float eps) {
+    const int c = blockIdx.x * blockDim.x + threadIdx.x;
+    if (c >= C) return;
+    float s1 = 0.f, s2 = 0.f;
+    for (int s = 0; s < S; ++s) {
+        s1 += partials[((long)c * S + s) * 2];
+        s2 += partials[((long)c * S + s) * 2 + 1];
+    }
+    const float mean = s1 * inv_m;
+    const float var = fmaxf(s2 * inv_m - mean * mean, 0.f);
+    const float invstd = rsqrtf(var + eps);
+    mean_out[c] = mean;
+    invstd_out[c] = invstd;
+    const float g = gamma ? gamma[c] : 1.f;
+    const float b = beta ? beta[c] : 0.f;
+    const float scale = invstd * g;
+    scale_shift[2 * c] = scale;
+    scale_shift[2 * c + 1] = b - mean * scale;
+}
+
+template <typename T>
+__global__ void __launch_bounds__(256)
+bn_apply_relu_kernel(const T* __restrict__ x,
+                     const float* __restrict__ scale_shift,
+                     T* __restrict__ y, int N, int C, int HW, int S) {
+    const int c = blockIdx.x;
+    const int s = blockIdx.y;
+    const int chunk = (N + S - 1) / S;
+    const int n1 = min(N, (s + 1) * chunk);
+    const long chan_off = (long)c * HW;
+    const long samp_stride = (long)C * HW;
+    const float scale = scale_shift[2 * c];
+    const float shift = scale_shift[2 * c + 1];
+    for (int n = s * chunk; n < n1; ++n) {
+        const long off = n * samp_stride + chan_off;
+        for (int hw = threadIdx.x; hw < HW; hw += blockDim.x)
+            st_f32(y + off + hw, fmaxf(ld_f32(x + off + hw) * scale + shift,
+                                       0.f));
+    }
+}
+
 // ----------------------------------------------------------------- backward
 // relu mask from pre = xhat*g+b; dx = invstd*g*(dym - s1/M - xhat*s2/M).
 // STAGE: the channel's x and (relu-masked) dy are kept in LDS from the
@@ -244,6 +325,36 @@ std::vector<at::Tensor> bn_relu_fwd(at::Tensor x, at::Tensor gamma,
     auto invstd = at::empty({C}, opts);
     auto stream = at::hip::getCurrentHIPStream();
     const long stage_bytes = (long)N * HW * x.element_size();
+    const long M = (long)N * HW;
+    if (M > 64 * 1024) {
+        // large batch (stats/eval pass): grid-parallel 3-kernel path
+        const int S = std::max(1, std::min(N, (4096 + C - 1) / C));
+        auto partials = at::empty({(long)C * S * 2}, opts);
+        auto scale_shift = at::empty({2L * C}, opts);
+        DISPATCH_FT(x.scalar_type(), {
+            hipLaunchKernelGGL(bn_sums_kernel<scalar_t>, dim3(C, S),
+                               dim3(256), 0, stream,
+                               (const scalar_t*)x.data_ptr(),
+                               partials.data_ptr<float>(), N, C, HW, S);
+            hipLaunchKernelGGL(bn_finalize_kernel,
+                               dim3((C + 255) / 256), dim3(256), 0, stream,
+                               partials.data_ptr<float>(),
+                               gamma.defined() ? gamma.data_ptr<float>()
+                                               : nullptr,
+                               beta.defined() ? beta.data_ptr<float>()
+                                              : nullptr,
+                               mean.data_ptr<float>(),
+                               invstd.data_ptr<float>(),
+                               scale_shift.data_ptr<float>(), C, S,
+                               (float)(1.0 / M), (float)eps);
+            hipLaunchKernelGGL(bn_apply_relu_kernel<scalar_t>, dim3(C, S),
+                               dim3(256), 0, stream,
+                               (const scalar_t*)x.data_ptr(),
+                               scale_shift.data_ptr<float>(),
+                               (scalar_t*)y.data_ptr(), N, C, HW, S);
+        });
+        return {y, mean, invstd};
+    }
     DISPATCH_FT(x.scalar_type(), {
         if (stage_bytes <= 64 * 1024)
             hipLaunchKernelGGL((bn_relu_fwd_kernel<scalar_t, true>), dim3(C),

@@ -916,3 +916,25 @@ def test_fast_eval_matches_loader_eval(base_cfg, monkeypatch):
     for k, v in means['0'].items():
         assert k in means['1'], k
         assert abs(means['1'][k] - v) < 5e-3, (k, means['1'][k], v)
+
+
+@needs_gpu
+def test_fused_bn_relu_large_batch():
+    """The grid-parallel large-batch BN path (stats/eval batches) matches
+    the one-block-per-channel kernel and torch at N=2500."""
+    from heterofl_amd.ops import require_native
+    import torch.nn.functional as F
+    ext = require_native()
+    torch.manual_seed(0)
+    for C, HW in [(64, 1024), (512, 16)]:
+        N = 2500
+        side = int(HW ** 0.5)
+        x = torch.randn(N, C, side, side, device='cuda:0')
+        g = torch.rand(C, device='cuda:0') + 0.5
+        b = torch.randn(C, device='cuda:0')
+        y, mean, invstd = ext.bn_relu_fwd(x, g, b, 1e-5)
+        ref = F.relu(F.batch_norm(x, None, None, g, b, training=True,
+                                  eps=1e-5))
+        assert (y - ref).abs().max().item() < 1e-3, (C, HW)
+        ref_m = x.mean(dim=(0, 2, 3))
+        assert (mean - ref_m).abs().max().item() < 1e-4
