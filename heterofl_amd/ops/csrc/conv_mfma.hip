@@ -854,6 +854,159 @@ conv_fwd_staged_kernel(const T* __restrict__ x, const float* __restrict__ w,
 }
 
 
+// -------------------------- staged fwd, multi-sample (small output planes)
+// L4-family shapes have OH*OW < BP (e.g. 4x4 = 16 pixels), so a BP tile
+// spans ns = BP/OHW WHOLE samples: stage each sample's full (padded) input
+// plane for the k-slab's channel span and build the B tile from LDS — the
+// same im2col-in-LDS idea as conv_fwd_staged_kernel, which requires
+// OHW %% BP == 0 and therefore left these shapes on the scattered-gather
+// kernel (1.9s of kstats_r02).
+constexpr int MS_ROWS = 40;          // ns * ((OH-1)*stride + khw)
+
+template <typename T, typename TA, typename TB>
+__global__ void __launch_bounds__(256)
+conv_fwd_staged_ms_kernel(const T* __restrict__ x,
+                          const float* __restrict__ w,
+                          const float* __restrict__ bias,
+                          const T* __restrict__ residual, T* __restrict__ y,
+                          float* __restrict__ partial, ConvGeom gm,
+                          int splitk) {
+    __shared__ TA a_lds[BM][LDK];
+    __shared__ TB b_lds[BP][LDK];
+    __shared__ float win[WIN_CH][MS_ROWS][WIN_W];
+    const int g = blockIdx.z % gm.G;
+    const int sp = blockIdx.z / gm.G;
+    const int m0 = blockIdx.x * BM;
+    const int p0 = blockIdx.y * BP;
+    const int kk2 = gm.khw * gm.khw;
+    const int K = gm.Cin * kk2;
+    const int M = gm.Cout;
+    const int OHW = gm.OH * gm.OW;
+    const int P = gm.N * OHW;
+    const int HW = gm.H * gm.W;
+    const int tid = threadIdx.x;
+    const int l = tid & (WAVE - 1);
+    const int wave = tid / WAVE;
+    const int wm = (wave >> 1) * 32;
+    const int wp = (wave & 1) * 32;
+    const int nkc = ((K + BK - 1) / BK + splitk - 1) / splitk;
+    const int ks = sp * nkc * BK;
+    const int ke = min(K, ks + nkc * BK);
+    // tile geometry: ns whole samples, full output planes
+    const int ns = BP / OHW;
+    const int n0 = p0 / OHW;
+    const int rows_in = (gm.OH - 1) * gm.stride + gm.khw;
+    const long gch = ((long)g * gm.Cin) * HW;
+    const long sampstride = (long)gm.G * gm.Cin * HW;
+    const int tile_p = min(BP, P - p0);
+    const int mm_a = tid >> 2, kkb = (tid & 3) * 8;
+    const int pp_b = tid >> 2;
+    const int ls_b = pp_b / OHW;
+    const int pix_b = pp_b - ls_b * OHW;
+    const int ohl_b = (pix_b / gm.OW) * gm.stride;
+    const int iwl_b = (pix_b - (pix_b / gm.OW) * gm.OW) * gm.stride;
+    const int rowb_b = ls_b * rows_in;
+    float va[8];
+
+    f32x4 acc[2][2] = {};
+    for (int k0 = ks; k0 < ke; k0 += BK) {
+        const int cin0 = k0 / kk2;
+        const int cin1 = min(gm.Cin - 1, (k0 + BK - 1) / kk2);
+        const int nch = cin1 - cin0 + 1;
+        // stage ns sample planes (coalesced along the input width)
+        const int wtot = nch * ns * rows_in * gm.W;
+        for (int e = tid; e < wtot; e += 256) {
+            const int ww = e % gm.W;
+            const int rr = (e / gm.W) % rows_in;
+            const int lsl = (e / (gm.W * rows_in)) % ns;
+            const int cc = e / (gm.W * rows_in * ns);
+            const int ih = rr - gm.pad;
+            const int n = n0 + lsl;
+            win[cc][lsl * rows_in + rr][ww + gm.pad] =
+                (ih >= 0 && ih < gm.H && n < gm.N)
+                    ? ld_f32(x + (long)n * sampstride + gch
+                             + (long)(cin0 + cc) * HW + ih * gm.W + ww)
+                    : 0.f;
+        }
+        for (int e = tid; e < nch * ns * rows_in * gm.pad * 2; e += 256) {
+            const int side = e & 1;
+            const int pe = e >> 1;
+            const int pcol = pe % gm.pad;
+            const int rr = (pe / gm.pad) % (ns * rows_in);
+            const int cc = pe / (gm.pad * ns * rows_in);
+            win[cc][rr][side ? gm.pad + gm.W + pcol : pcol] = 0.f;
+        }
+        {   // A tile (weights, coalesced fp32)
+            const int m = m0 + mm_a;
+            const float* wrow = w + (long)(g * gm.Cout + m) * K + k0 + kkb;
+#pragma unroll
+            for (int j = 0; j < 8; ++j) {
+                const int k = k0 + kkb + j;
+                va[j] = (m < M && k < K) ? wrow[j] : 0.f;
+            }
+            st8_lds(&a_lds[mm_a][kkb], va);
+        }
+        __syncthreads();
+        {   // B tile from the LDS windows
+            float vb[8];
+            int k = k0 + kkb;
+            int cin = k / kk2, r = k - cin * kk2;
+            int kh = r / gm.khw, kw = r - kh * gm.khw;
+#pragma unroll
+            for (int j = 0; j < 8; ++j) {
+                vb[j] = (k + j < K && pp_b < tile_p)
+                            ? win[cin - cin0][rowb_b + ohl_b + kh]
+                                 [iwl_b + kw]
+                            : 0.f;
+                if (++kw == gm.khw) {
+                    kw = 0;
+                    if (++kh == gm.khw) {
+                        kh = 0;
+                        ++cin;
+                    }
+                }
+            }
+            st8_lds(&b_lds[pp_b][kkb], vb);
+        }
+        __syncthreads();
+#pragma unroll
+        for (int fm = 0; fm < 2; ++fm)
+#pragma unroll
+            for (int fp = 0; fp < 2; ++fp)
+                acc[fm][fp] = mfma_tile2<TA, TB>(
+                    &a_lds[wm + fm * 16 + (l & 15)][0],
+                    &b_lds[wp + fp * 16 + (l & 15)][0], acc[fm][fp]);
+        __syncthreads();
+    }
+    const long slab = (long)sp * gm.N * gm.G * gm.Cout * OHW;
+    const long ystride = (long)gm.G * gm.Cout * OHW;
+    const long ygch = (long)g * gm.Cout * OHW;
+#pragma unroll
+    for (int fm = 0; fm < 2; ++fm)
+#pragma unroll
+        for (int fp = 0; fp < 2; ++fp) {
+            const int pp = wp + fp * 16 + (l & 15);
+            const int lsl = pp / OHW;
+            const int pix = pp - lsl * OHW;
+            const long yb = (long)(n0 + lsl) * ystride + ygch + pix;
+#pragma unroll
+            for (int r = 0; r < 4; ++r) {
+                const int m = m0 + wm + fm * 16 + (l >> 4) * 4 + r;
+                if (m < M && pp < tile_p) {
+                    const long off = yb + (long)m * OHW;
+                    float v = acc[fm][fp][r];
+                    if (splitk == 1) {
+                        if (bias) v += bias[g * gm.Cout + m];
+                        if (residual) v += ld_f32(residual + off);
+                        st_f32(y + off, v);
+                    } else {
+                        partial[slab + off] = v;
+                    }
+                }
+            }
+        }
+}
+
 // ----------------------------------- staged stride-1 bwd-data (LDS window)
 // Same idea as conv_fwd_staged_kernel: a BP-pixel dx tile within one
 // sample spanning whole input rows has a dense dy window; stage it
@@ -1189,14 +1342,37 @@ at::Tensor conv_fwd(at::Tensor x, at::Tensor w, at::Tensor bias,
     if (splitk > 1)
         partial = at::empty({(long)splitk * y.numel()},
                             x.options().dtype(at::kFloat));
+    const int OHW = gm.OH * gm.OW;
     const bool staged = (P % BP == 0 || gm.N * gm.OH * gm.OW >= BP)
                         && (gm.OH * gm.OW) % BP == 0 && BP % gm.OW == 0
                         && (gm.W + 2 * gm.pad) <= WIN_W - 2
                         && ((BP / gm.OW - 1) * gm.stride + gm.khw) <= WIN_ROWS
                         && (BK / (gm.khw * gm.khw) + 2) <= WIN_CH
                         && std::getenv("HETEROFL_CONV_NO_STAGED") == nullptr;
+    // small output planes: a BP tile spans BP/OHW whole samples
+    const bool staged_ms = !staged && OHW < BP && BP % OHW == 0
+                          && (gm.W + 2 * gm.pad) <= WIN_W - 2
+                          && (BP / OHW) * ((gm.OH - 1) * gm.stride + gm.khw)
+                                 <= MS_ROWS
+                          && (BK / (gm.khw * gm.khw) + 2) <= WIN_CH
+                          && std::getenv("HETEROFL_CONV_NO_STAGED") == nullptr;
     DISPATCH_CONV_FT(x.scalar_type(), {
-        if (staged && !fp8) {
+        if (staged_ms && !fp8) {
+            hipLaunchKernelGGL((conv_fwd_staged_ms_kernel<scalar_t, scalar_t,
+                                                          scalar_t>),
+                               grid, dim3(256), 0, stream,
+                               (const scalar_t*)x.data_ptr(),
+                               w.data_ptr<float>(),
+                               bias.defined() ? bias.data_ptr<float>()
+                                              : nullptr,
+                               residual.defined()
+                                   ? (const scalar_t*)residual.data_ptr()
+                                   : nullptr,
+                               (scalar_t*)y.data_ptr(),
+                               splitk > 1 ? partial.data_ptr<float>()
+                                          : nullptr,
+                               gm, splitk);
+        } else if (staged && !fp8) {
             hipLaunchKernelGGL((conv_fwd_staged_kernel<scalar_t, scalar_t,
                                                        scalar_t>),
                                grid, dim3(256), 0, stream,
