@@ -1155,6 +1155,160 @@ conv_bwd_data_staged_kernel(const T* __restrict__ dy,
 }
 
 
+// ------------------- staged stride-1 bwd-data, multi-sample (small planes)
+// Same extension as conv_fwd_staged_ms_kernel: a BP-pixel dx tile spanning
+// BP/HW whole samples stages each sample's full padded dy plane per J-step.
+template <typename T, typename TA, typename TB>
+__global__ void __launch_bounds__(256)
+conv_bwd_data_staged_ms_kernel(const T* __restrict__ dy,
+                               const float* __restrict__ w,
+                               T* __restrict__ dx,
+                               float* __restrict__ partial, ConvGeom gm,
+                               int splitk) {
+    __shared__ TA a_lds[BM][LDK];
+    __shared__ TB b_lds[BP][LDK];
+    __shared__ float win[WIN_CH][MS_ROWS][WIN_W];
+    const int g = blockIdx.z % gm.G;
+    const int sp = blockIdx.z / gm.G;
+    const int c0 = blockIdx.x * BM;
+    const int q0 = blockIdx.y * BP;
+    const int kk2 = gm.khw * gm.khw;
+    const int J = gm.Cout * kk2;
+    const int K = gm.Cin * kk2;
+    const int HW = gm.H * gm.W;
+    const int OHW = gm.OH * gm.OW;
+    const int Q = gm.N * HW;
+    const int tid = threadIdx.x;
+    const int l = tid & (WAVE - 1);
+    const int wave = tid / WAVE;
+    const int wm = (wave >> 1) * 32;
+    const int wp = (wave & 1) * 32;
+    const int njc = ((J + BK - 1) / BK + splitk - 1) / splitk;
+    const int js = sp * njc * BK;
+    const int je = min(J, js + njc * BK);
+    const int ns = BP / HW;
+    const int n0 = q0 / HW;
+    // dy rows needed for a full dx plane: ohs = ih + pad - kh over ih in
+    // [0, H) -> rows [pad - (khw-1), H - 1 + pad]; stage rows_w rows from
+    // ohs_min with zero-fill outside [0, OH)
+    const int rows_w = gm.H + gm.khw - 1;
+    const int ohs_min = gm.pad - (gm.khw - 1);
+    const long dgch = (long)g * gm.Cout * OHW;
+    const long dystride = (long)gm.G * gm.Cout * OHW;
+    const int tile_q = min(BP, Q - q0);
+    const int cc_a = tid >> 2, jjb = (tid & 3) * 8;
+    const int ls_b = cc_a / HW;
+    const int pix_b = cc_a - ls_b * HW;
+    const int iwl = pix_b - (pix_b / gm.W) * gm.W;
+    const int rowl = (pix_b / gm.W) + (gm.khw - 1);
+    const int rowb = ls_b * rows_w;
+    float va[8];
+
+    f32x4 acc[2][2] = {};
+    for (int j0 = js; j0 < je; j0 += BK) {
+        const int co0 = j0 / kk2;
+        const int co1 = min(gm.Cout - 1, (j0 + BK - 1) / kk2);
+        const int nch = co1 - co0 + 1;
+        const int wtot = nch * ns * rows_w * gm.OW;
+        for (int e = tid; e < wtot; e += 256) {
+            const int ww = e % gm.OW;
+            const int rr = (e / gm.OW) % rows_w;
+            const int lsl = (e / (gm.OW * rows_w)) % ns;
+            const int cc = e / (gm.OW * rows_w * ns);
+            const int ohs = ohs_min + rr;
+            const int n = n0 + lsl;
+            win[cc][lsl * rows_w + rr][ww + gm.pad] =
+                (ohs >= 0 && ohs < gm.OH && n < gm.N)
+                    ? ld_f32(dy + (long)n * dystride + dgch
+                             + (long)(co0 + cc) * OHW + ohs * gm.OW + ww)
+                    : 0.f;
+        }
+        for (int e = tid; e < nch * ns * rows_w * gm.pad * 2; e += 256) {
+            const int side = e & 1;
+            const int pe = e >> 1;
+            const int pcol = pe % gm.pad;
+            const int rr = (pe / gm.pad) % (ns * rows_w);
+            const int cc = pe / (gm.pad * ns * rows_w);
+            win[cc][rr][side ? gm.pad + gm.OW + pcol : pcol] = 0.f;
+        }
+        {   // A tile: w rows = input channel
+            const int c = c0 + cc_a;
+            int j = j0 + jjb;
+            int cout = j / kk2, r = j - cout * kk2;
+            int kh = r / gm.khw, kw = r - kh * gm.khw;
+#pragma unroll
+            for (int j8 = 0; j8 < 8; ++j8) {
+                va[j8] = (c < gm.Cin && j + j8 < J)
+                             ? w[(long)(g * gm.Cout + cout) * K + c * kk2
+                                 + kh * gm.khw + kw]
+                             : 0.f;
+                if (++kw == gm.khw) {
+                    kw = 0;
+                    if (++kh == gm.khw) {
+                        kh = 0;
+                        ++cout;
+                    }
+                }
+            }
+            st8_lds(&a_lds[cc_a][jjb], va);
+        }
+        __syncthreads();
+        {   // B tile from the windows
+            float vb[8];
+            int j = j0 + jjb;
+            int cout = j / kk2, r = j - cout * kk2;
+            int kh = r / gm.khw, kw = r - kh * gm.khw;
+#pragma unroll
+            for (int j8 = 0; j8 < 8; ++j8) {
+                vb[j8] = (j + j8 < J && cc_a < tile_q)
+                             ? win[cout - co0][rowb + rowl - kh]
+                                  [iwl + 2 * gm.pad - kw]
+                             : 0.f;
+                if (++kw == gm.khw) {
+                    kw = 0;
+                    if (++kh == gm.khw) {
+                        kh = 0;
+                        ++cout;
+                    }
+                }
+            }
+            st8_lds(&b_lds[cc_a][jjb], vb);
+        }
+        __syncthreads();
+#pragma unroll
+        for (int fm = 0; fm < 2; ++fm)
+#pragma unroll
+            for (int fp = 0; fp < 2; ++fp)
+                acc[fm][fp] = mfma_tile2<TA, TB>(
+                    &a_lds[wm + fm * 16 + (l & 15)][0],
+                    &b_lds[wp + fp * 16 + (l & 15)][0], acc[fm][fp]);
+        __syncthreads();
+    }
+    const long slab = (long)sp * gm.N * gm.G * gm.Cin * HW;
+    const long xstride = (long)gm.G * gm.Cin * HW;
+    const long xgch = (long)g * gm.Cin * HW;
+#pragma unroll
+    for (int fm = 0; fm < 2; ++fm)
+#pragma unroll
+        for (int fp = 0; fp < 2; ++fp) {
+            const int qq = wp + fp * 16 + (l & 15);
+            const int lsl = qq / HW;
+            const int pix = qq - lsl * HW;
+            const long xb = (long)(n0 + lsl) * xstride + xgch + pix;
+#pragma unroll
+            for (int r = 0; r < 4; ++r) {
+                const int c = c0 + wm + fm * 16 + (l >> 4) * 4 + r;
+                if (c < gm.Cin && qq < tile_q) {
+                    const long off = xb + (long)c * HW;
+                    if (splitk == 1)
+                        st_f32(dx + off, acc[fm][fp][r]);
+                    else
+                        partial[slab + off] = acc[fm][fp][r];
+                }
+            }
+        }
+}
+
 // -------------------------------------- staged bwd-weight (x window in LDS)
 // The k-tile (hence the input-channel span) is FIXED per block, so the x
 // window only advances rows as the P loop walks output rows: stage the
@@ -1349,8 +1503,16 @@ at::Tensor conv_fwd(at::Tensor x, at::Tensor w, at::Tensor bias,
                         && ((BP / gm.OW - 1) * gm.stride + gm.khw) <= WIN_ROWS
                         && (BK / (gm.khw * gm.khw) + 2) <= WIN_CH
                         && std::getenv("HETEROFL_CONV_NO_STAGED") == nullptr;
-    // small output planes: a BP tile spans BP/OHW whole samples
-    const bool staged_ms = !staged && OHW < BP && BP % OHW == 0
+    // small output planes: a BP tile spans BP/OHW whole samples.
+    // MEASURED NEGATIVE at the training shapes (convbench r02c: L4 fwd
+    // 107->124us, L4d 67->101us — the 32 KB window slab costs more
+    // occupancy than the gather kernel loses to scattered loads); kept
+    // opt-in for future large-batch study
+    static const bool ms_on = [] {
+        const char* e = std::getenv("HETEROFL_CONV_MS");
+        return e && e[0] == '1';
+    }();
+    const bool staged_ms = ms_on && !staged && OHW < BP && BP % OHW == 0
                           && (gm.W + 2 * gm.pad) <= WIN_W - 2
                           && (BP / OHW) * ((gm.OH - 1) * gm.stride + gm.khw)
                                  <= MS_ROWS
@@ -1475,9 +1637,30 @@ at::Tensor conv_bwd_data(at::Tensor dy, at::Tensor w, int64_t groups,
                      && (BP / gm.W + gm.khw - 1) <= WIN_ROWS
                      && (BK / (gm.khw * gm.khw) + 2) <= WIN_CH
                      && std::getenv("HETEROFL_CONV_NO_STAGED") == nullptr;
+    const int HWd = gm.H * gm.W;
+    static const bool ms_on_d = [] {
+        const char* e = std::getenv("HETEROFL_CONV_MS");
+        return e && e[0] == '1';
+    }();
+    const bool stg_ms = ms_on_d && !stg && !s2 && gm.stride == 1
+                        && HWd < BP && BP % HWd == 0
+                        && (gm.OW + 2 * gm.pad) <= WIN_W - 2
+                        && (BP / HWd) * (gm.H + gm.khw - 1) <= MS_ROWS
+                        && (BK / (gm.khw * gm.khw) + 2) <= WIN_CH
+                        && std::getenv("HETEROFL_CONV_NO_STAGED") == nullptr;
     DISPATCH_CONV_FT(dy.scalar_type(), {
         const bool q = fp8 && dy.scalar_type() == at::kBFloat16;
-        if (stg && !q) {
+        if (stg_ms && !q) {
+            hipLaunchKernelGGL((conv_bwd_data_staged_ms_kernel<scalar_t,
+                                                               scalar_t,
+                                                               scalar_t>),
+                               grid, dim3(256), 0, stream,
+                               (const scalar_t*)dyc.data_ptr(),
+                               w.data_ptr<float>(), (scalar_t*)dx.data_ptr(),
+                               splitk > 1 ? partial.data_ptr<float>()
+                                          : nullptr,
+                               gm, splitk);
+        } else if (stg && !q) {
             hipLaunchKernelGGL((conv_bwd_data_staged_kernel<scalar_t,
                                                             scalar_t,
                                                             scalar_t>),
