@@ -143,7 +143,11 @@ class BNormReLU(nn.Module):
     def forward(self, x):
         if self.norm == 'none':
             if self.training and self.rate != 1.0:
-                x = x / self.rate
+                if native_ops.use_native(x):
+                    from ..ops.fused import fused_scaler
+                    x = fused_scaler(x, self.rate)
+                else:
+                    x = x / self.rate
             return F.relu(x)
         kind = 'bn' if self.norm == 'bn' else 'gn'
         if native_ops.use_native(x):
@@ -172,7 +176,12 @@ class BScaler(nn.Module):
         self.rate = rate
 
     def forward(self, x):
-        return x / self.rate if self.training else x
+        if not self.training or self.rate == 1.0:
+            return x
+        if native_ops.use_native(x):
+            from ..ops.fused import fused_scaler
+            return fused_scaler(x, self.rate)
+        return x / self.rate
 
 
 # ------------------------------------------------------------------ models

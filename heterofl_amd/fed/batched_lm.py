@@ -90,7 +90,12 @@ class SScaler(nn.Module):
         self.rate = rate
 
     def forward(self, x):
-        return x / self.rate if self.training else x
+        if not self.training or self.rate == 1.0:
+            return x
+        if native_ops.use_native(x):
+            from ..ops.fused import fused_scaler
+            return fused_scaler(x, self.rate)
+        return x / self.rate
 
 
 class SEmbedding(nn.Module):

@@ -1489,7 +1489,17 @@ at::Tensor conv_fwd(at::Tensor x, at::Tensor w, at::Tensor bias,
     const int kchunks = (K + BK - 1) / BK;
     const int tiles = ((M + BM - 1) / BM) * ((P + BP - 1) / BP) * gm.G;
     int splitk = 1;
-    while (tiles * splitk < split_target() && splitk * 2 <= kchunks / 2) splitk *= 2;
+    // same rule as bwdW (measured, L3): short reductions with enough tiles
+    // run best unsplit — the split's partial-write + reduce traffic beats
+    // the extra parallelism (HETEROFL_SPLIT_NOSKIP=1 restores the old
+    // always-split heuristic for A/B)
+    static const bool noskip_f = [] {
+        const char* e = std::getenv("HETEROFL_SPLIT_NOSKIP");
+        return e && e[0] == '1';
+    }();
+    if (noskip_f || K > 1024 || tiles < 512)
+        while (tiles * splitk < split_target() && splitk * 2 <= kchunks / 2)
+            splitk *= 2;
     dim3 grid((M + BM - 1) / BM, (P + BP - 1) / BP, gm.G * splitk);
     auto stream = at::hip::getCurrentHIPStream();
     at::Tensor partial;
@@ -1624,7 +1634,13 @@ at::Tensor conv_bwd_data(at::Tensor dy, at::Tensor w, int64_t groups,
     const int nyp = (qtile + BP - 1) / BP;
     const int tiles = ((gm.Cin + BM - 1) / BM) * nyp * (s2 ? 4 : 1) * gm.G;
     int splitk = 1;
-    while (tiles * splitk < split_target() && splitk * 2 <= jchunks / 2) splitk *= 2;
+    static const bool noskip_d = [] {
+        const char* e = std::getenv("HETEROFL_SPLIT_NOSKIP");
+        return e && e[0] == '1';
+    }();
+    if (noskip_d || J > 1024 || tiles < 512)
+        while (tiles * splitk < split_target() && splitk * 2 <= jchunks / 2)
+            splitk *= 2;
     dim3 grid((gm.Cin + BM - 1) / BM, nyp * (s2 ? 4 : 1), gm.G * splitk);
     auto stream = at::hip::getCurrentHIPStream();
     at::Tensor partial;

@@ -383,3 +383,27 @@ class _FusedResDropout(torch.autograd.Function):
 
 def fused_res_dropout(src, h, rate, p):
     return _FusedResDropout.apply(src, h, rate, p)
+
+
+class _FusedScaler(torch.autograd.Function):
+    """Standalone Scaler kernel (K5 of SURVEY §2b; reference
+    src/modules/modules.py:9-11): y = x / rate in training.  Runs the
+    drop_scale kernel with p=0 (pure 1/rate scale) both directions — used
+    by the norm='none' ablation, where the Scaler cannot cancel into a
+    following train-mode norm."""
+
+    @staticmethod
+    def forward(ctx, x, rate):
+        ext = require_native()
+        ctx.rate = rate
+        return ext.drop_scale_bwd(x.contiguous(), torch.Tensor(), rate, 0.0)
+
+    @staticmethod
+    def backward(ctx, dy):
+        ext = require_native()
+        return ext.drop_scale_bwd(dy.contiguous(), torch.Tensor(),
+                                  ctx.rate, 0.0), None
+
+
+def fused_scaler(x, rate):
+    return _FusedScaler.apply(x, rate)
