@@ -193,7 +193,8 @@ class SMultiheadAttention(nn.Module):
         q = q.permute(0, 1, 3, 2, 4).reshape(R * B * H, S, d)
         k = k.permute(0, 1, 3, 2, 4).reshape(R * B * H, S, d)
         v = v.permute(0, 1, 3, 2, 4).reshape(R * B * H, S, d)
-        if native_ops.use_native(q) and S <= 64 and d <= 32:
+        if native_ops.use_native(q) and d <= 32:
+            # S<=64 one-kernel path; larger S blockwise flash-style
             from ..ops.fused import fused_attention
             out = fused_attention(q, k, v, self.temperature)
         else:

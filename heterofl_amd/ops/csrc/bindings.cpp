@@ -46,6 +46,12 @@ std::vector<at::Tensor> attn_fwd(at::Tensor q, at::Tensor k, at::Tensor v,
 std::vector<at::Tensor> attn_bwd(at::Tensor dout, at::Tensor q, at::Tensor k,
                                  at::Tensor v, at::Tensor p,
                                  double temperature);
+std::vector<at::Tensor> attn_fwd_block(at::Tensor q, at::Tensor k,
+                                       at::Tensor v, double temperature);
+std::vector<at::Tensor> attn_bwd_block(at::Tensor dout, at::Tensor q,
+                                       at::Tensor k, at::Tensor v,
+                                       at::Tensor out, at::Tensor lse,
+                                       double temperature);
 std::vector<at::Tensor> ln_fwd(at::Tensor x, at::Tensor gamma,
                                at::Tensor beta, int64_t R, double eps);
 std::vector<at::Tensor> ln_bwd(at::Tensor dy, at::Tensor x, at::Tensor gamma,
@@ -88,6 +94,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("head_bwd", &head_bwd, "fused head backward");
     m.def("attn_fwd", &attn_fwd, "fused small-S attention forward");
     m.def("attn_bwd", &attn_bwd, "fused small-S attention backward");
+    m.def("attn_fwd_block", &attn_fwd_block,
+          "blockwise (flash-style) attention forward, any S");
+    m.def("attn_bwd_block", &attn_bwd_block,
+          "blockwise attention backward (deterministic two-pass)");
     m.def("ln_fwd", &ln_fwd, "fused per-client LayerNorm forward");
     m.def("ln_bwd", &ln_bwd, "fused per-client LayerNorm backward");
     m.def("lm_ce_fwd", &lm_ce_fwd, "fused vocab-masked LM CE forward");

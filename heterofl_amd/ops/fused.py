@@ -237,24 +237,38 @@ class GraphClipSGD:
 
 
 class _FusedAttention(torch.autograd.Function):
-    """Whole-(batch*head) fused attention for bptt<=64, head_dim<=32
-    (ops/csrc/attention.hip): QK^T/temp -> softmax -> PV in one kernel;
-    saves the bf16 softmax matrix for the one-kernel backward."""
+    """Fused attention (ops/csrc/attention.hip), head_dim<=32.  S<=64 runs
+    the whole-(batch*head) single-kernel path saving the bf16 softmax
+    matrix; larger S runs the blockwise (flash-style) path — 64-row KV
+    tiles with an online softmax, per-row LSE saved, and a deterministic
+    two-pass backward that recomputes P (SURVEY §5 long-context
+    readiness)."""
 
     @staticmethod
     def forward(ctx, q, k, v, temperature):
         ext = require_native()
         q, k, v = q.contiguous(), k.contiguous(), v.contiguous()
-        out, p = ext.attn_fwd(q, k, v, temperature)
-        ctx.save_for_backward(q, k, v, p)
+        if q.size(1) <= 64:
+            out, p = ext.attn_fwd(q, k, v, temperature)
+            ctx.save_for_backward(q, k, v, p)
+            ctx.blockwise = False
+        else:
+            out, lse = ext.attn_fwd_block(q, k, v, temperature)
+            ctx.save_for_backward(q, k, v, out, lse)
+            ctx.blockwise = True
         ctx.temperature = temperature
         return out
 
     @staticmethod
     def backward(ctx, dout):
         ext = require_native()
-        q, k, v, p = ctx.saved_tensors
-        dq, dk, dv = ext.attn_bwd(dout, q, k, v, p, ctx.temperature)
+        if ctx.blockwise:
+            q, k, v, out, lse = ctx.saved_tensors
+            dq, dk, dv = ext.attn_bwd_block(dout, q, k, v, out, lse,
+                                            ctx.temperature)
+        else:
+            q, k, v, p = ctx.saved_tensors
+            dq, dk, dv = ext.attn_bwd(dout, q, k, v, p, ctx.temperature)
         return dq, dk, dv, None
 
 
