@@ -707,7 +707,14 @@ class BatchedClientTrainer:
                     opt.step()
                 if logger is not None:
                     with torch.no_grad():
-                        pred = scores.argmax(dim=2)
+                        # accuracy over the MASKED scores, like the
+                        # reference metric (the model's forward rewrites
+                        # score before Metric sees it,
+                        # src/models/resnet.py:152-157)
+                        sc = scores
+                        if masks is not None:
+                            sc = sc.masked_fill(masks.unsqueeze(0) == 0, 0)
+                        pred = sc.argmax(dim=2)
                         acc = (pred == yb).float().mean(0) * 100.0
                         for i in range(R):
                             logger.append({'Local-Loss': losses[i].item(),
