@@ -114,6 +114,9 @@ class BatchedLMClientTrainer:
                 losses.sum().backward()
                 if native:
                     fopt.step(1.0, lr, cfg['momentum'], cfg['weight_decay'])
+                    if use_shadows:
+                        from ..ops.fused import bump_rng
+                        bump_rng(device)
                 else:
                     per_client_clip_(params, R, 1.0)
                     opt.step()

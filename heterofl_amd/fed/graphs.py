@@ -267,6 +267,11 @@ class LMGraphedStep:
             if not self._capturing:
                 self.opt.bind(grads)
             self.opt.launch(1.0, self.lr, self.momentum, self.weight_decay)
+            if self._shadows:
+                # one seed-cell advance per step: the fused dropout sites
+                # draw (seed, salt, index)-keyed masks, fresh every replay
+                from ..ops.fused import bump_rng
+                bump_rng(self.device)
         else:
             raw = torch.autograd.grad(losses.sum(), self.params)
             with torch.no_grad():

@@ -65,11 +65,12 @@ std::vector<at::Tensor> head_bwd(at::Tensor dscores, at::Tensor pooled,
                                  int64_t W, bool bf16_feat, bool want_db);
 void rng_bump(at::Tensor seed);
 std::vector<at::Tensor> gelu_drop_fwd(at::Tensor x, at::Tensor seed,
-                                      double rate, double p);
+                                      int64_t salt, double rate, double p);
 at::Tensor gelu_drop_bwd(at::Tensor dy, at::Tensor x, at::Tensor mask,
                          double rate, double p);
 std::vector<at::Tensor> res_drop_fwd(at::Tensor src, at::Tensor h,
-                                     at::Tensor seed, double rate, double p);
+                                     at::Tensor seed, int64_t salt,
+                                     double rate, double p);
 at::Tensor drop_scale_bwd(at::Tensor dt, at::Tensor mask, double rate,
                           double p);
 

@@ -4,9 +4,12 @@
 // (reference semantics: src/models/transformer.py:112-119):
 //   gelu_drop:  y = dropout(gelu(x / rate), p)          [Scaler+GELU+drop]
 //   res_drop:   t = src + dropout(h / rate, p)          [Scaler+drop+residual]
-// The dropout mask is drawn by a counter-hash RNG keyed on a device u64
-// seed cell (bumped by its own tiny kernel so hipGraph replays draw fresh
-// masks) and SAVED as uint8 for the exact one-kernel backward.
+// The dropout mask is drawn by a counter-hash RNG keyed on (device u64
+// seed cell, per-call-site salt, element index).  The seed cell advances
+// ONCE per training step (rng_bump, launched by the step driver) so
+// hipGraph replays draw fresh masks; the per-site salt decorrelates the
+// many dropout sites within a step without per-call bump launches.
+// Masks are SAVED as uint8 for the exact one-kernel backward.
 #include "common.h"
 
 __device__ __forceinline__ uint32_t mix32(uint32_t h) {
@@ -18,9 +21,11 @@ __device__ __forceinline__ uint32_t mix32(uint32_t h) {
     return h;
 }
 
-__device__ __forceinline__ uint32_t rng_at(unsigned long long seed, long i) {
+__device__ __forceinline__ uint32_t rng_at(unsigned long long seed,
+                                           uint32_t salt, long i) {
     uint32_t h = (uint32_t)(seed & 0xFFFFFFFFull) * 0x9E3779B1u
                  + (uint32_t)(seed >> 32) * 0x85EBCA77u
+                 + salt * 0x27D4EB2Fu
                  + (uint32_t)i * 0xC2B2AE3Du + (uint32_t)(i >> 32);
     return mix32(h);
 }
@@ -42,7 +47,8 @@ template <typename T>
 __global__ void __launch_bounds__(256)
 gelu_drop_fwd_kernel(const T* __restrict__ x, T* __restrict__ y,
                      unsigned char* __restrict__ mask,
-                     const unsigned long long* __restrict__ seed, long n,
+                     const unsigned long long* __restrict__ seed,
+                     unsigned salt, long n,
                      float inv_rate, float p, float inv_keep) {
     const unsigned long long sd = *seed;
     const uint32_t thresh = (uint32_t)(p * 4294967296.0);
@@ -50,7 +56,7 @@ gelu_drop_fwd_kernel(const T* __restrict__ x, T* __restrict__ y,
          i += (long)gridDim.x * blockDim.x) {
         const float g = gelu_f(ld_f32(x + i) * inv_rate);
         unsigned char keep = 1;
-        if (thresh) keep = rng_at(sd, i) >= thresh;
+        if (thresh) keep = rng_at(sd, salt, i) >= thresh;
         if (mask) mask[i] = keep;
         st_f32(y + i, keep ? g * inv_keep : 0.f);
     }
@@ -75,14 +81,15 @@ template <typename T>
 __global__ void __launch_bounds__(256)
 res_drop_fwd_kernel(const T* __restrict__ src, const T* __restrict__ h,
                     T* __restrict__ t, unsigned char* __restrict__ mask,
-                    const unsigned long long* __restrict__ seed, long n,
+                    const unsigned long long* __restrict__ seed,
+                    unsigned salt, long n,
                     float inv_rate, float p, float inv_keep) {
     const unsigned long long sd = *seed;
     const uint32_t thresh = (uint32_t)(p * 4294967296.0);
     for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < n;
          i += (long)gridDim.x * blockDim.x) {
         unsigned char keep = 1;
-        if (thresh) keep = rng_at(sd, i) >= thresh;
+        if (thresh) keep = rng_at(sd, salt, i) >= thresh;
         if (mask) mask[i] = keep;
         const float hv = keep ? ld_f32(h + i) * inv_rate * inv_keep : 0.f;
         st_f32(t + i, ld_f32(src + i) + hv);
@@ -125,7 +132,7 @@ void rng_bump(at::Tensor seed) {
 }
 
 std::vector<at::Tensor> gelu_drop_fwd(at::Tensor x, at::Tensor seed,
-                                      double rate, double p) {
+                                      int64_t salt, double rate, double p) {
     TORCH_CHECK(x.is_cuda() && x.is_contiguous());
     const long n = x.numel();
     auto y = at::empty_like(x);
@@ -138,11 +145,11 @@ std::vector<at::Tensor> gelu_drop_fwd(at::Tensor x, at::Tensor seed,
                            (const scalar_t*)x.data_ptr(),
                            (scalar_t*)y.data_ptr(),
                            dropping ? mask.data_ptr<unsigned char>() : nullptr,
-                           (const unsigned long long*)seed.data_ptr(), n,
+                           (const unsigned long long*)seed.data_ptr(),
+                           (unsigned)salt, n,
                            (float)(1.0 / rate), (float)p,
                            (float)(1.0 / (1.0 - p)));
     });
-    if (dropping) rng_bump(seed);
     return {y, mask};
 }
 
@@ -166,7 +173,8 @@ at::Tensor gelu_drop_bwd(at::Tensor dy, at::Tensor x, at::Tensor mask,
 }
 
 std::vector<at::Tensor> res_drop_fwd(at::Tensor src, at::Tensor h,
-                                     at::Tensor seed, double rate, double p) {
+                                     at::Tensor seed, int64_t salt,
+                                     double rate, double p) {
     TORCH_CHECK(src.is_cuda() && src.is_contiguous() && h.is_contiguous());
     TORCH_CHECK(src.numel() == h.numel());
     const long n = src.numel();
@@ -182,11 +190,11 @@ std::vector<at::Tensor> res_drop_fwd(at::Tensor src, at::Tensor h,
                            (const scalar_t*)h.data_ptr(),
                            (scalar_t*)t.data_ptr(),
                            dropping ? mask.data_ptr<unsigned char>() : nullptr,
-                           (const unsigned long long*)seed.data_ptr(), n,
+                           (const unsigned long long*)seed.data_ptr(),
+                           (unsigned)salt, n,
                            (float)(1.0 / rate), (float)p,
                            (float)(1.0 / (1.0 - p)));
     });
-    if (dropping) rng_bump(seed);
     return {t, mask};
 }
 

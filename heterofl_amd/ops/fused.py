@@ -333,6 +333,23 @@ def fused_lm_ce(logits, tokens, mask, R):
 
 # ------------------------------------------------------- LM glue fusions
 _rng_cells = {}
+_rng_salt = [0]
+
+
+def _next_salt():
+    """Per-call-site salt: decorrelates the many dropout sites inside one
+    step without a bump launch per call.  In eager mode the counter also
+    advances across steps; in a captured graph the salts freeze at capture
+    and the per-step rng_bump (launched by the step driver) moves the seed
+    cell instead."""
+    _rng_salt[0] = (_rng_salt[0] + 1) & 0x7FFFFFFF
+    return _rng_salt[0]
+
+
+def bump_rng(device):
+    """Advance the device RNG seed cell (once per training step)."""
+    ext = require_native()
+    ext.rng_bump(_rng_cell(device))
 
 
 def _rng_cell(device):
@@ -356,7 +373,8 @@ class _FusedGeluDropout(torch.autograd.Function):
     def forward(ctx, x, rate, p):
         ext = require_native()
         x = x.contiguous()
-        y, mask = ext.gelu_drop_fwd(x, _rng_cell(x.device), rate, p)
+        y, mask = ext.gelu_drop_fwd(x, _rng_cell(x.device), _next_salt(),
+                                    rate, p)
         ctx.save_for_backward(x, mask)
         ctx.meta = (rate, p)
         return y
@@ -381,7 +399,8 @@ class _FusedResDropout(torch.autograd.Function):
     def forward(ctx, src, h, rate, p):
         ext = require_native()
         t, mask = ext.res_drop_fwd(src.contiguous(), h.contiguous(),
-                                   _rng_cell(src.device), rate, p)
+                                   _rng_cell(src.device), _next_salt(),
+                                   rate, p)
         ctx.save_for_backward(mask)
         ctx.meta = (rate, p)
         return t
