@@ -894,14 +894,14 @@ def test_fast_eval_matches_loader_eval(base_cfg, monkeypatch):
     from heterofl_amd.logger import Logger
     from heterofl_amd.models import make_model
     from heterofl_amd.utils import process_dataset, make_optimizer
-    cfg = make_cfg(base_cfg, '1_6_0.5_non-iid-2_fix_a1_bn_1_1',
+    cfg = make_cfg(base_cfg, '1_5_0.4_non-iid-2_fix_a1_bn_1_1',
                    data_name='CIFAR10', model_name='resnet18')
     cfg['device'] = 'cuda:0'
     cfg['engine'] = 'batched'
     torch.manual_seed(0)
     ds = fetch_dataset('CIFAR10', synthetic=True, synthetic_size=240)
     process_dataset(ds, cfg)
-    data_split, label_split = split_dataset(ds, 6, 'non-iid-2', 10)
+    data_split, label_split = split_dataset(ds, 5, 'non-iid-2', 10)
     model = make_model(cfg).to('cuda:0')
     runner = FedRunner(cfg, ds, data_split, label_split, model,
                        make_optimizer(model, cfg['lr'], cfg))
