@@ -170,6 +170,16 @@ class BNormReLU(nn.Module):
                                       self.groups, r)
 
 
+class BMaxPool2(nn.Module):
+    """2x2/stride-2 MaxPool: hand kernel on GPU (K6), torch on CPU."""
+
+    def forward(self, x):
+        if native_ops.use_native(x):
+            from ..ops.fused import fused_maxpool2
+            return fused_maxpool2(x)
+        return F.max_pool2d(x, 2)
+
+
 class BScaler(nn.Module):
     def __init__(self, rate):
         super().__init__()
@@ -297,7 +307,7 @@ class BatchedConv(nn.Module):
             blocks.append(BNormReLU(R, out_ch, norm, rate, scale))
             blocks.append(nn.Identity())
             if i != len(hidden_size) - 1:
-                blocks.append(nn.MaxPool2d(2))
+                blocks.append(BMaxPool2())
             in_ch = out_ch
         self.blocks = nn.Sequential(*blocks)
         # final linear occupies the tail indices like the local model

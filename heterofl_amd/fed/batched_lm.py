@@ -271,9 +271,15 @@ class BatchedTransformer(nn.Module):
 
     def forward(self, tokens):
         """tokens (R, B, S) -> logits (R, B, S, V)."""
-        mask = torch.bernoulli(
-            torch.full(tokens.shape, self.mask_rate, device=tokens.device))
-        src = tokens.masked_fill(mask == 1, self.num_tokens).detach()
+        if native_ops.use_native(tokens):
+            # one-kernel Bernoulli masking (K11), graph-replay safe
+            from ..ops.fused import fused_token_mask
+            src = fused_token_mask(tokens, self.mask_rate, self.num_tokens)
+        else:
+            mask = torch.bernoulli(
+                torch.full(tokens.shape, self.mask_rate,
+                           device=tokens.device))
+            src = tokens.masked_fill(mask == 1, self.num_tokens).detach()
         x = self.transformer_embedding(src)
         for layer in self.transformer_encoder['layers']:
             x = layer(x)

@@ -73,6 +73,11 @@ std::vector<at::Tensor> res_drop_fwd(at::Tensor src, at::Tensor h,
                                      double rate, double p);
 at::Tensor drop_scale_bwd(at::Tensor dt, at::Tensor mask, double rate,
                           double p);
+at::Tensor token_mask(at::Tensor tokens, at::Tensor seed, int64_t salt,
+                      double rate, int64_t mask_id);
+std::vector<at::Tensor> maxpool2_fwd(at::Tensor x);
+at::Tensor maxpool2_bwd(at::Tensor dy, at::Tensor arg, int64_t H,
+                        int64_t W);
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("bn_relu_fwd", &bn_relu_fwd, "fused sBN+ReLU forward");
@@ -110,4 +115,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
           "fused scaler+dropout+residual-add fwd");
     m.def("drop_scale_bwd", &drop_scale_bwd,
           "dropout-mask * 1/rate backward");
+    m.def("token_mask", &token_mask, "Bernoulli token masking (in-kernel RNG)");
+    m.def("maxpool2_fwd", &maxpool2_fwd, "2x2/2 MaxPool forward (saves argmax)");
+    m.def("maxpool2_bwd", &maxpool2_bwd, "2x2/2 MaxPool backward");
 }

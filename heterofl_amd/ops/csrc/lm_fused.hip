@@ -215,3 +215,33 @@ at::Tensor drop_scale_bwd(at::Tensor dt, at::Tensor mask, double rate,
     });
     return dh;
 }
+
+// ------------------------- K11: in-kernel Bernoulli token masking --------
+// reference: src/models/transformer.py:149-151 — Bernoulli(mask_rate) mask
+// -> fill with the <mask> token id.  One kernel instead of bernoulli +
+// masked_fill (+ fill); same (seed, salt, index) RNG as the dropouts, so
+// hipGraph replays draw fresh masks.
+__global__ void __launch_bounds__(256)
+token_mask_kernel(const long* __restrict__ in, long* __restrict__ out,
+                  const unsigned long long* __restrict__ seed, unsigned salt,
+                  long n, float rate, long mask_id) {
+    const unsigned long long sd = *seed;
+    const uint32_t thresh = (uint32_t)(rate * 4294967296.0);
+    for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+         i += (long)gridDim.x * blockDim.x)
+        out[i] = rng_at(sd, salt, i) < thresh ? mask_id : in[i];
+}
+
+at::Tensor token_mask(at::Tensor tokens, at::Tensor seed, int64_t salt,
+                      double rate, int64_t mask_id) {
+    TORCH_CHECK(tokens.is_cuda() && tokens.is_contiguous() &&
+                tokens.scalar_type() == at::kLong);
+    const long n = tokens.numel();
+    auto out = at::empty_like(tokens);
+    auto stream = at::hip::getCurrentHIPStream();
+    hipLaunchKernelGGL(token_mask_kernel, lmf_grid(n), dim3(256), 0, stream,
+                       tokens.data_ptr<long>(), out.data_ptr<long>(),
+                       (const unsigned long long*)seed.data_ptr(),
+                       (unsigned)salt, n, (float)rate, (long)mask_id);
+    return out;
+}

@@ -440,3 +440,35 @@ class _FusedScaler(torch.autograd.Function):
 
 def fused_scaler(x, rate):
     return _FusedScaler.apply(x, rate)
+
+
+def fused_token_mask(tokens, rate, mask_id):
+    """Bernoulli(rate) token masking in one kernel (K11; reference:
+    src/models/transformer.py:149-151) — (seed, salt, index)-keyed RNG,
+    graph-replay safe."""
+    ext = require_native()
+    return ext.token_mask(tokens.contiguous(), _rng_cell(tokens.device),
+                          _next_salt(), rate, mask_id)
+
+
+class _FusedMaxPool2(torch.autograd.Function):
+    """2x2/stride-2 MaxPool (K6; reference src/models/conv.py:33): forward
+    saves the winning-corner index, backward scatters to it."""
+
+    @staticmethod
+    def forward(ctx, x):
+        ext = require_native()
+        y, arg = ext.maxpool2_fwd(x.contiguous())
+        ctx.save_for_backward(arg)
+        ctx.hw = (x.size(2), x.size(3))
+        return y
+
+    @staticmethod
+    def backward(ctx, dy):
+        ext = require_native()
+        (arg,) = ctx.saved_tensors
+        return ext.maxpool2_bwd(dy, arg, *ctx.hw)
+
+
+def fused_maxpool2(x):
+    return _FusedMaxPool2.apply(x)

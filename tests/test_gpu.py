@@ -986,3 +986,32 @@ def test_blockwise_attention_agrees_with_small_s_path():
     out1, _ = ext.attn_fwd(q, k, v, d ** 0.5)
     out2, _ = ext.attn_fwd_block(q, k, v, d ** 0.5)
     assert (out1 - out2).abs().max().item() < 2e-2
+
+
+@needs_gpu
+def test_fused_maxpool_and_token_mask():
+    """K6 MaxPool kernel vs F.max_pool2d (fwd + exact scatter bwd) and K11
+    in-kernel Bernoulli token masking statistics/value correctness."""
+    from heterofl_amd.ops.fused import fused_maxpool2, fused_token_mask
+    import torch.nn.functional as F
+    torch.manual_seed(0)
+    for dt in (torch.float32, torch.bfloat16):
+        for shape in ((6, 40, 32, 32), (3, 8, 7, 7)):
+            x = torch.randn(*shape, device='cuda:0', dtype=dt,
+                            requires_grad=True)
+            x2 = x.detach().clone().requires_grad_(True)
+            y = fused_maxpool2(x)
+            ref = F.max_pool2d(x2, 2)
+            assert torch.equal(y, ref), (dt, shape)
+            g = torch.randn_like(y)
+            y.backward(g)
+            ref.backward(g)
+            assert torch.equal(x.grad, x2.grad), (dt, shape)
+    toks = torch.randint(0, 1000, (10, 4, 64), device='cuda:0')
+    out = fused_token_mask(toks, 0.15, 1000)
+    frac = (out == 1000).float().mean().item()
+    assert 0.10 < frac < 0.20, frac
+    keep = out != 1000
+    assert torch.equal(out[keep], toks[keep])
+    out2 = fused_token_mask(toks, 0.15, 1000)
+    assert not torch.equal(out == 1000, out2 == 1000)  # fresh draw (salt)
