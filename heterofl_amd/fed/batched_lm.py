@@ -166,6 +166,14 @@ class STransformerEmbedding(nn.Module):
 
     def forward(self, src):
         R, B, S = src.shape
+        if native_ops.use_native(src):
+            from ..ops.fused import fused_embed_pos
+            emb = self.embedding
+            pe = self.positional_embedding['positional_embedding']
+            r = self.scaler.rate if self.training else 1.0
+            x = fused_embed_pos(src, emb.weight, pe.weight, emb.w16,
+                                pe.w16, r)
+            return self.dropout(self.norm(x))
         pos = torch.arange(S, device=src.device).view(1, 1, S).expand(R, B, S)
         x = self.scaler(self.embedding(src)) + \
             self.scaler(self.positional_embedding['positional_embedding'](pos))

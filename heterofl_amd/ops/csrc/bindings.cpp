@@ -75,6 +75,10 @@ at::Tensor drop_scale_bwd(at::Tensor dt, at::Tensor mask, double rate,
                           double p);
 at::Tensor token_mask(at::Tensor tokens, at::Tensor seed, int64_t salt,
                       double rate, int64_t mask_id);
+at::Tensor embed_pos_fwd(at::Tensor ids, at::Tensor table, at::Tensor pos,
+                         double rate);
+std::vector<at::Tensor> embed_pos_bwd(at::Tensor dy, at::Tensor ids,
+                                      int64_t V, int64_t P, double rate);
 std::vector<at::Tensor> maxpool2_fwd(at::Tensor x);
 at::Tensor maxpool2_bwd(at::Tensor dy, at::Tensor arg, int64_t H,
                         int64_t W);
@@ -116,6 +120,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("drop_scale_bwd", &drop_scale_bwd,
           "dropout-mask * 1/rate backward");
     m.def("token_mask", &token_mask, "Bernoulli token masking (in-kernel RNG)");
+    m.def("embed_pos_fwd", &embed_pos_fwd,
+          "fused token+positional embedding gather with Scaler");
+    m.def("embed_pos_bwd", &embed_pos_bwd,
+          "deterministic embedding backward (fp32 master grads)");
     m.def("maxpool2_fwd", &maxpool2_fwd, "2x2/2 MaxPool forward (saves argmax)");
     m.def("maxpool2_bwd", &maxpool2_bwd, "2x2/2 MaxPool backward");
 }

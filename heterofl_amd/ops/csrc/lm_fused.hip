@@ -245,3 +245,152 @@ at::Tensor token_mask(at::Tensor tokens, at::Tensor seed, int64_t salt,
                        (unsigned)salt, n, (float)rate, (long)mask_id);
     return out;
 }
+
+// ----------------- K9: fused embedding + positional + Scaler gather ------
+// reference: src/models/transformer.py:29-37 — scaler(tok_emb(src)) +
+// scaler(pos_emb(pos)); one gather-add-scale kernel forward, and a
+// DETERMINISTIC backward (per-(r,vocab-row) reduction in fixed (b,s) scan
+// order — no atomics, so hipGraph replays stay bit-identical) that writes
+// fp32 grads straight for the masters (no shadow-upcast kernels).
+template <typename TT>
+__global__ void __launch_bounds__(256)
+embed_pos_fwd_kernel(const long* __restrict__ ids,
+                     const TT* __restrict__ table,
+                     const TT* __restrict__ pos, TT* __restrict__ out,
+                     int R, int B, int S, int E, int V, int P,
+                     float inv_rate) {
+    const long rows = (long)R * B * S;
+    for (long i = (long)blockIdx.x * (blockDim.x / WAVE)
+                  + threadIdx.x / WAVE;
+         i < rows; i += (long)gridDim.x * (blockDim.x / WAVE)) {
+        const int l = threadIdx.x & (WAVE - 1);
+        const int r = (int)(i / ((long)B * S));
+        const int s = (int)(i % S);
+        const long id = ids[i];
+        const TT* src = table + ((long)r * V + id) * E;
+        const TT* pp = pos + ((long)r * P + s) * E;
+        TT* o = out + i * E;
+        for (int e = l; e < E; e += WAVE)
+            o[e] = (TT)((ld_f32(src + e) + ld_f32(pp + e)) * inv_rate);
+    }
+}
+
+// dtable[r, v, :] = inv_rate * sum over (b,s) with ids==v of dy rows.
+// One wave owns one (r, v) pair; scans the B*S ids in fixed order.
+template <typename TT>
+__global__ void __launch_bounds__(256)
+embed_bwd_table_kernel(const long* __restrict__ ids,
+                       const TT* __restrict__ dy,
+                       float* __restrict__ dtable, int R, int BS, int E,
+                       int V, float inv_rate) {
+    const long pairs = (long)R * V;
+    const int l = threadIdx.x & (WAVE - 1);
+    for (long p = (long)blockIdx.x * (blockDim.x / WAVE)
+                  + threadIdx.x / WAVE;
+         p < pairs; p += (long)gridDim.x * (blockDim.x / WAVE)) {
+        const int r = (int)(p / V);
+        const long v = p % V;
+        float acc[16];
+        const int ne = (E + WAVE - 1) / WAVE;
+#pragma unroll
+        for (int j = 0; j < 16; ++j) acc[j] = 0.f;
+        bool any = false;
+        const long* idr = ids + (long)r * BS;
+        for (int t = 0; t < BS; ++t) {
+            if (idr[t] == v) {
+                any = true;
+                const TT* dyr = dy + ((long)r * BS + t) * E;
+                for (int j = 0; j < ne; ++j) {
+                    const int e = l + j * WAVE;
+                    if (e < E) acc[j] += ld_f32(dyr + e);
+                }
+            }
+        }
+        if (any)
+            for (int j = 0; j < ne; ++j) {
+                const int e = l + j * WAVE;
+                if (e < E)
+                    dtable[((long)r * V + v) * E + e] = acc[j] * inv_rate;
+            }
+    }
+}
+
+// dpos[r, s, :] = inv_rate * sum over b of dy[r, b, s, :]
+template <typename TT>
+__global__ void __launch_bounds__(256)
+embed_bwd_pos_kernel(const TT* __restrict__ dy, float* __restrict__ dpos,
+                     int R, int B, int S, int E, int P, float inv_rate) {
+    const long rows = (long)R * S;
+    const int l = threadIdx.x & (WAVE - 1);
+    for (long p = (long)blockIdx.x * (blockDim.x / WAVE)
+                  + threadIdx.x / WAVE;
+         p < rows; p += (long)gridDim.x * (blockDim.x / WAVE)) {
+        const int r = (int)(p / S);
+        const int s = (int)(p % S);
+        for (int e = l; e < E; e += WAVE) {
+            float acc = 0.f;
+            for (int b = 0; b < B; ++b)
+                acc += ld_f32(dy + (((long)r * B + b) * S + s) * E + e);
+            dpos[((long)r * P + s) * E + e] = acc * inv_rate;
+        }
+    }
+}
+
+#define DISPATCH_EMB(t, ...)                                                  \
+    if ((t) == at::kFloat) { using emb_t = float; __VA_ARGS__; }              \
+    else if ((t) == at::kBFloat16) { using emb_t = __hip_bfloat16; __VA_ARGS__; } \
+    else { TORCH_CHECK(false, "unsupported dtype"); }
+
+at::Tensor embed_pos_fwd(at::Tensor ids, at::Tensor table, at::Tensor pos,
+                         double rate) {
+    TORCH_CHECK(ids.is_cuda() && ids.is_contiguous() &&
+                table.is_contiguous() && pos.is_contiguous());
+    const int R = table.size(0), V = table.size(1), E = table.size(2);
+    const int P = pos.size(1);
+    const int B = ids.size(1), S = ids.size(2);
+    TORCH_CHECK(E <= 16 * WAVE, "embedding dim too large for bwd acc");
+    auto out = at::empty({(long)R, B, S, E}, table.options());
+    auto stream = at::hip::getCurrentHIPStream();
+    const long rows = (long)R * B * S;
+    long blocks = (rows + 3) / 4;
+    if (blocks > 4096) blocks = 4096;
+    DISPATCH_EMB(table.scalar_type(), {
+        hipLaunchKernelGGL(embed_pos_fwd_kernel<emb_t>,
+                           dim3((unsigned)blocks), dim3(256), 0, stream,
+                           ids.data_ptr<long>(),
+                           (const emb_t*)table.data_ptr(),
+                           (const emb_t*)pos.data_ptr(),
+                           (emb_t*)out.data_ptr(), R, B, S, E, V, P,
+                           (float)(1.0 / rate));
+    });
+    return out;
+}
+
+std::vector<at::Tensor> embed_pos_bwd(at::Tensor dy, at::Tensor ids,
+                                      int64_t V, int64_t P, double rate) {
+    const int R = ids.size(0), B = ids.size(1), S = ids.size(2);
+    const int E = dy.size(3);
+    auto dyc = dy.contiguous();
+    auto opts = dy.options().dtype(at::kFloat);
+    auto dtable = at::zeros({(long)R, V, (long)E}, opts);
+    auto dpos = at::zeros({(long)R, P, (long)E}, opts);
+    auto stream = at::hip::getCurrentHIPStream();
+    long pb = ((long)R * V + 3) / 4;
+    if (pb > 4096) pb = 4096;
+    long sb = ((long)R * S + 3) / 4;
+    if (sb > 4096) sb = 4096;
+    DISPATCH_EMB(dy.scalar_type(), {
+        hipLaunchKernelGGL(embed_bwd_table_kernel<emb_t>,
+                           dim3((unsigned)pb), dim3(256), 0, stream,
+                           ids.data_ptr<long>(),
+                           (const emb_t*)dyc.data_ptr(),
+                           dtable.data_ptr<float>(), R, B * S, E, (int)V,
+                           (float)(1.0 / rate));
+        hipLaunchKernelGGL(embed_bwd_pos_kernel<emb_t>,
+                           dim3((unsigned)sb), dim3(256), 0, stream,
+                           (const emb_t*)dyc.data_ptr(),
+                           dpos.data_ptr<float>(), R, B, S, E, (int)P,
+                           (float)(1.0 / rate));
+    });
+    return {dtable, dpos};
+}

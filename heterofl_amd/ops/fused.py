@@ -472,3 +472,35 @@ class _FusedMaxPool2(torch.autograd.Function):
 
 def fused_maxpool2(x):
     return _FusedMaxPool2.apply(x)
+
+
+class _FusedEmbedPos(torch.autograd.Function):
+    """Fused token-embedding + positional-embedding gather with the Scaler
+    folded in (K9; reference: src/models/transformer.py:29-37).  Reads the
+    bf16 shadow tables when present; the deterministic backward writes fp32
+    grads straight for the fp32 masters (no shadow-upcast kernels, no
+    atomics — bit-stable under graph replay)."""
+
+    @staticmethod
+    def forward(ctx, ids, table, pos, t16, p16, rate):
+        ext = require_native()
+        use16 = t16 is not None and p16 is not None
+        t = t16 if use16 else table
+        p = p16 if use16 else pos
+        out = ext.embed_pos_fwd(ids.contiguous(), t.contiguous(),
+                                p.contiguous(), rate)
+        ctx.save_for_backward(ids)
+        ctx.meta = (table.size(1), pos.size(1), rate)
+        return out
+
+    @staticmethod
+    def backward(ctx, dy):
+        ext = require_native()
+        (ids,) = ctx.saved_tensors
+        V, P, rate = ctx.meta
+        dtable, dpos = ext.embed_pos_bwd(dy, ids, V, P, rate)
+        return None, dtable, dpos, None, None, None
+
+
+def fused_embed_pos(ids, table, pos, t16, p16, rate):
+    return _FusedEmbedPos.apply(ids, table, pos, t16, p16, rate)

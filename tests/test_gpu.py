@@ -1015,3 +1015,47 @@ def test_fused_maxpool_and_token_mask():
     assert torch.equal(out[keep], toks[keep])
     out2 = fused_token_mask(toks, 0.15, 1000)
     assert not torch.equal(out == 1000, out2 == 1000)  # fresh draw (salt)
+
+
+@needs_gpu
+def test_fused_embed_pos_matches_eager():
+    """K9 fused embedding+positional+Scaler gather vs the eager chain —
+    forward and the deterministic fp32 master grads, fp32 and bf16-shadow
+    routes."""
+    from heterofl_amd.ops.fused import fused_embed_pos
+    torch.manual_seed(0)
+    R, B, S, E, V, P = 3, 2, 64, 96, 300, 64
+    rate = 0.5
+    for shadows in (False, True):
+        table = torch.randn(R, V, E, device='cuda:0',
+                            requires_grad=True)
+        pos = torch.randn(R, P, E, device='cuda:0', requires_grad=True)
+        ids = torch.randint(0, V, (R, B, S), device='cuda:0')
+        t16 = table.detach().to(torch.bfloat16).contiguous() if shadows \
+            else None
+        p16 = pos.detach().to(torch.bfloat16).contiguous() if shadows \
+            else None
+        y = fused_embed_pos(ids, table, pos, t16, p16, rate)
+        # eager reference on the same effective (possibly quantized) tables
+        t_ref = (t16.float() if shadows else table.detach()) \
+            .clone().requires_grad_(True)
+        p_ref = (p16.float() if shadows else pos.detach()) \
+            .clone().requires_grad_(True)
+        pidx = torch.arange(S, device='cuda:0').view(1, 1, S).expand(R, B, S)
+        off = (torch.arange(R, device='cuda:0') * V).view(R, 1, 1)
+        poff = (torch.arange(R, device='cuda:0') * P).view(R, 1, 1)
+        ref = (torch.nn.functional.embedding((ids + off).reshape(-1),
+                                             t_ref.reshape(R * V, E))
+               .reshape(R, B, S, E) / rate
+               + torch.nn.functional.embedding((pidx + poff).reshape(-1),
+                                               p_ref.reshape(R * P, E))
+               .reshape(R, B, S, E) / rate)
+        tol = 2e-2 if shadows else 1e-4
+        assert (y.float() - ref).abs().max().item() < tol, shadows
+        g = torch.randn_like(ref)
+        y.backward(g.to(y.dtype))
+        ref.backward(g)
+        terr = (table.grad - t_ref.grad).abs().max().item()
+        perr = (pos.grad - p_ref.grad).abs().max().item()
+        assert terr < tol * 4, (shadows, terr)
+        assert perr < tol * 4, (shadows, perr)
