@@ -269,12 +269,23 @@ class Federation:
         generator for multi-rank determinism).  ``slots`` restricts the
         tensor slicing to this rank's clients (index maps are still built
         for every slot — combine needs only the local ones, but they are
-        cheap); other slots get None."""
+        cheap); other slots get None.
+
+        On GPU with the native extension, every dense-prefix slice is
+        packed by ONE HIP kernel launch over a cached descriptor table
+        (K13 of SURVEY §2b) instead of per-parameter narrow+clone calls;
+        transformer per-head GATHER slices and non-fp32 buffers keep the
+        per-tensor path."""
         if resample:
             self.make_model_rate(generator)
         param_idx = self.split_model(user_idx)
-        want = range(len(user_idx)) if slots is None else slots
+        want = list(range(len(user_idx))) if slots is None else list(slots)
         local_parameters = [None] * len(user_idx)
+        packed = self._pack_distribute(user_idx, param_idx, want)
+        if packed is not None:
+            for m, lp in packed.items():
+                local_parameters[m] = lp
+            return local_parameters, param_idx
         for m in want:
             lp = OrderedDict()
             for k, v in self.global_parameters.items():
@@ -285,6 +296,85 @@ class Federation:
                     lp[k] = v.clone()
             local_parameters[m] = lp
         return local_parameters, param_idx
+
+    def _pack_distribute(self, user_idx, param_idx, want):
+        """One-kernel slice pack (see distribute).  Returns
+        {slot: OrderedDict} or None when inapplicable.  Descriptors and
+        destination buffers are cached per (slot -> rate) assignment —
+        global master data_ptrs are stable (finalize writes in place)."""
+        from .. import ops as native_ops
+        gp = self.global_parameters
+        first = next(iter(gp.values()))
+        if not (first.is_cuda and native_ops.use_native(first)
+                and native_ops.native_available()):
+            return None
+        key = tuple((m, self.model_rate[user_idx[m]]) for m in want)
+        cache = getattr(self, '_pack_cache', None)
+        if cache is None:
+            cache = self._pack_cache = {}
+        hit = cache.get(key)
+        if hit is None:
+            import numpy as np
+            desc_dt = np.dtype([('src', 'u8'), ('dst', 'u8'), ('rows', 'i4'),
+                                ('cols', 'i4'), ('stride', 'i4'),
+                                ('pad', 'i4')])
+            descs = []
+            outs = {}
+            extras = []   # (slot, key, ps) handled per-tensor (gather/non-f32)
+            max_rows = 1
+            for m in want:
+                lp = OrderedDict()
+                for k, v in gp.items():
+                    ptype = k.split('.')[-1]
+                    ps = param_idx[m][k] if ('weight' in ptype
+                                             or 'bias' in ptype) else None
+                    sliceable = (v.dtype == torch.float32
+                                 and ps is not None and ps.out is not None
+                                 and ps.out.is_dense_prefix()
+                                 and (ps.inp is None
+                                      or ps.inp.is_dense_prefix()))
+                    if ('weight' in ptype or 'bias' in ptype) and sliceable:
+                        if v.dim() > 1 and ps.inp is not None:
+                            rows = ps.out.n
+                            cols = ps.inp.n * (v.numel() // (v.size(0)
+                                                             * v.size(1)))
+                            stride = v.numel() // v.size(0)
+                            # inner-dims contiguity: prefix on dim 1 of a
+                            # 4-D conv weight is contiguous only if we copy
+                            # (inp.n * kh * kw) elems from each row start —
+                            # true because dims 2+ are always FULL
+                            dst = torch.empty(
+                                (rows,) + (ps.inp.n,) + tuple(v.shape[2:]),
+                                dtype=v.dtype, device=v.device)
+                        else:
+                            rows = 1
+                            cols = ps.out.n
+                            stride = cols
+                            dst = torch.empty(ps.out.n, dtype=v.dtype,
+                                              device=v.device)
+                        descs.append((v.data_ptr(), dst.data_ptr(), rows,
+                                      cols, stride, 0))
+                        max_rows = max(max_rows, rows)
+                        lp[k] = dst
+                    else:
+                        extras.append((m, k, ps))
+                        lp[k] = None
+                outs[m] = lp
+            blob = torch.from_numpy(
+                np.array(descs, dtype=desc_dt).view(np.uint8)).to(
+                    first.device)
+            hit = cache[key] = (blob, len(descs), max_rows, outs, extras)
+        blob, n, max_rows, outs, extras = hit
+        ext = native_ops.require_native()
+        ext.pack_slices(blob, n, max_rows)
+        for m, k, ps in extras:
+            v = gp[k]
+            ptype = k.split('.')[-1]
+            if 'weight' in ptype or 'bias' in ptype:
+                outs[m][k] = self._slice_value(v, ps)
+            else:
+                outs[m][k] = v.clone()
+        return outs
 
     # --------------------------------------------------------------- combine
     def _output_layer_kind(self, k, weight_keys, bias_keys):

@@ -80,6 +80,7 @@ at::Tensor embed_pos_fwd(at::Tensor ids, at::Tensor table, at::Tensor pos,
 std::vector<at::Tensor> embed_pos_bwd(at::Tensor dy, at::Tensor ids,
                                       int64_t V, int64_t P, double rate);
 std::vector<at::Tensor> maxpool2_fwd(at::Tensor x);
+void pack_slices(at::Tensor blob, int64_t n, int64_t max_rows);
 at::Tensor maxpool2_bwd(at::Tensor dy, at::Tensor arg, int64_t H,
                         int64_t W);
 
@@ -125,5 +126,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("embed_pos_bwd", &embed_pos_bwd,
           "deterministic embedding backward (fp32 master grads)");
     m.def("maxpool2_fwd", &maxpool2_fwd, "2x2/2 MaxPool forward (saves argmax)");
+    m.def("pack_slices", &pack_slices,
+          "one-launch distribute slice pack from a descriptor table");
     m.def("maxpool2_bwd", &maxpool2_bwd, "2x2/2 MaxPool backward");
 }

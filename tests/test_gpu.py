@@ -1050,7 +1050,9 @@ def test_fused_embed_pos_matches_eager():
                + torch.nn.functional.embedding((pidx + poff).reshape(-1),
                                                p_ref.reshape(R * P, E))
                .reshape(R, B, S, E) / rate)
-        tol = 2e-2 if shadows else 1e-4
+        # bf16 route: the kernel rounds the fused (a+b)/rate once to bf16,
+        # the fp32 reference does not — one output ulp at |y|<=8 is 0.0625
+        tol = 7e-2 if shadows else 1e-4
         assert (y.float() - ref).abs().max().item() < tol, shadows
         g = torch.randn_like(ref)
         y.backward(g.to(y.dtype))
@@ -1059,3 +1061,35 @@ def test_fused_embed_pos_matches_eager():
         perr = (pos.grad - p_ref.grad).abs().max().item()
         assert terr < tol * 4, (shadows, terr)
         assert perr < tol * 4, (shadows, perr)
+
+
+@needs_gpu
+def test_pack_distribute_matches_slice(base_cfg, monkeypatch):
+    """K13 one-kernel slice pack == per-tensor narrow/clone distribute,
+    resnet18 heterogeneous rates and transformer (per-head GATHER slices
+    take the per-tensor path inside the packed route)."""
+    from heterofl_amd.fed.federation import Federation
+    from heterofl_amd.models import make_model
+    for model_name, data_name in (('resnet18', 'CIFAR10'),
+                                  ('transformer', 'WikiText2')):
+        cfg = make_cfg(base_cfg, '1_4_1_iid_fix_a1-e1_bn_1_1',
+                       data_name=data_name, model_name=model_name)
+        if model_name == 'transformer':
+            cfg['num_tokens'] = 300
+            cfg['bptt'] = 64
+        torch.manual_seed(0)
+        model = make_model(cfg, model_rate=1.0).to('cuda:0')
+        label_split = {i: list(range(10)) for i in range(4)}
+        fed = Federation(model.state_dict(), cfg['model_rate'], label_split,
+                         cfg)
+        user_idx = [0, 1, 2, 3]
+        lp_pack, pidx = fed.distribute(user_idx, resample=False)
+        assert getattr(fed, '_pack_cache', None), 'pack path did not run'
+        monkeypatch.setenv('HETEROFL_FORCE_EAGER', '1')
+        lp_ref, _ = fed.distribute(user_idx, resample=False)
+        monkeypatch.delenv('HETEROFL_FORCE_EAGER')
+        for m in range(4):
+            for k in lp_ref[m]:
+                a, b = lp_pack[m][k], lp_ref[m][k]
+                assert a.shape == b.shape, (model_name, m, k)
+                assert torch.equal(a, b), (model_name, m, k)
