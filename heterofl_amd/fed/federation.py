@@ -311,9 +311,17 @@ class Federation:
         key = tuple((m, self.model_rate[user_idx[m]]) for m in want)
         cache = getattr(self, '_pack_cache', None)
         if cache is None:
-            cache = self._pack_cache = {}
+            from collections import OrderedDict as _OD
+            cache = self._pack_cache = _OD()
         hit = cache.get(key)
+        if hit is not None:
+            cache.move_to_end(key)
         if hit is None:
+            # dynamic mode draws a fresh rate assignment every round: cap
+            # the descriptor/buffer cache (each entry owns the slots' dst
+            # tensors) so it cannot grow with the round count
+            while len(cache) >= 48:
+                cache.popitem(last=False)
             import numpy as np
             desc_dt = np.dtype([('src', 'u8'), ('dst', 'u8'), ('rows', 'i4'),
                                 ('cols', 'i4'), ('stride', 'i4'),
@@ -360,13 +368,17 @@ class Federation:
                         extras.append((m, k, ps))
                         lp[k] = None
                 outs[m] = lp
-            blob = torch.from_numpy(
-                np.array(descs, dtype=desc_dt).view(np.uint8)).to(
-                    first.device)
+            if descs:
+                blob = torch.from_numpy(
+                    np.array(descs, dtype=desc_dt).view(np.uint8)).to(
+                        first.device)
+            else:
+                blob = torch.empty(0, dtype=torch.uint8, device=first.device)
             hit = cache[key] = (blob, len(descs), max_rows, outs, extras)
         blob, n, max_rows, outs, extras = hit
-        ext = native_ops.require_native()
-        ext.pack_slices(blob, n, max_rows)
+        if n:
+            ext = native_ops.require_native()
+            ext.pack_slices(blob, n, max_rows)
         for m, k, ps in extras:
             v = gp[k]
             ptype = k.split('.')[-1]
