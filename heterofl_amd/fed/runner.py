@@ -348,6 +348,9 @@ class FedRunner:
             import torch.distributed as dist
             parts, weights = [], []
             for _, m in bn_mods:
+                if m not in state:  # rank had zero batches (tiny dataset)
+                    z = torch.zeros(m.weight.numel(), device=device)
+                    state[m] = [z, z.clone(), 0]
                 mu, var, k = state[m]
                 parts.append(mu * k)
                 parts.append(var * k)
