@@ -416,3 +416,51 @@ def test_native_stats_pass_matches_eager(base_cfg, monkeypatch):
         if 'running' in k or 'num_batches' in k:
             diff = (sd_n[k].float() - sd_e[k].float()).abs().max().item()
             assert diff < 1e-4, (k, diff)
+
+
+def test_eval_every_knob(base_cfg, tmp_path, monkeypatch):
+    """HETEROFL_EVAL_EVERY=k evaluates every k-th round plus the final
+    round; skipped rounds never update the best-checkpoint pivot."""
+    import os
+    monkeypatch.chdir(tmp_path)
+    monkeypatch.setenv('HETEROFL_EVAL_EVERY', '2')
+    monkeypatch.setenv('HETEROFL_MAX_ROUNDS', '3')
+    monkeypatch.setenv('HETEROFL_SYNTHETIC_SIZE', '40')
+    from heterofl_amd.entry import run_fed_experiment
+    from heterofl_amd.config import default_config
+    from heterofl_amd.utils import load
+    cfg = default_config()
+    cfg.update({'data_name': 'MNIST', 'model_name': 'conv', 'device': 'cpu',
+                'engine': 'sequential', 'synthetic': True,
+                'num_experiments': 1, 'init_seed': 0, 'resume_mode': 0})
+    cfg['control'] = {'fed': '1', 'num_users': '4', 'frac': '0.5',
+                      'data_split_mode': 'iid', 'model_split_mode': 'fix',
+                      'model_mode': 'a1', 'norm': 'bn', 'scale': '1',
+                      'mask': '1'}
+    cfg['control_name'] = '1_4_0.5_iid_fix_a1_bn_1_1'
+    metric_name = {'train': {'Local': ['Local-Loss', 'Local-Accuracy']},
+                   'test': {'Local': ['Local-Loss', 'Local-Accuracy'],
+                            'Global': ['Global-Loss', 'Global-Accuracy']}}
+    run_fed_experiment(dict(cfg), 'Global-Accuracy', +1, metric_name)
+    tag = '0_MNIST_label_conv_1_4_0.5_iid_fix_a1_bn_1_1'
+    ck = load('./output/model/{}_checkpoint.pt'.format(tag))
+    hist = ck['logger'].history
+    # rounds 2 and 3 (final) evaluated; round 1 skipped
+    assert len(hist.get('test/Global-Accuracy', [])) == 2
+    assert os.path.exists('./output/model/{}_best.pt'.format(tag))
+
+
+def test_learnable_synthetic_mode(monkeypatch):
+    """HETEROFL_SYNTHETIC_MODE=learnable: deterministic class-template data,
+    train/test share the template bank but draw disjoint samples."""
+    import torch
+    from heterofl_amd.data import fetch_dataset
+    monkeypatch.setenv('HETEROFL_SYNTHETIC_MODE', 'learnable')
+    a = fetch_dataset('CIFAR10', synthetic=True, synthetic_size=300)
+    b = fetch_dataset('CIFAR10', synthetic=True, synthetic_size=300)
+    assert torch.equal(a['train'].img, b['train'].img)  # deterministic
+    assert a['train'].img.shape == (300, 32, 32, 3)
+    assert not torch.equal(a['train'].img[:60], a['test'].img)  # disjoint draw
+    # labels roughly balanced over 10 classes
+    counts = torch.bincount(torch.tensor(a['train'].target), minlength=10)
+    assert int(counts.min()) > 10
